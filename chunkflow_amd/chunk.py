@@ -1,0 +1,288 @@
+"""Chunk: an ndarray-or-tensor with a global voxel offset (the data model).
+
+Mirrors the behavior of the reference's chunkflow/chunk/base.py for the hot
+path methods — create :139-199 (sin/zero/random synthetic patterns), cutout
+:761-781, blend :792-807, crop_margin :691-726, save :783-790,
+mask_using_last_channel :685-689 — re-designed MI355X-first: a Chunk may be
+backed by a numpy array (host, pipeline edges) or a torch CUDA tensor
+(resident in HBM for the whole operator chain). Device-side arithmetic for
+the hot path lives in the HIP extension (chunkflow_amd/csrc), not here; the
+host methods below are the reference semantics at the pipeline boundary.
+"""
+from __future__ import annotations
+
+from numbers import Number
+from typing import Union
+
+import numpy as np
+
+from .cartesian import BoundingBox, Cartesian
+
+try:
+    import torch
+except ImportError:  # torch is a hard dep in practice; keep import-time soft
+    torch = None
+
+
+def _is_tensor(x) -> bool:
+    return torch is not None and isinstance(x, torch.Tensor)
+
+
+class Chunk:
+    def __init__(self, array, voxel_offset=None, voxel_size=None):
+        if isinstance(array, Chunk):
+            if voxel_offset is None:
+                voxel_offset = array.voxel_offset
+            if voxel_size is None:
+                voxel_size = array.voxel_size
+            array = array.array
+        assert isinstance(array, np.ndarray) or _is_tensor(array)
+        if array.ndim == 2:
+            array = array[None, ...]
+        self.array = array
+        if voxel_offset is None:
+            voxel_offset = Cartesian(0, 0, 0)
+        if len(voxel_offset) == 4:
+            assert voxel_offset[0] == 0
+            voxel_offset = voxel_offset[1:]
+        self.voxel_offset = Cartesian.from_collection(voxel_offset)
+        self.voxel_size = (Cartesian.from_collection(voxel_size)
+                           if voxel_size is not None else None)
+
+    # --- construction ------------------------------------------------------
+    @classmethod
+    def create(cls, size=(64, 64, 64), dtype=np.uint8,
+               voxel_offset=(0, 0, 0), voxel_size=None,
+               pattern='sin', high=255):
+        """Synthetic chunks for tests/benchmarks (reference
+        chunk/base.py:139-199; uint8/float patterns only — the integer
+        'sin'/'random' patterns that need connected-components labeling are
+        out of the hot-path scope)."""
+        dtype = np.dtype(dtype)
+        if pattern == 'zero':
+            arr = np.zeros(size, dtype=dtype)
+        elif pattern == 'sin':
+            ix, iy, iz = np.meshgrid(
+                *[np.linspace(0, 1, n) for n in size[-3:]], indexing='ij')
+            arr = np.abs(np.sin(4 * (ix + iy + iz)))
+            if len(size) == 4:
+                arr = np.repeat(arr[None, ...], size[0], axis=0)
+            if dtype == np.uint8:
+                arr = (arr * 255).astype(dtype)
+            elif np.issubdtype(dtype, np.floating):
+                arr = arr.astype(dtype)
+            else:
+                raise NotImplementedError(
+                    f'sin pattern for {dtype} needs segmentation labeling '
+                    '(out of hot-path scope)')
+        elif pattern == 'random':
+            if np.issubdtype(dtype, np.floating):
+                arr = np.random.rand(*size).astype(dtype)
+            elif np.issubdtype(dtype, np.integer):
+                arr = np.random.randint(high, size=size, dtype=dtype)
+            else:
+                raise NotImplementedError(dtype)
+        else:
+            raise NotImplementedError(pattern)
+        return cls(arr, voxel_offset=voxel_offset, voxel_size=voxel_size)
+
+    @classmethod
+    def from_bbox(cls, bbox: BoundingBox, dtype=np.uint8, pattern='zero',
+                  voxel_size=None):
+        return cls.create(size=tuple(bbox.shape), dtype=dtype,
+                          pattern=pattern, voxel_offset=bbox.start,
+                          voxel_size=voxel_size)
+
+    # --- properties --------------------------------------------------------
+    @property
+    def shape(self):
+        return tuple(self.array.shape)
+
+    @property
+    def ndim(self):
+        return self.array.ndim
+
+    @property
+    def dtype(self):
+        if _is_tensor(self.array):
+            return self.array.dtype
+        return self.array.dtype
+
+    @property
+    def is_device(self) -> bool:
+        return _is_tensor(self.array) and self.array.is_cuda
+
+    @property
+    def ndoffset(self) -> tuple:
+        if self.ndim == 4:
+            return (0,) + tuple(self.voxel_offset)
+        return tuple(self.voxel_offset)
+
+    @property
+    def slices(self) -> tuple:
+        return tuple(slice(o, o + s)
+                     for o, s in zip(self.ndoffset, self.shape))
+
+    @property
+    def bbox(self) -> BoundingBox:
+        return BoundingBox.from_delta(self.voxel_offset, self.shape[-3:])
+
+    # --- host/device movement ---------------------------------------------
+    def numpy(self) -> 'Chunk':
+        """Host view/copy of this chunk (D2H when device-backed)."""
+        if _is_tensor(self.array):
+            return Chunk(self.array.detach().cpu().numpy(),
+                         voxel_offset=self.voxel_offset,
+                         voxel_size=self.voxel_size)
+        return self
+
+    def to_device(self, device='cuda') -> 'Chunk':
+        """Move to GPU HBM (H2D once; the operator chain then stays
+        device-resident)."""
+        if self.is_device:
+            return self
+        arr = self.array
+        if isinstance(arr, np.ndarray):
+            t = torch.from_numpy(np.ascontiguousarray(arr))
+        else:
+            t = arr
+        return Chunk(t.to(device, non_blocking=True),
+                     voxel_offset=self.voxel_offset,
+                     voxel_size=self.voxel_size)
+
+    def clone(self) -> 'Chunk':
+        arr = self.array.clone() if _is_tensor(self.array) \
+            else self.array.copy()
+        return Chunk(arr, voxel_offset=self.voxel_offset,
+                     voxel_size=self.voxel_size)
+
+    def astype(self, dtype) -> 'Chunk':
+        if _is_tensor(self.array):
+            tdt = {'float32': torch.float32, 'float16': torch.float16,
+                   'uint8': torch.uint8}[np.dtype(dtype).name]
+            return Chunk(self.array.to(tdt), voxel_offset=self.voxel_offset,
+                         voxel_size=self.voxel_size)
+        return Chunk(self.array.astype(dtype), voxel_offset=self.voxel_offset,
+                     voxel_size=self.voxel_size)
+
+    # --- hot-path host semantics (reference chunk/base.py) -----------------
+    def cutout(self, x) -> 'Chunk':
+        """Region-of-interest in GLOBAL coordinates (chunk/base.py:761-781)."""
+        if isinstance(x, BoundingBox):
+            slices = x.slices
+        else:
+            slices = x
+        if len(slices) == self.ndim - 1:
+            slices = (slice(0, self.shape[0]),) + tuple(slices)
+        internal = tuple(
+            slice(s.start - o, s.stop - o)
+            for s, o in zip(slices, self.ndoffset))
+        arr = self.array[internal]
+        return Chunk(arr, voxel_offset=tuple(s.start for s in slices[-3:]),
+                     voxel_size=self.voxel_size)
+
+    def save(self, patch: 'Chunk'):
+        """Replace a subvolume (chunk/base.py:783-790)."""
+        internal = tuple(slice(s.start - o, s.stop - o)
+                         for s, o in zip(patch.slices, self.ndoffset))
+        self.array[internal] = patch.array
+
+    def blend(self, patch: 'Chunk'):
+        """out[region] += patch, clipped to the intersection
+        (chunk/base.py:792-807)."""
+        internal = tuple(
+            slice(max(s.start - o, 0), min(s.stop - o, h))
+            for s, o, h in zip(patch.slices, self.ndoffset, self.shape))
+        shape = tuple(s.stop - s.start for s in internal)
+        pstart = tuple(i.start - (s.start - o) for s, o, i in
+                       zip(patch.slices, self.ndoffset, internal))
+        pslices = tuple(slice(s, s + h) for s, h in zip(pstart, shape))
+        self.array[internal] += patch.array[pslices]
+
+    def crop_margin(self, margin_size=None, output_bbox=None) -> 'Chunk':
+        """Slice off a 3- or 6-tuple margin (chunk/base.py:691-726).
+        Host path: numpy/tensor view. Device chunks get a contiguous copy via
+        the HIP crop kernel in the inference operator chain."""
+        if margin_size:
+            sz, sy, sx = self.shape[-3:]
+            if len(margin_size) == 3:
+                m = tuple(margin_size) * 2
+            elif len(margin_size) == 6:
+                m = tuple(margin_size)
+            else:
+                raise ValueError('margin_size must have 3 or 6 elements')
+            arr = self.array[..., m[0]:sz - m[3], m[1]:sy - m[4],
+                             m[2]:sx - m[5]]
+            offset = tuple(o + mm for o, mm in
+                           zip(self.voxel_offset, m[:3]))
+            return Chunk(arr, voxel_offset=offset, voxel_size=self.voxel_size)
+        assert output_bbox is not None
+        return self.cutout(output_bbox.slices)
+
+    def mask_using_last_channel(self, threshold: float = 0.3) -> 'Chunk':
+        """(C,z,y,x) -> (C-1,z,y,x) masked where last channel >= threshold
+        (chunk/base.py:685-689)."""
+        mask = self.array[-1] < threshold
+        ret = self.array[:-1]
+        ret = ret * mask
+        return Chunk(ret, voxel_offset=self.voxel_offset,
+                     voxel_size=self.voxel_size)
+
+    def threshold(self, threshold: float) -> 'Chunk':
+        arr = self.array > threshold
+        if arr.ndim == 4:
+            assert arr.shape[0] == 1
+            arr = arr[0]
+        if _is_tensor(arr):
+            arr = arr.to(torch.uint8)
+        else:
+            arr = arr.astype(np.uint8)
+        return Chunk(arr, voxel_offset=self.voxel_offset,
+                     voxel_size=self.voxel_size)
+
+    # --- arithmetic passthrough --------------------------------------------
+    def __imul__(self, other):
+        self.array *= other.array if isinstance(other, Chunk) else other
+        return self
+
+    def __itruediv__(self, other):
+        self.array /= other.array if isinstance(other, Chunk) else other
+        return self
+
+    def __isub__(self, other):
+        self.array -= other.array if isinstance(other, Chunk) else other
+        return self
+
+    def __eq__(self, other):
+        if isinstance(other, Chunk):
+            return self.array == other.array
+        return self.array == other
+
+    def __getitem__(self, idx):
+        return self.array[idx]
+
+    def __setitem__(self, idx, value):
+        self.array[idx] = value
+
+    def __len__(self):
+        return len(self.array)
+
+    def __array__(self):
+        assert isinstance(self.array, np.ndarray)
+        return self.array
+
+    def min(self):
+        return self.array.min()
+
+    def max(self):
+        return self.array.max()
+
+    # --- simple file IO at the pipeline edge (npy/npz) ---------------------
+    def to_npy(self, file_name: str):
+        np.save(file_name, self.numpy().array)
+
+    @classmethod
+    def from_npy(cls, file_name: str, voxel_offset=(0, 0, 0),
+                 voxel_size=None):
+        return cls(np.load(file_name), voxel_offset=voxel_offset,
+                   voxel_size=voxel_size)

@@ -1,0 +1,701 @@
+// chunkflow_amd hot-path kernels for MI355X (gfx950, CDNA4).
+//
+// Every kernel here is HBM-bandwidth-bound streaming work over f32 volumes
+// laid out C-order z-y-x (x fastest): coalesced float4 lines along x,
+// 256-thread workgroups (4 waves of 64), grid-stride over (channel, z, y)
+// "lines" so a 512^3 chunk launches tens of thousands of workgroups (>>256
+// CUs, fills all 8 XCDs). No LDS is needed: there is zero reuse beyond the
+// overlap halo, which the line decomposition already exploits via L2.
+// Compiled with -ffp-contract=off so the blend's multiply-then-add matches
+// the reference's numpy arithmetic (two roundings, not one fused fma).
+//
+// Semantics follow seung-lab/chunkflow v1.1.7 (citations in
+// include/chunkflow_amd.h and per kernel below).
+#include <hip/hip_runtime.h>
+
+#include <cmath>
+#include <cstdio>
+#include <cstring>
+#include <string>
+#include <vector>
+
+#include "../../include/chunkflow_amd.h"
+
+#define CFX_VERSION 1
+
+// ---------------------------------------------------------------------------
+// error handling
+// ---------------------------------------------------------------------------
+static thread_local std::string g_err;
+
+extern "C" const char* cfx_last_error(void) { return g_err.c_str(); }
+extern "C" int cfx_version(void) { return CFX_VERSION; }
+
+#define CFX_CHECK(expr)                                                      \
+    do {                                                                     \
+        hipError_t _e = (expr);                                              \
+        if (_e != hipSuccess) {                                              \
+            g_err = std::string(#expr) + ": " + hipGetErrorString(_e);       \
+            return -1;                                                       \
+        }                                                                    \
+    } while (0)
+
+// ---------------------------------------------------------------------------
+// context
+// ---------------------------------------------------------------------------
+struct ProfEntry {
+    hipEvent_t e0, e1;
+    int kid;
+    double bytes;
+};
+
+struct cfx_ctx {
+    int device = 0;
+    hipStream_t stream = nullptr;  // legacy default stream unless adopted
+    bool profile = false;
+    std::vector<ProfEntry> pending;
+    unsigned long long prof_count[CFX_K_COUNT] = {};
+    double prof_ms[CFX_K_COUNT] = {};
+    double prof_bytes[CFX_K_COUNT] = {};
+    unsigned int* dev_max = nullptr;  // scratch for cfx_max
+};
+
+extern "C" cfx_ctx* cfx_init(int device) {
+    hipError_t e = hipSetDevice(device);
+    if (e != hipSuccess) {
+        g_err = std::string("hipSetDevice: ") + hipGetErrorString(e);
+        return nullptr;
+    }
+    cfx_ctx* ctx = new cfx_ctx();
+    ctx->device = device;
+    if (hipMalloc(&ctx->dev_max, sizeof(unsigned int)) != hipSuccess) {
+        g_err = "hipMalloc scratch failed";
+        delete ctx;
+        return nullptr;
+    }
+    return ctx;
+}
+
+extern "C" void cfx_destroy(cfx_ctx* ctx) {
+    if (!ctx) return;
+    for (auto& p : ctx->pending) {
+        hipEventDestroy(p.e0);
+        hipEventDestroy(p.e1);
+    }
+    if (ctx->dev_max) hipFree(ctx->dev_max);
+    delete ctx;
+}
+
+extern "C" int cfx_set_stream(cfx_ctx* ctx, void* hip_stream) {
+    ctx->stream = reinterpret_cast<hipStream_t>(hip_stream);
+    return 0;
+}
+
+extern "C" int cfx_sync(cfx_ctx* ctx) {
+    CFX_CHECK(hipStreamSynchronize(ctx->stream));
+    return 0;
+}
+
+// profiling helpers ---------------------------------------------------------
+static int prof_begin(cfx_ctx* ctx, hipEvent_t* e0) {
+    if (!ctx->profile) return 0;
+    CFX_CHECK(hipEventCreate(e0));
+    CFX_CHECK(hipEventRecord(*e0, ctx->stream));
+    return 0;
+}
+
+static int prof_end(cfx_ctx* ctx, hipEvent_t e0, int kid, double bytes) {
+    if (!ctx->profile) return 0;
+    hipEvent_t e1;
+    CFX_CHECK(hipEventCreate(&e1));
+    CFX_CHECK(hipEventRecord(e1, ctx->stream));
+    ctx->pending.push_back({e0, e1, kid, bytes});
+    return 0;
+}
+
+extern "C" int cfx_profile_enable(cfx_ctx* ctx, int enable) {
+    ctx->profile = enable != 0;
+    return 0;
+}
+
+extern "C" int cfx_profile_reset(cfx_ctx* ctx) {
+    CFX_CHECK(hipStreamSynchronize(ctx->stream));
+    for (auto& p : ctx->pending) {
+        hipEventDestroy(p.e0);
+        hipEventDestroy(p.e1);
+    }
+    ctx->pending.clear();
+    memset(ctx->prof_count, 0, sizeof(ctx->prof_count));
+    memset(ctx->prof_ms, 0, sizeof(ctx->prof_ms));
+    memset(ctx->prof_bytes, 0, sizeof(ctx->prof_bytes));
+    return 0;
+}
+
+extern "C" int cfx_profile_get(cfx_ctx* ctx, int kernel_id,
+                               unsigned long long* count, double* total_ms,
+                               double* bytes) {
+    if (kernel_id < 0 || kernel_id >= CFX_K_COUNT) {
+        g_err = "bad kernel id";
+        return -1;
+    }
+    CFX_CHECK(hipStreamSynchronize(ctx->stream));
+    for (auto& p : ctx->pending) {
+        float ms = 0.f;
+        CFX_CHECK(hipEventElapsedTime(&ms, p.e0, p.e1));
+        ctx->prof_count[p.kid] += 1;
+        ctx->prof_ms[p.kid] += ms;
+        ctx->prof_bytes[p.kid] += p.bytes;
+        hipEventDestroy(p.e0);
+        hipEventDestroy(p.e1);
+    }
+    ctx->pending.clear();
+    *count = ctx->prof_count[kernel_id];
+    *total_ms = ctx->prof_ms[kernel_id];
+    *bytes = ctx->prof_bytes[kernel_id];
+    return 0;
+}
+
+// ---------------------------------------------------------------------------
+// elementwise u8 -> f32 kernels (normalize-intensity, int->unit cast)
+// ---------------------------------------------------------------------------
+// out = (float)in * a + b; vectorized uchar4 -> float4. a/b are chosen by the
+// host so the arithmetic matches the reference exactly:
+//   normalize-intensity: x/127.5 - 1   (flow.py:1664-1666)
+//   unit cast:           x/255        (inferencer.py:395-399)
+// f32 division by a constant and multiplication by its reciprocal differ in
+// the last ulp, so we divide, like numpy does.
+__global__ void k_u8_to_f32(const unsigned char* __restrict__ in,
+                            float* __restrict__ out, long long n4,
+                            float divisor, float bias) {
+    long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+    long long stride = gridDim.x * (long long)blockDim.x;
+    const uchar4* in4 = reinterpret_cast<const uchar4*>(in);
+    float4* out4 = reinterpret_cast<float4*>(out);
+    for (; i < n4; i += stride) {
+        uchar4 v = in4[i];
+        float4 r;
+        r.x = (float)v.x / divisor + bias;
+        r.y = (float)v.y / divisor + bias;
+        r.z = (float)v.z / divisor + bias;
+        r.w = (float)v.w / divisor + bias;
+        out4[i] = r;
+    }
+}
+
+__global__ void k_u8_to_f32_tail(const unsigned char* __restrict__ in,
+                                 float* __restrict__ out, long long start,
+                                 long long n, float divisor, float bias) {
+    long long i = start + blockIdx.x * (long long)blockDim.x + threadIdx.x;
+    if (i < n) out[i] = (float)in[i] / divisor + bias;
+}
+
+static int u8_to_f32(cfx_ctx* ctx, const unsigned char* in, float* out,
+                     long long n, float divisor, float bias, int kid) {
+    hipEvent_t e0;
+    if (prof_begin(ctx, &e0)) return -1;
+    long long n4 = n / 4;
+    if (n4 > 0) {
+        int threads = 256;
+        long long want = (n4 + threads - 1) / threads;
+        int blocks = (int)std::min<long long>(want, 8192);
+        hipLaunchKernelGGL(k_u8_to_f32, dim3(blocks), dim3(threads), 0,
+                           ctx->stream, in, out, n4, divisor, bias);
+    }
+    if (n % 4) {
+        hipLaunchKernelGGL(k_u8_to_f32_tail, dim3(1), dim3(256), 0,
+                           ctx->stream, in, out, n4 * 4, n, divisor, bias);
+    }
+    CFX_CHECK(hipGetLastError());
+    if (prof_end(ctx, e0, kid, (double)n * 5.0)) return -1;  // 1B read + 4B write
+    return 0;
+}
+
+extern "C" int cfx_normalize_intensity(cfx_ctx* ctx, const unsigned char* in,
+                                       float* out, long long n) {
+    return u8_to_f32(ctx, in, out, n, 127.5f, -1.0f, CFX_K_NORMALIZE);
+}
+
+extern "C" int cfx_cast_u8_f32_div(cfx_ctx* ctx, const unsigned char* in,
+                                   float* out, long long n, float divisor) {
+    return u8_to_f32(ctx, in, out, n, divisor, 0.0f, CFX_K_CAST);
+}
+
+// ---------------------------------------------------------------------------
+// patch extraction: gather windows of the f32 chunk into the batch buffer
+// ---------------------------------------------------------------------------
+// One launch per batch. Block b of the grid walks "lines" (patch, channel,
+// z, y); threads stride x. Patch starts ride in kernarg (<=64 per launch).
+struct ExtractArgs {
+    int n;                 // number of patches this launch
+    int starts[64 * 3];    // chunk-local (z, y, x) starts
+};
+
+template <bool VEC>
+__global__ void k_extract(const float* __restrict__ chunk, int C, int D,
+                          int H, int W, ExtractArgs args, int pz, int py,
+                          int px, float* __restrict__ out) {
+    long long n_lines = (long long)args.n * C * pz * py;
+    for (long long line = blockIdx.x; line < n_lines; line += gridDim.x) {
+        int y = (int)(line % py);
+        long long t = line / py;
+        int z = (int)(t % pz);
+        t /= pz;
+        int c = (int)(t % C);
+        int p = (int)(t / C);
+        int z0 = args.starts[p * 3 + 0];
+        int y0 = args.starts[p * 3 + 1];
+        int x0 = args.starts[p * 3 + 2];
+        const float* src =
+            chunk + (((long long)c * D + z0 + z) * H + y0 + y) * W + x0;
+        float* dst =
+            out + ((((long long)p * C + c) * pz + z) * py + y) * px;
+        if (VEC) {
+            const float4* s4 = reinterpret_cast<const float4*>(src);
+            float4* d4 = reinterpret_cast<float4*>(dst);
+            for (int x = threadIdx.x; x < px / 4; x += blockDim.x)
+                d4[x] = s4[x];
+        } else {
+            for (int x = threadIdx.x; x < px; x += blockDim.x)
+                dst[x] = src[x];
+        }
+    }
+}
+
+extern "C" int cfx_extract_patches(cfx_ctx* ctx, const float* chunk,
+                                   int channels, const int chunk_dims[3],
+                                   const int* starts_zyx, int n_patches,
+                                   const int patch_size[3], float* out) {
+    int D = chunk_dims[0], H = chunk_dims[1], W = chunk_dims[2];
+    int pz = patch_size[0], py = patch_size[1], px = patch_size[2];
+    long long pvox = (long long)pz * py * px;
+    for (int base = 0; base < n_patches; base += 64) {
+        int n = std::min(64, n_patches - base);
+        ExtractArgs args;
+        args.n = n;
+        bool vec = (W % 4 == 0) && (px % 4 == 0);
+        for (int i = 0; i < n; ++i) {
+            args.starts[i * 3 + 0] = starts_zyx[(base + i) * 3 + 0];
+            args.starts[i * 3 + 1] = starts_zyx[(base + i) * 3 + 1];
+            args.starts[i * 3 + 2] = starts_zyx[(base + i) * 3 + 2];
+            vec = vec && (args.starts[i * 3 + 2] % 4 == 0);
+        }
+        hipEvent_t e0;
+        if (prof_begin(ctx, &e0)) return -1;
+        long long n_lines = (long long)n * channels * pz * py;
+        int blocks = (int)std::min<long long>(n_lines, 8192);
+        float* dst = out + (long long)base * channels * pvox;
+        if (vec)
+            hipLaunchKernelGGL(k_extract<true>, dim3(blocks), dim3(256), 0,
+                               ctx->stream, chunk, channels, D, H, W, args,
+                               pz, py, px, dst);
+        else
+            hipLaunchKernelGGL(k_extract<false>, dim3(blocks), dim3(256), 0,
+                               ctx->stream, chunk, channels, D, H, W, args,
+                               pz, py, px, dst);
+        CFX_CHECK(hipGetLastError());
+        double bytes = (double)n * channels * pvox * 8.0;  // read + write f32
+        if (prof_end(ctx, e0, CFX_K_EXTRACT, bytes)) return -1;
+    }
+    return 0;
+}
+
+// ---------------------------------------------------------------------------
+// blend-accumulate: out[c, region] += patch[c, region'] * mask[region']
+// ---------------------------------------------------------------------------
+// The highest-traffic kernel of the whole path (~13.6 GB per
+// 512^3 chunk). One launch per patch (keeps the accumulate deterministic and
+// atomics-free: within a launch each output voxel is written exactly once;
+// patches overlap only ACROSS launches, which the stream serializes in the
+// reference's own blend order). Lines along x, float4 when aligned.
+template <bool VEC, bool MASKED>
+__global__ void k_blend(float* __restrict__ out, int OD, int OH, int OW,
+                        const float* __restrict__ patch, int PD, int PH,
+                        int PW, const float* __restrict__ mask, int C,
+                        int dz0, int dy0, int dx0,   // region start in out
+                        int pz0, int py0, int px0,   // region start in patch
+                        int rz, int ry, int rx) {
+    long long n_lines = (long long)C * rz * ry;
+    for (long long line = blockIdx.x; line < n_lines; line += gridDim.x) {
+        int y = (int)(line % ry);
+        long long t = line / ry;
+        int z = (int)(t % rz);
+        int c = (int)(t / rz);
+        float* o =
+            out + (((long long)c * OD + dz0 + z) * OH + dy0 + y) * OW + dx0;
+        const float* p =
+            patch + (((long long)c * PD + pz0 + z) * PH + py0 + y) * PW + px0;
+        const float* m = MASKED
+            ? mask + (((long long)(pz0 + z)) * PH + py0 + y) * PW + px0
+            : nullptr;
+        if (VEC) {
+            float4* o4 = reinterpret_cast<float4*>(o);
+            const float4* p4 = reinterpret_cast<const float4*>(p);
+            const float4* m4 = reinterpret_cast<const float4*>(m);
+            for (int x = threadIdx.x; x < rx / 4; x += blockDim.x) {
+                float4 ov = o4[x];
+                float4 pv = p4[x];
+                if (MASKED) {
+                    float4 mv = m4[x];
+                    ov.x += pv.x * mv.x;
+                    ov.y += pv.y * mv.y;
+                    ov.z += pv.z * mv.z;
+                    ov.w += pv.w * mv.w;
+                } else {
+                    ov.x += pv.x;
+                    ov.y += pv.y;
+                    ov.z += pv.z;
+                    ov.w += pv.w;
+                }
+                o4[x] = ov;
+            }
+        } else {
+            for (int x = threadIdx.x; x < rx; x += blockDim.x)
+                o[x] += MASKED ? p[x] * m[x] : p[x];
+        }
+    }
+}
+
+// clip the patch span against the output bounds (Chunk.blend semantics,
+// chunk/base.py:796-807); returns false when the intersection is empty
+static bool clip_region(const int out_dims[3], const int patch_dims[3],
+                        const int offset[3], int* d0, int* p0, int* r) {
+    for (int a = 0; a < 3; ++a) {
+        int lo = offset[a] > 0 ? offset[a] : 0;
+        int hi = offset[a] + patch_dims[a] < out_dims[a]
+                     ? offset[a] + patch_dims[a]
+                     : out_dims[a];
+        if (hi <= lo) return false;
+        d0[a] = lo;
+        p0[a] = lo - offset[a];
+        r[a] = hi - lo;
+    }
+    return true;
+}
+
+static int blend_one(cfx_ctx* ctx, float* out, int C, const int out_dims[3],
+                     const float* patch, const int patch_dims[3],
+                     const int offset[3], const float* mask) {
+    int d0[3], p0[3], r[3];
+    if (!clip_region(out_dims, patch_dims, offset, d0, p0, r)) return 0;
+    bool vec = (out_dims[2] % 4 == 0) && (patch_dims[2] % 4 == 0) &&
+               (d0[2] % 4 == 0) && (p0[2] % 4 == 0) && (r[2] % 4 == 0);
+    long long n_lines = (long long)C * r[0] * r[1];
+    int blocks = (int)std::min<long long>(n_lines, 8192);
+    hipEvent_t e0;
+    if (prof_begin(ctx, &e0)) return -1;
+#define CFX_LAUNCH_BLEND(V, M)                                              \
+    hipLaunchKernelGGL((k_blend<V, M>), dim3(blocks), dim3(256), 0,         \
+                       ctx->stream, out, out_dims[0], out_dims[1],          \
+                       out_dims[2], patch, patch_dims[0], patch_dims[1],    \
+                       patch_dims[2], mask, C, d0[0], d0[1], d0[2], p0[0],  \
+                       p0[1], p0[2], r[0], r[1], r[2])
+    if (vec && mask) CFX_LAUNCH_BLEND(true, true);
+    else if (vec) CFX_LAUNCH_BLEND(true, false);
+    else if (mask) CFX_LAUNCH_BLEND(false, true);
+    else CFX_LAUNCH_BLEND(false, false);
+#undef CFX_LAUNCH_BLEND
+    CFX_CHECK(hipGetLastError());
+    double rv = (double)r[0] * r[1] * r[2];
+    // algorithmic bytes: out read+write + patch read (per channel) + mask read
+    double bytes = rv * C * 12.0 + (mask ? rv * 4.0 : 0.0);
+    if (prof_end(ctx, e0, CFX_K_BLEND, bytes)) return -1;
+    return 0;
+}
+
+extern "C" int cfx_blend_accumulate(cfx_ctx* ctx, float* out, int channels,
+                                    const int out_dims[3], const float* patch,
+                                    const int patch_dims[3],
+                                    const int offset_zyx[3],
+                                    const float* mask) {
+    return blend_one(ctx, out, channels, out_dims, patch, patch_dims,
+                     offset_zyx, mask);
+}
+
+// ---------------------------------------------------------------------------
+// chunk-mask build: zero, blend the patch mask at every offset, reciprocal
+// ---------------------------------------------------------------------------
+__global__ void k_reciprocal(float* __restrict__ buf, long long n) {
+    long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+    long long stride = gridDim.x * (long long)blockDim.x;
+    for (; i < n; i += stride) buf[i] = 1.0f / buf[i];
+}
+
+extern "C" int cfx_reciprocal(cfx_ctx* ctx, float* buf, long long n) {
+    hipEvent_t e0;
+    if (prof_begin(ctx, &e0)) return -1;
+    int threads = 256;
+    long long want = (n + threads - 1) / threads;
+    int blocks = (int)std::min<long long>(want, 8192);
+    hipLaunchKernelGGL(k_reciprocal, dim3(blocks), dim3(threads), 0,
+                       ctx->stream, buf, n);
+    CFX_CHECK(hipGetLastError());
+    if (prof_end(ctx, e0, CFX_K_RECIPROCAL, (double)n * 8.0)) return -1;
+    return 0;
+}
+
+extern "C" int cfx_build_chunk_mask(cfx_ctx* ctx, float* mask_out,
+                                    const int out_dims[3],
+                                    const float* patch_mask,
+                                    const int patch_dims[3],
+                                    const int* offsets_zyx, int n) {
+    long long nvox =
+        (long long)out_dims[0] * out_dims[1] * out_dims[2];
+    CFX_CHECK(hipMemsetAsync(mask_out, 0, nvox * sizeof(float), ctx->stream));
+    for (int i = 0; i < n; ++i) {
+        if (blend_one(ctx, mask_out, 1, out_dims, patch_mask, patch_dims,
+                      offsets_zyx + i * 3, nullptr))
+            return -1;
+    }
+    return cfx_reciprocal(ctx, mask_out, nvox);
+}
+
+// ---------------------------------------------------------------------------
+// mask-normalize multiply: out[c, i] *= mask[i]
+// ---------------------------------------------------------------------------
+template <bool VEC>
+__global__ void k_maskmul(float* __restrict__ out,
+                          const float* __restrict__ mask, long long n) {
+    int c = blockIdx.y;
+    long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+    long long stride = gridDim.x * (long long)blockDim.x;
+    if (VEC) {
+        float4* o4 = reinterpret_cast<float4*>(out + c * n * 4);
+        const float4* m4 = reinterpret_cast<const float4*>(mask);
+        for (; i < n; i += stride) {
+            float4 o = o4[i];
+            float4 m = m4[i];
+            o.x *= m.x;
+            o.y *= m.y;
+            o.z *= m.z;
+            o.w *= m.w;
+            o4[i] = o;
+        }
+    } else {
+        float* o = out + c * n;
+        for (; i < n; i += stride) o[i] *= mask[i];
+    }
+}
+
+extern "C" int cfx_multiply_mask(cfx_ctx* ctx, float* out, const float* mask,
+                                 int channels, long long n_voxels) {
+    hipEvent_t e0;
+    if (prof_begin(ctx, &e0)) return -1;
+    int threads = 256;
+    bool vec = (n_voxels % 4 == 0);
+    long long n = vec ? n_voxels / 4 : n_voxels;
+    long long want = (n + threads - 1) / threads;
+    int blocks = (int)std::min<long long>(want, 4096);
+    dim3 grid(blocks, channels);
+    if (vec)
+        hipLaunchKernelGGL(k_maskmul<true>, grid, dim3(threads), 0,
+                           ctx->stream, out, mask, n);
+    else
+        hipLaunchKernelGGL(k_maskmul<false>, grid, dim3(threads), 0,
+                           ctx->stream, out, mask, n);
+    CFX_CHECK(hipGetLastError());
+    double bytes = (double)n_voxels * (channels * 8.0 + 4.0);
+    if (prof_end(ctx, e0, CFX_K_MASKMUL, bytes)) return -1;
+    return 0;
+}
+
+// ---------------------------------------------------------------------------
+// max reduce (sanity assert: all(out < 1.0001), inferencer.py:463-466)
+// ---------------------------------------------------------------------------
+// order-preserving bit transform so unsigned atomicMax works for any float
+__host__ __device__ inline unsigned int f32_ord(float f) {
+    unsigned int u;
+    __builtin_memcpy(&u, &f, sizeof(u));
+    return (u & 0x80000000u) ? ~u : (u | 0x80000000u);
+}
+
+__global__ void k_max(const float* __restrict__ buf, long long n,
+                      unsigned int* __restrict__ result) {
+    long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+    long long stride = gridDim.x * (long long)blockDim.x;
+    float m = -INFINITY;
+    for (; i < n; i += stride) m = fmaxf(m, buf[i]);
+    // wave reduce (64 lanes), then one atomic per wave
+    for (int off = 32; off > 0; off >>= 1)
+        m = fmaxf(m, __shfl_down(m, off, 64));
+    if ((threadIdx.x & 63) == 0) atomicMax(result, f32_ord(m));
+}
+
+extern "C" int cfx_max(cfx_ctx* ctx, const float* buf, long long n,
+                       float* host_max) {
+    unsigned int init = f32_ord(-INFINITY);
+    CFX_CHECK(hipMemcpyAsync(ctx->dev_max, &init, sizeof(init),
+                             hipMemcpyHostToDevice, ctx->stream));
+    hipEvent_t e0;
+    if (prof_begin(ctx, &e0)) return -1;
+    int threads = 256;
+    long long want = (n + threads - 1) / threads;
+    int blocks = (int)std::min<long long>(want, 4096);
+    hipLaunchKernelGGL(k_max, dim3(blocks), dim3(threads), 0, ctx->stream,
+                       buf, n, ctx->dev_max);
+    CFX_CHECK(hipGetLastError());
+    if (prof_end(ctx, e0, CFX_K_MAX, (double)n * 4.0)) return -1;
+    unsigned int out;
+    CFX_CHECK(hipMemcpyAsync(&out, ctx->dev_max, sizeof(out),
+                             hipMemcpyDeviceToHost, ctx->stream));
+    CFX_CHECK(hipStreamSynchronize(ctx->stream));
+    unsigned int u = (out & 0x80000000u) ? (out & 0x7fffffffu) : ~out;
+    *host_max = __builtin_bit_cast(float, u);
+    return 0;
+}
+
+// ---------------------------------------------------------------------------
+// crop-margin: contiguous copy dropping margins (chunk/base.py:691-726)
+// ---------------------------------------------------------------------------
+template <bool VEC>
+__global__ void k_crop(const float* __restrict__ in, int D, int H, int W,
+                       float* __restrict__ out, int C, int oD, int oH, int oW,
+                       int m0, int m1, int m2) {
+    long long n_lines = (long long)C * oD * oH;
+    for (long long line = blockIdx.x; line < n_lines; line += gridDim.x) {
+        int y = (int)(line % oH);
+        long long t = line / oH;
+        int z = (int)(t % oD);
+        int c = (int)(t / oD);
+        const float* src =
+            in + (((long long)c * D + m0 + z) * H + m1 + y) * W + m2;
+        float* dst = out + (((long long)c * oD + z) * oH + y) * oW;
+        if (VEC) {
+            const float4* s4 = reinterpret_cast<const float4*>(src);
+            float4* d4 = reinterpret_cast<float4*>(dst);
+            for (int x = threadIdx.x; x < oW / 4; x += blockDim.x)
+                d4[x] = s4[x];
+        } else {
+            for (int x = threadIdx.x; x < oW; x += blockDim.x)
+                dst[x] = src[x];
+        }
+    }
+}
+
+extern "C" int cfx_crop_margin(cfx_ctx* ctx, const float* in, float* out,
+                               int channels, const int in_dims[3],
+                               const int margins[6]) {
+    int D = in_dims[0], H = in_dims[1], W = in_dims[2];
+    int oD = D - margins[0] - margins[3];
+    int oH = H - margins[1] - margins[4];
+    int oW = W - margins[2] - margins[5];
+    if (oD <= 0 || oH <= 0 || oW <= 0) {
+        g_err = "crop_margin: empty output";
+        return -1;
+    }
+    bool vec = (W % 4 == 0) && (oW % 4 == 0) && (margins[2] % 4 == 0);
+    long long n_lines = (long long)channels * oD * oH;
+    int blocks = (int)std::min<long long>(n_lines, 8192);
+    hipEvent_t e0;
+    if (prof_begin(ctx, &e0)) return -1;
+    if (vec)
+        hipLaunchKernelGGL(k_crop<true>, dim3(blocks), dim3(256), 0,
+                           ctx->stream, in, D, H, W, out, channels, oD, oH,
+                           oW, margins[0], margins[1], margins[2]);
+    else
+        hipLaunchKernelGGL(k_crop<false>, dim3(blocks), dim3(256), 0,
+                           ctx->stream, in, D, H, W, out, channels, oD, oH,
+                           oW, margins[0], margins[1], margins[2]);
+    CFX_CHECK(hipGetLastError());
+    double bytes = (double)channels * oD * oH * oW * 8.0;
+    if (prof_end(ctx, e0, CFX_K_CROP, bytes)) return -1;
+    return 0;
+}
+
+// ---------------------------------------------------------------------------
+// myelin mask: out[c] = in[c] * (in[C-1] < threshold)  (chunk/base.py:685-689)
+// ---------------------------------------------------------------------------
+__global__ void k_myelin(const float* __restrict__ in,
+                         float* __restrict__ out, long long n, int cout,
+                         float threshold) {
+    long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+    long long stride = gridDim.x * (long long)blockDim.x;
+    const float* last = in + (long long)cout * n;
+    for (; i < n; i += stride) {
+        float keep = last[i] < threshold ? 1.0f : 0.0f;
+        for (int c = 0; c < cout; ++c)
+            out[c * n + i] = in[c * n + i] * keep;
+    }
+}
+
+extern "C" int cfx_mask_using_last_channel(cfx_ctx* ctx, const float* in,
+                                           float* out, int channels,
+                                           const int dims[3],
+                                           float threshold) {
+    long long n = (long long)dims[0] * dims[1] * dims[2];
+    int cout = channels - 1;
+    hipEvent_t e0;
+    if (prof_begin(ctx, &e0)) return -1;
+    int threads = 256;
+    long long want = (n + threads - 1) / threads;
+    int blocks = (int)std::min<long long>(want, 8192);
+    hipLaunchKernelGGL(k_myelin, dim3(blocks), dim3(threads), 0, ctx->stream,
+                       in, out, n, cout, threshold);
+    CFX_CHECK(hipGetLastError());
+    double bytes = (double)n * (channels + cout) * 4.0;
+    if (prof_end(ctx, e0, CFX_K_MYELIN, bytes)) return -1;
+    return 0;
+}
+
+// ---------------------------------------------------------------------------
+// host-side patch mask (f64 pipeline; patch/patch_mask.py:15-68)
+// ---------------------------------------------------------------------------
+// The product's Python host path computes this with numpy (bit-identical to
+// the reference); this C implementation completes the C-ABI for non-Python
+// hosts and is tested against the same golden CRCs to <=1e-12 relative.
+extern "C" int cfx_make_patch_mask(const int patch_size[3],
+                                   const int overlap[3], float* out) {
+    const int pz = patch_size[0], py = patch_size[1], px = patch_size[2];
+    const long long n = (long long)pz * py * px;
+    std::vector<double> bump((size_t)n);
+    // bump map on centered grids
+    std::vector<double> gx(px), gy(py), gz(pz);
+    for (int i = 0; i < px; ++i)
+        gx[i] = (i + 1.0) / (px + 1.0) * 2.0 - 1.0;
+    for (int i = 0; i < py; ++i)
+        gy[i] = (i + 1.0) / (py + 1.0) * 2.0 - 1.0;
+    for (int i = 0; i < pz; ++i)
+        gz[i] = (i + 1.0) / (pz + 1.0) * 2.0 - 1.0;
+    double bmin = 1e300, bmax = -1e300;
+    for (int z = 0; z < pz; ++z)
+        for (int y = 0; y < py; ++y)
+            for (int x = 0; x < px; ++x) {
+                double v = std::exp(-1.0 / (1.0 - gx[x] * gx[x]) +
+                                    -1.0 / (1.0 - gy[y] * gy[y]) +
+                                    -1.0 / (1.0 - gz[z] * gz[z]));
+                bump[((long long)z * py + y) * px + x] = v;
+                if (v < bmin) bmin = v;
+                if (v > bmax) bmax = v;
+            }
+    // np.interp remap of (bmin, bmax) -> (1, 1e6)
+    const double slope = (1e6 - 1.0) / (bmax - bmin);
+    for (long long i = 0; i < n; ++i) {
+        double v = bump[i];
+        if (v <= bmin) v = 1.0;
+        else if (v >= bmax) v = 1e6;
+        else v = slope * (v - bmin) + 1.0;
+        bump[i] = v;
+    }
+    // 3x3x3 shifted self-accumulation, normalize by the center crop
+    const int sz = pz - overlap[0], sy = py - overlap[1], sx = px - overlap[2];
+    const int BD = pz + 2 * sz, BH = py + 2 * sy, BW = px + 2 * sx;
+    std::vector<double> base((size_t)BD * BH * BW, 0.0);
+    for (int nz = 0; nz < 3; ++nz)
+        for (int ny = 0; ny < 3; ++ny)
+            for (int nx = 0; nx < 3; ++nx)
+                for (int z = 0; z < pz; ++z)
+                    for (int y = 0; y < py; ++y) {
+                        double* b = &base[((long long)(nz * sz + z) * BH +
+                                           ny * sy + y) * BW + nx * sx];
+                        const double* s = &bump[((long long)z * py + y) * px];
+                        for (int x = 0; x < px; ++x) b[x] += s[x];
+                    }
+    for (int z = 0; z < pz; ++z)
+        for (int y = 0; y < py; ++y)
+            for (int x = 0; x < px; ++x) {
+                long long i = ((long long)z * py + y) * px + x;
+                bump[i] /= base[((long long)(sz + z) * BH + sy + y) * BW +
+                                sx + x];
+            }
+    for (long long i = 0; i < n; ++i) out[i] = (float)bump[i];
+    return 0;
+}

@@ -1,0 +1,417 @@
+"""The `chunkflow` CLI: chained operators over a task stream.
+
+Keeps the reference's command and flag surface verbatim for the hot-path
+operators (reference chunkflow/flow/flow.py):
+  generate-tasks        :73-183   (local grid decomposition; cloud/SQS
+                                   distribution is replaced by in-node
+                                   multi-GPU dispatch — see dispatch.py)
+  create-chunk          :652-678
+  normalize-intensity   :1650-1669
+  inference             :1852-1933
+  crop-margin           :2053-2084
+  plugin                :1751-1800
+  connected-components  :1803-1829 (host path; GPU kernel is a 'next' row)
+  log-summary           :1633-1647 (voxels/sec report shape)
+plus npy load/save at the pipeline edge.
+"""
+import json
+import os
+from time import time
+
+import click
+import numpy as np
+
+from .cartesian import BoundingBox, BoundingBoxes, Cartesian
+from .chunk import Chunk
+from .plugin import Plugin
+from .runtime import (CartesianParam, DEFAULT_CHUNK_NAME, default_none,
+                      generator, get_initial_task, main, operator, state)
+
+
+@main.command('generate-tasks')
+@click.option('--roi-start', '-s', type=click.INT, default=None, nargs=3,
+              callback=default_none, help='(z y x), start of the chunks')
+@click.option('--roi-stop', '-r', type=click.INT, nargs=3, default=None,
+              callback=default_none, help='stop coordinate of region of interest')
+@click.option('--roi-size', '-z', type=click.INT, nargs=3, default=None,
+              callback=default_none, help='size of region of interest')
+@click.option('--chunk-size', '-c', type=click.INT, default=None, nargs=3,
+              help='(z y x), size/shape of chunks')
+@click.option('--bounding-box', '-b', type=str, default=None,
+              help='the string representation of a bounding box')
+@click.option('--grid-size', '-g', type=click.INT, default=None, nargs=3,
+              callback=default_none, help='(z y x), grid size of output blocks')
+@click.option('--file-path', '-f', default=None, type=str,
+              help='output tasks as an numpy array formated as npy.')
+@click.option('--respect-chunk-size/--respect-stop', default=True,
+              help='for the last bounding box, make the chunk size '
+                   'consistent or cut off at the stopping boundary.')
+@click.option('--task-index-start', '-i', type=click.INT, default=0,
+              help='starting index of task list.')
+@click.option('--task-index-stop', '-p', type=click.INT, default=None,
+              help='stop index of task list.')
+@generator
+def generate_tasks(roi_start, roi_stop, roi_size, chunk_size, bounding_box,
+                   grid_size, file_path, respect_chunk_size,
+                   task_index_start, task_index_stop):
+    """Generate a batch of tasks."""
+    if bounding_box is not None:
+        bboxes = [BoundingBox.from_string(bounding_box)]
+        if chunk_size is None:
+            chunk_size = tuple(bboxes[0].shape)
+    else:
+        bboxes = BoundingBoxes.from_manual_setup(
+            chunk_size, roi_start=roi_start, roi_stop=roi_stop,
+            roi_size=roi_size, grid_size=grid_size,
+            respect_chunk_size=respect_chunk_size)
+    if task_index_start:
+        if task_index_stop is None:
+            task_index_stop = len(bboxes)
+        bboxes = bboxes[task_index_start:task_index_stop]
+    if file_path:
+        arr = np.array([list(b.start) + list(b.stop) for b in bboxes],
+                       dtype=np.int64)
+        np.save(file_path, arr)
+    bbox_num = len(bboxes)
+    print(f'total number of tasks: {bbox_num}')
+    for bbox_index, bbox in enumerate(bboxes):
+        task = get_initial_task()
+        task['bbox'] = bbox
+        task['bbox_index'] = bbox_index
+        task['bbox_num'] = bbox_num
+        task['log']['bbox'] = bbox.string
+        yield task
+
+
+@main.command('create-chunk')
+@click.option('--size', '-s', type=click.INT, nargs=3,
+              default=(64, 64, 64), help='the size of created chunk')
+@click.option('--dtype', '-d',
+              type=click.Choice(['uint8', 'uint32', 'uint16', 'uint64',
+                                 'float32', 'float64']),
+              default='uint8', help='the data type of chunk')
+@click.option('--pattern', '-p', type=click.Choice(['sin', 'zero', 'random']),
+              default='sin', help='ways to generate array.')
+@click.option('--voxel-offset', '-t', type=click.INT, nargs=3,
+              default=(0, 0, 0), help='offset in voxel number.')
+@click.option('--voxel-size', '-e', type=click.INT, nargs=3,
+              default=(1, 1, 1), help='voxel size in nm')
+@click.option('--output-chunk-name', '-o', type=str, default='chunk',
+              help='name of created chunk')
+@operator
+def create_chunk(tasks, size, dtype, pattern, voxel_offset, voxel_size,
+                 output_chunk_name):
+    """Create a fake chunk for easy test."""
+    print(f'creating chunk: {output_chunk_name}')
+    for task in tasks:
+        if task is not None:
+            if 'bbox' in task:
+                task[output_chunk_name] = Chunk.from_bbox(
+                    task['bbox'], dtype=np.dtype(dtype), pattern=pattern,
+                    voxel_size=voxel_size)
+            else:
+                task[output_chunk_name] = Chunk.create(
+                    size=size, dtype=np.dtype(dtype), pattern=pattern,
+                    voxel_offset=voxel_offset, voxel_size=voxel_size)
+        yield task
+
+
+@main.command('normalize-intensity')
+@click.option('--name', type=str, default='normalize-intensity',
+              help='name of operator')
+@click.option('--input-chunk-name', '-i', type=str,
+              default=DEFAULT_CHUNK_NAME, help='input chunk name')
+@click.option('--output-chunk-name', '-o', type=str,
+              default=DEFAULT_CHUNK_NAME, help='output chunk name')
+@operator
+def normalize_intensity(tasks, name, input_chunk_name, output_chunk_name):
+    """transform gray image to float (-1:1). x=(x-127.5) - 1.0"""
+    import torch
+    ops = None
+    for task in tasks:
+        if task is not None:
+            start = time()
+            chunk = task[input_chunk_name]
+            if torch.cuda.is_available():
+                # device path: upload once, HIP kernel (flow.py:1664-1666)
+                from .ops import HipOps
+                if ops is None:
+                    ops = HipOps(0)
+                chunk = chunk.to_device()
+                assert chunk.array.dtype == torch.uint8
+                out = ops.normalize_intensity(chunk.array)
+            else:
+                arr = chunk.numpy().array
+                assert np.issubdtype(arr.dtype, np.uint8)
+                out = arr.astype('float32')
+                out /= 127.5
+                out -= 1.0
+            task[output_chunk_name] = Chunk(
+                out, voxel_offset=chunk.voxel_offset,
+                voxel_size=chunk.voxel_size)
+            task['log']['timer'][name] = time() - start
+        yield task
+
+
+@main.command('inference')
+@click.option('--name', type=str, default='inference',
+              help='name of this operator')
+@click.option('--convnet-model', '-m', type=str, default=None,
+              help='convnet model path or type.')
+@click.option('--convnet-weight-path', '-w', type=str, default=None,
+              help='convnet weight path')
+@click.option('--input-patch-size', '-s', type=click.INT, nargs=3,
+              required=True, help='input patch size')
+@click.option('--output-patch-size', '-z', type=click.INT, nargs=3,
+              default=None, callback=default_none, help='output patch size')
+@click.option('--output-patch-overlap', '-v', type=click.INT, nargs=3,
+              default=(4, 64, 64), help='patch overlap')
+@click.option('--output-crop-margin', type=click.INT, nargs=3, default=None,
+              callback=default_none, help='margin size of output chunk cropping.')
+@click.option('--patch-num', '-n', default=None, callback=default_none,
+              type=click.INT, nargs=3, help='patch number in z,y,x.')
+@click.option('--num-input-channels', type=click.INT, default=1,
+              help='number of input channels')
+@click.option('--num-output-channels', '-c', type=click.INT, default=3,
+              help='number of output channels')
+@click.option('--dtype', '-d', type=click.Choice(['float32', 'float16']),
+              default='float32', help='output data type.')
+@click.option('--framework', '-f',
+              type=click.Choice(['universal', 'identity', 'pytorch']),
+              default='universal', help='inference framework')
+@click.option('--batch-size', '-b', type=click.INT, default=1,
+              help='mini batch size of input patch.')
+@click.option('--bump', type=click.Choice(['wu', 'zung']), default='wu',
+              help='bump function type (only support wu now!).')
+@click.option('--mask-output-chunk/--no-mask-output-chunk', default=False,
+              help='mask output chunk will make the whole chunk like one '
+                   'output patch. This will also work with non-aligned chunk size.')
+@click.option('--mask-myelin-threshold', '-y', default=None, type=click.FLOAT,
+              help='mask myelin if netoutput have myelin channel.')
+@click.option('--augment/--no-augment', default=False,
+              help='transform the input patch and transform back the output patch')
+@click.option('--input-chunk-name', '-i', type=str, default='chunk',
+              help='input chunk name')
+@click.option('--output-chunk-name', '-o', type=str, default='chunk',
+              help='output chunk name')
+@operator
+def inference(tasks, name, convnet_model, convnet_weight_path,
+              input_patch_size, output_patch_size, output_patch_overlap,
+              output_crop_margin, patch_num, num_input_channels,
+              num_output_channels, dtype, framework, batch_size, bump,
+              mask_output_chunk, mask_myelin_threshold, augment,
+              input_chunk_name, output_chunk_name):
+    """Perform convolutional network inference for chunks."""
+    from .inferencer import Inferencer
+    with Inferencer(
+            convnet_model, convnet_weight_path,
+            input_patch_size=input_patch_size,
+            output_patch_size=output_patch_size,
+            num_input_channels=num_input_channels,
+            num_output_channels=num_output_channels,
+            output_patch_overlap=output_patch_overlap,
+            output_crop_margin=output_crop_margin,
+            patch_num=patch_num, framework=framework, dtype=dtype,
+            batch_size=batch_size, bump=bump, augment=augment,
+            mask_output_chunk=mask_output_chunk,
+            mask_myelin_threshold=mask_myelin_threshold,
+            dry_run=state['dry_run']) as inferencer:
+        for task in tasks:
+            if task is not None:
+                if 'log' not in task:
+                    task['log'] = {'timer': {}}
+                start = time()
+                task[output_chunk_name] = inferencer(task[input_chunk_name])
+                task['log']['timer'][name] = time() - start
+                task['log']['compute_device'] = inferencer.compute_device
+            yield task
+
+
+@main.command('crop-margin')
+@click.option('--name', type=str, default='crop-margin',
+              help='name of this operator')
+@click.option('--margin-size', '-m', type=click.INT, nargs=6, default=None,
+              callback=default_none,
+              help='crop the chunk margin. It should have 6 values.')
+@click.option('--crop-bbox/--no-crop-bbox', default=False,
+              help='adjust the bounding box or not.')
+@click.option('--input-chunk-name', '-i', type=str, default='chunk',
+              help='input chunk name.')
+@click.option('--output-chunk-name', '-o', type=str, default='chunk',
+              help='output chunk name.')
+@operator
+def crop_margin(tasks, name, margin_size, crop_bbox, input_chunk_name,
+                output_chunk_name):
+    """Crop the margin of chunk."""
+    ops = None
+    for task in tasks:
+        if task is not None:
+            start = time()
+            chunk = task[input_chunk_name]
+            if margin_size:
+                if chunk.is_device:
+                    # HIP contiguous-copy kernel (chunk/base.py:691-726)
+                    from .ops import HipOps
+                    if ops is None:
+                        ops = HipOps(0)
+                    out = ops.crop_margin(chunk.array, list(margin_size))
+                    offset = tuple(o + m for o, m in
+                                   zip(chunk.voxel_offset, margin_size[:3]))
+                    task[output_chunk_name] = Chunk(
+                        out, voxel_offset=offset,
+                        voxel_size=chunk.voxel_size)
+                else:
+                    task[output_chunk_name] = chunk.crop_margin(
+                        margin_size=margin_size)
+            else:
+                task[output_chunk_name] = chunk.cutout(task['bbox'].slices)
+            task['log']['timer'][name] = time() - start
+        yield task
+
+
+@main.command('connected-components')
+@click.option('--name', type=str, default='connected-components',
+              help='threshold a map and get the targets.')
+@click.option('--input-chunk-name', '-i', type=str,
+              default=DEFAULT_CHUNK_NAME, help='input chunk name')
+@click.option('--output-chunk-name', '-o', type=str,
+              default=DEFAULT_CHUNK_NAME, help='output chunk name')
+@click.option('--threshold', '-t', type=click.FLOAT, default=None,
+              help='threshold to cut the map.')
+@click.option('--connectivity', '-c', type=click.Choice(['6', '18', '26']),
+              default='6', help='number of neighboring voxels used.')
+@operator
+def connected_components(tasks, name, input_chunk_name, output_chunk_name,
+                         threshold, connectivity):
+    """Threshold the probability map to get a segmentation."""
+    from .connected import connected_component
+    connectivity = int(connectivity)
+    for task in tasks:
+        if task is not None:
+            start = time()
+            task[output_chunk_name] = connected_component(
+                task[input_chunk_name], threshold=threshold,
+                connectivity=connectivity)
+            task['log']['timer'][name] = time() - start
+        yield task
+
+
+@main.command('plugin')
+@click.option('--name', type=str, default='plugin-1',
+              help='name of plugin. Multiple plugins should have different names.')
+@click.option('--input-names', '-i', type=str, default=None,
+              help='input names with delimiter of comma')
+@click.option('--output-names', '-o', type=str, default=None,
+              help='output names with dilimiter of comma')
+@click.option('--file', '-f', type=str, help='python file to call.')
+@click.option('--args', '-a', type=str, default=None,
+              help='arguments of plugin.')
+@operator
+def plugin(tasks, name, input_names, output_names, file, args):
+    """Insert custom program as a plugin."""
+    op = Plugin(file, name=name)
+    for task in tasks:
+        if task is not None:
+            start = time()
+            if input_names is not None:
+                inputs = [None if n == 'None' else task[n]
+                          for n in input_names.split(',')]
+            else:
+                inputs = []
+            outputs = op(inputs, args=args)
+            if isinstance(outputs, (list, tuple)):
+                names = output_names.split(',')
+                assert len(outputs) == len(names)
+                for n, o in zip(names, outputs):
+                    task[n] = o
+            elif output_names is not None:
+                assert ',' not in output_names
+                task[output_names] = outputs
+            else:
+                assert outputs is None
+            task['log']['timer'][name] = time() - start
+        yield task
+
+
+@main.command('save-npy')
+@click.option('--name', type=str, default='save-npy', help='name of operator')
+@click.option('--file-name', '-f', type=str, required=True,
+              help='output .npy path; {bbox} expands to the task bbox string')
+@click.option('--input-chunk-name', '-i', type=str,
+              default=DEFAULT_CHUNK_NAME, help='input chunk name')
+@operator
+def save_npy(tasks, name, file_name, input_chunk_name):
+    """Save a chunk as .npy (D2H at the pipeline edge)."""
+    for task in tasks:
+        if task is not None:
+            start = time()
+            fname = file_name
+            if '{bbox}' in fname and 'bbox' in task:
+                fname = fname.replace('{bbox}', task['bbox'].string)
+            task[input_chunk_name].to_npy(fname)
+            task['log']['timer'][name] = time() - start
+        yield task
+
+
+@main.command('load-npy')
+@click.option('--name', type=str, default='load-npy', help='name of operator')
+@click.option('--file-name', '-f', type=str, required=True,
+              help='input .npy path')
+@click.option('--voxel-offset', '-t', type=click.INT, nargs=3,
+              default=(0, 0, 0), help='voxel offset')
+@click.option('--output-chunk-name', '-o', type=str,
+              default=DEFAULT_CHUNK_NAME, help='output chunk name')
+@operator
+def load_npy(tasks, name, file_name, voxel_offset, output_chunk_name):
+    """Load a chunk from .npy."""
+    for task in tasks:
+        if task is not None:
+            start = time()
+            task[output_chunk_name] = Chunk.from_npy(
+                file_name, voxel_offset=voxel_offset)
+            task['log']['timer'][name] = time() - start
+        yield task
+
+
+@main.command('save-log')
+@click.option('--output-path', '-o', type=str, required=True,
+              help='directory for per-task log JSON files')
+@operator
+def save_log(tasks, output_path):
+    """Persist each task's timer log as JSON (reference upload_log shape)."""
+    os.makedirs(output_path, exist_ok=True)
+    for idx, task in enumerate(tasks):
+        if task is not None:
+            tag = task.get('bbox').string if 'bbox' in task else str(idx)
+            with open(os.path.join(output_path, f'{tag}.json'), 'w') as f:
+                json.dump(task['log'], f)
+        yield task
+
+
+@main.command('log-summary')
+@click.option('--log-dir', '-l', type=click.Path(exists=True, dir_okay=True),
+              required=True, help='directory of json log files.')
+@click.option('--output-size', '-s', type=click.INT, nargs=3, default=None,
+              callback=default_none, help='output size for voxels/sec')
+@generator
+def log_summary(log_dir, output_size):
+    """Compute the statistics of large scale run (the reference's
+    per-operator mean/max/min + voxels/sec report, log_summary.py:16-74)."""
+    import glob
+    timers = {}
+    for fname in glob.glob(os.path.join(log_dir, '*.json')):
+        with open(fname) as f:
+            log = json.load(f)
+        for op_name, secs in log.get('timer', {}).items():
+            timers.setdefault(op_name, []).append(secs)
+    total_mean = 0.0
+    for op_name, vals in sorted(timers.items()):
+        arr = np.array(vals)
+        print(f'{op_name}: mean={arr.mean():.3f}s max={arr.max():.3f}s '
+              f'min={arr.min():.3f}s n={len(arr)}')
+        total_mean += arr.mean()
+    if output_size is not None and total_mean > 0:
+        voxels = int(np.prod(output_size))
+        print(f'speed: {voxels / total_mean / 1e3:.1f} kv/s '
+              f'({voxels / total_mean / 1e6:.2f} mv/s)')
+    yield get_initial_task()

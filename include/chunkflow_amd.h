@@ -1,0 +1,114 @@
+/* chunkflow_amd C-ABI — the MI355X (gfx950) hot-path extension beneath the
+ * `chunkflow inference` operator surface.
+ *
+ * Each entry point replaces a reference hot-path function (SURVEY.md §8a/§8b;
+ * reference = seung-lab/chunkflow v1.1.7):
+ *   cfx_make_patch_mask      <- patch/patch_mask.py:15-68  (host, f64 internally)
+ *   cfx_cast_u8_f32_div      <- inferencer.py:395-399      (int chunk -> f32 / dtype_max)
+ *   cfx_normalize_intensity  <- flow/flow.py:1650-1669     (u8 -> f32, x/127.5 - 1)
+ *   cfx_extract_patches      <- inferencer.py:408-411 + chunk/base.py:761-781
+ *   cfx_blend_accumulate     <- inferencer.py:436-455 + chunk/base.py:792-807
+ *                               (fuses the engine's patch-mask multiply,
+ *                                pytorch.py:113, when mask != NULL)
+ *   cfx_build_chunk_mask     <- inferencer.py:294-333
+ *   cfx_reciprocal           <- inferencer.py:333
+ *   cfx_multiply_mask        <- inferencer.py:460-461 (ufunc chunk/base.py:418-453)
+ *   cfx_max                  <- inferencer.py:463-466 (assert < 1.0001)
+ *   cfx_crop_margin          <- chunk/base.py:691-726 (flow.py:2053-2084)
+ *   cfx_mask_using_last_channel <- chunk/base.py:685-689 (inferencer.py:468-477)
+ *
+ * Conventions: the caller owns every buffer (device pointers come from
+ * torch-ROCm tensors' data_ptr()); layout is contiguous C-order z-y-x with x
+ * fastest; all calls are stream-ordered on the context stream (adopt torch's
+ * stream via cfx_set_stream); int return = 0 on success, nonzero with
+ * cfx_last_error() set; one context per GPU rank, single-threaded per
+ * context.
+ */
+#ifndef CHUNKFLOW_AMD_H
+#define CHUNKFLOW_AMD_H
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+typedef struct cfx_ctx cfx_ctx;
+
+/* ---- lifecycle -------------------------------------------------------- */
+cfx_ctx* cfx_init(int device);
+void     cfx_destroy(cfx_ctx* ctx);
+const char* cfx_last_error(void);
+int      cfx_version(void);
+/* adopt an existing HIP stream (e.g. torch.cuda.current_stream().cuda_stream);
+ * stream may be NULL for the legacy default stream */
+int cfx_set_stream(cfx_ctx* ctx, void* hip_stream);
+int cfx_sync(cfx_ctx* ctx);
+
+/* ---- host precompute --------------------------------------------------- */
+/* Wu bump-weight patch mask, float64 pipeline, f32 result (pz*py*px floats).
+ * Runs on the HOST; upload the result once per geometry. */
+int cfx_make_patch_mask(const int patch_size[3], const int overlap[3],
+                        float* out);
+
+/* ---- device kernels (all pointers are DEVICE pointers) ----------------- */
+/* out[i] = (float)in[i] / 127.5f - 1.0f */
+int cfx_normalize_intensity(cfx_ctx* ctx, const unsigned char* in, float* out,
+                            long long n);
+/* out[i] = (float)in[i] / divisor  (divisor = integer dtype max) */
+int cfx_cast_u8_f32_div(cfx_ctx* ctx, const unsigned char* in, float* out,
+                        long long n, float divisor);
+/* gather n_patches windows of an f32 chunk (channels, D, H, W) into a batch
+ * buffer (n_patches, channels, pz, py, px); starts_zyx is a HOST array of
+ * n_patches*3 ints, chunk-local coordinates */
+int cfx_extract_patches(cfx_ctx* ctx, const float* chunk, int channels,
+                        const int chunk_dims[3], const int* starts_zyx,
+                        int n_patches, const int patch_size[3], float* out);
+/* out[c, region] += patch[c, region'] * mask[region'], clipped to the output
+ * bounds; offset_zyx is the patch start relative to the output buffer origin;
+ * mask may be NULL (plain accumulate, e.g. a pre-masked universal plugin) */
+int cfx_blend_accumulate(cfx_ctx* ctx, float* out, int channels,
+                         const int out_dims[3], const float* patch,
+                         const int patch_dims[3], const int offset_zyx[3],
+                         const float* mask);
+/* zero mask_out (out_dims), blend patch_mask at each of n offsets (HOST
+ * array, n*3 ints, relative to the output origin), then reciprocal */
+int cfx_build_chunk_mask(cfx_ctx* ctx, float* mask_out, const int out_dims[3],
+                         const float* patch_mask, const int patch_dims[3],
+                         const int* offsets_zyx, int n);
+int cfx_reciprocal(cfx_ctx* ctx, float* buf, long long n);
+/* out[c*n + i] *= mask[i] for every channel c (mask-normalize) */
+int cfx_multiply_mask(cfx_ctx* ctx, float* out, const float* mask,
+                      int channels, long long n_voxels);
+/* synchronous max over n floats (the <1.0001 sanity assert) */
+int cfx_max(cfx_ctx* ctx, const float* buf, long long n, float* host_max);
+/* contiguous copy dropping margins[6] = -z,-y,-x,+z,+y,+x */
+int cfx_crop_margin(cfx_ctx* ctx, const float* in, float* out, int channels,
+                    const int in_dims[3], const int margins[6]);
+/* out[(channels-1), dims] = in[:channels-1] * (in[channels-1] < threshold) */
+int cfx_mask_using_last_channel(cfx_ctx* ctx, const float* in, float* out,
+                                int channels, const int dims[3],
+                                float threshold);
+
+/* ---- kernel timing (HIP events on the context stream) ------------------ */
+enum cfx_kernel_id {
+    CFX_K_BLEND = 0,
+    CFX_K_EXTRACT = 1,
+    CFX_K_NORMALIZE = 2,
+    CFX_K_CAST = 3,
+    CFX_K_RECIPROCAL = 4,
+    CFX_K_MASKMUL = 5,
+    CFX_K_CROP = 6,
+    CFX_K_MAX = 7,
+    CFX_K_MYELIN = 8,
+    CFX_K_COUNT = 9
+};
+int cfx_profile_enable(cfx_ctx* ctx, int enable);
+int cfx_profile_reset(cfx_ctx* ctx);
+/* drains pending events (syncs the stream); bytes = accumulated ALGORITHMIC
+ * bytes of the profiled launches (computed from region sizes, not counters) */
+int cfx_profile_get(cfx_ctx* ctx, int kernel_id, unsigned long long* count,
+                    double* total_ms, double* bytes);
+
+#ifdef __cplusplus
+}
+#endif
+#endif /* CHUNKFLOW_AMD_H */

@@ -11,6 +11,8 @@ if REPO_ROOT not in sys.path:
 def pytest_configure(config):
     config.addinivalue_line(
         'markers', 'gpu: needs a real MI355X GPU (run with -m gpu)')
+    config.addinivalue_line(
+        'markers', 'slow: takes more than a few seconds on CPU')
 
 
 def pytest_collection_modifyitems(config, items):

@@ -1,0 +1,95 @@
+"""CLI surface tests: chained operators, flags, plugin op, task stream."""
+import os
+import re
+
+import numpy as np
+import pytest
+import torch
+from click.testing import CliRunner
+
+from chunkflow_amd.flow import main
+
+GPU = torch.cuda.is_available()
+
+
+def run_cli(args):
+    runner = CliRunner()
+    result = runner.invoke(main, args, catch_exceptions=False)
+    assert result.exit_code == 0, result.output
+    return result
+
+
+def test_generate_tasks_stream():
+    r = run_cli(['generate-tasks', '--roi-size', '512', '1024', '2048',
+                 '--chunk-size', '512', '512', '512'])
+    assert 'total number of tasks: 8' in r.output
+
+
+def test_create_chunk_save_npy(tmp_path):
+    out = tmp_path / 'c.npy'
+    run_cli(['create-chunk', '--size', '16', '24', '32', '--dtype', 'uint8',
+             '--pattern', 'sin', 'save-npy', '-f', str(out)])
+    arr = np.load(out)
+    assert arr.shape == (16, 24, 32) and arr.dtype == np.uint8
+
+
+@pytest.mark.skipif(GPU, reason='CPU plumbing pipeline; GPU covered in -m gpu')
+def test_cpu_pipeline_config1_shape(tmp_path, golden):
+    """BASELINE config 1 (plumbing, no GPU): create-chunk -> inference
+    (identity) -> crop-margin -> save-npy, small geometry."""
+    _, arrays = golden
+    out = tmp_path / 'out.npy'
+    run_cli(['create-chunk', '--size', '20', '68', '72',
+             '--dtype', 'uint8', '--pattern', 'sin',
+             'inference', '-s', '10', '32', '32',
+             '--output-patch-overlap', '2', '8', '8',
+             '--framework', 'identity', '--batch-size', '3',
+             '--num-output-channels', '3', '--mask-output-chunk',
+             'crop-margin', '-m', '1', '2', '3', '1', '2', '3',
+             'save-npy', '-f', str(out)])
+    got = np.load(out)
+    assert got.shape == (3, 18, 64, 66)
+    sin = arrays['sin_20x68x72_u8'].astype(np.float32) / 255.0
+    np.testing.assert_allclose(got[0], sin[1:-1, 2:-2, 3:-3],
+                               rtol=1e-5, atol=1e-5)
+
+
+@pytest.mark.skipif(GPU, reason='CPU plumbing pipeline')
+def test_cpu_pipeline_normalize_plugin_cc(tmp_path):
+    """normalize-intensity + plugin + connected-components chain on CPU."""
+    plugin = tmp_path / 'clip01.py'
+    plugin.write_text(
+        'import numpy as np\n'
+        'def execute(chunk):\n'
+        '    return np.clip(np.asarray(chunk.array), 0, 1)'
+        '.astype(np.float32)\n')
+    out = tmp_path / 'seg.npy'
+    run_cli(['create-chunk', '--size', '12', '20', '24', '--dtype', 'uint8',
+             'normalize-intensity',
+             'plugin', '-f', str(plugin), '-i', 'chunk', '-o', 'chunk',
+             'connected-components', '-t', '0.5', '-c', '6',
+             'save-npy', '-f', str(out)])
+    seg = np.load(out)
+    assert seg.dtype == np.uint32
+    assert seg.max() >= 1
+
+
+def test_dry_run_inference():
+    r = run_cli(['--dry-run', 'create-chunk', '--size', '20', '68', '72',
+                 'inference', '-s', '10', '32', '32',
+                 '--output-patch-overlap', '2', '8', '8',
+                 '--framework', 'identity', '--mask-output-chunk'])
+    assert r.exit_code == 0
+
+
+def test_log_summary(tmp_path):
+    logdir = tmp_path / 'log'
+    logdir.mkdir()
+    (logdir / 'a.json').write_text(
+        '{"timer": {"inference": 2.0, "crop-margin": 0.5}}')
+    (logdir / 'b.json').write_text(
+        '{"timer": {"inference": 4.0, "crop-margin": 0.7}}')
+    r = run_cli(['log-summary', '-l', str(logdir),
+                 '-s', '512', '512', '512'])
+    assert 'inference: mean=3.000s' in r.output
+    assert re.search(r'speed: .* kv/s', r.output)

@@ -1,0 +1,212 @@
+"""Product host-side tests (no GPU): Cartesian/BoundingBox grid math, Chunk
+data model, product patch mask vs golden CRCs, and the full Inferencer on the
+CPU plumbing path (TorchOps — only reachable on GPU-less machines) against
+the reference golden fixtures."""
+import os
+import zlib
+
+import numpy as np
+import pytest
+import torch
+
+from chunkflow_amd import BoundingBox, BoundingBoxes, Cartesian, Chunk
+from chunkflow_amd.patch_mask import make_patch_mask
+
+GPU = torch.cuda.is_available()
+
+
+def crc(arr):
+    return zlib.crc32(np.ascontiguousarray(arr).tobytes())
+
+
+# --- coordinates -----------------------------------------------------------
+def test_cartesian_arithmetic():
+    a = Cartesian(1, 2, 3)
+    assert a + (1, 1, 1) == Cartesian(2, 3, 4)
+    assert a * 2 == Cartesian(2, 4, 6)
+    assert Cartesian(20, 256, 256) - Cartesian(4, 64, 64) == \
+        Cartesian(16, 192, 192)
+    assert Cartesian(10, 10, 10) // 3 == Cartesian(3, 3, 3)
+    assert Cartesian(4, 64, 64) <= Cartesian(4, 64, 64)
+
+
+def test_bbox_roundtrip():
+    b = BoundingBox.from_delta((1, 2, 3), (10, 20, 30))
+    assert b.shape == Cartesian(10, 20, 30)
+    assert BoundingBox.from_string(b.string) == b
+    assert b.slices == (slice(1, 11), slice(2, 22), slice(3, 33))
+
+
+def test_generate_tasks_grid():
+    # 8 independent 512^3 chunks (the config-3 shard unit)
+    boxes = BoundingBoxes.from_manual_setup(
+        (512, 512, 512), roi_size=(512, 1024, 2048))
+    assert len(boxes) == 8
+    assert boxes[0].start == Cartesian(0, 0, 0)
+    assert boxes[-1].start == Cartesian(0, 512, 1536)
+    # clamped tail grid: ceil semantics
+    boxes = BoundingBoxes.from_manual_setup((64, 64, 64),
+                                            roi_size=(100, 64, 64))
+    assert len(boxes) == 2
+    assert boxes[1].start.z == 64
+    # respect_stop clips the tail box
+    boxes = BoundingBoxes.from_manual_setup(
+        (64, 64, 64), roi_size=(100, 64, 64), respect_chunk_size=False)
+    assert boxes[1].shape == Cartesian(36, 64, 64)
+
+
+# --- chunk data model ------------------------------------------------------
+def test_chunk_create_sin_matches_reference(golden):
+    meta, arrays = golden
+    c = Chunk.create(size=(20, 68, 72), dtype='uint8', pattern='sin')
+    np.testing.assert_array_equal(c.array, arrays['sin_20x68x72_u8'])
+    assert crc(c.array) == meta['cases']['sin_20x68x72_u8']['crc32']
+
+
+@pytest.mark.slow
+def test_chunk_create_sin_512(golden):
+    meta, _ = golden
+    c = Chunk.create(size=(512, 512, 512), dtype='uint8', pattern='sin')
+    assert crc(c.array) == meta['cases']['sin_512_u8']['crc32']
+
+
+def test_chunk_cutout_blend_crop():
+    arr = np.arange(4 * 5 * 6, dtype=np.float32).reshape(4, 5, 6)
+    c = Chunk(arr.copy(), voxel_offset=(10, 20, 30))
+    cut = c.cutout((slice(11, 13), slice(21, 24), slice(31, 35)))
+    np.testing.assert_array_equal(cut.array, arr[1:3, 1:4, 1:5])
+    assert cut.voxel_offset == Cartesian(11, 21, 31)
+
+    out = Chunk(np.zeros((2, 4, 5, 6), dtype=np.float32),
+                voxel_offset=(0, 0, 0))
+    patch = Chunk(np.ones((2, 3, 3, 3), dtype=np.float32),
+                  voxel_offset=(-1, 3, 4))  # clips on three sides
+    out.blend(patch)
+    assert out.array.sum() == 2 * 2 * 2 * 2
+    cropped = c.crop_margin(margin_size=(1, 1, 1))
+    np.testing.assert_array_equal(cropped.array, arr[1:-1, 1:-1, 1:-1])
+    assert cropped.voxel_offset == Cartesian(11, 21, 31)
+    cropped6 = c.crop_margin(margin_size=(1, 0, 0, 2, 1, 3))
+    np.testing.assert_array_equal(cropped6.array, arr[1:-2, 0:-1, 0:-3])
+
+
+def test_chunk_mask_using_last_channel():
+    arr = np.random.rand(4, 3, 3, 3).astype(np.float32)
+    c = Chunk(arr.copy())
+    out = c.mask_using_last_channel(threshold=0.5)
+    keep = arr[-1] < 0.5
+    np.testing.assert_array_equal(out.array, arr[:3] * keep)
+
+
+# --- product patch mask ----------------------------------------------------
+def test_product_patch_mask_golden(golden):
+    meta, arrays = golden
+    for name in ('mask_20x256x256_ov4x64x64', 'mask_10x32x32_ov2x8x8'):
+        case = meta['cases'][name]
+        m = make_patch_mask(tuple(case['patch_size']), tuple(case['overlap']))
+        assert crc(m) == case['crc32']
+
+
+def test_c_abi_patch_mask_bitexact():
+    from chunkflow_amd.hip import make_patch_mask_c
+    for geom in (((10, 32, 32), (2, 8, 8)), ((20, 128, 128), (4, 32, 32))):
+        c = make_patch_mask_c(*geom)
+        p = make_patch_mask(*geom)
+        np.testing.assert_array_equal(c, p)
+
+
+# --- full inferencer, CPU plumbing path (GPU-less machines only) -----------
+@pytest.mark.skipif(GPU, reason='CPU plumbing path is refused on a GPU box')
+class TestInferencerCPU:
+    def _run(self, golden, **kw):
+        from chunkflow_amd.inferencer import Inferencer
+        _, arrays = golden
+        chunk = Chunk(arrays['e2e_input_u8'].copy(),
+                      voxel_offset=kw.pop('voxel_offset', (0, 0, 0)))
+        inf = Inferencer(kw.pop('model', None), kw.pop('weights', None),
+                         (10, 32, 32), output_patch_overlap=(2, 8, 8),
+                         mask_output_chunk=True, **kw)
+        assert not inf.ops.is_hip
+        return inf(chunk)
+
+    def test_identity(self, golden):
+        _, arrays = golden
+        out = self._run(golden, framework='identity', batch_size=3)
+        np.testing.assert_allclose(out.numpy().array,
+                                   arrays['e2e_identity_out'],
+                                   rtol=1e-6, atol=1e-7)
+
+    def test_identity_offset(self, golden):
+        _, arrays = golden
+        out = self._run(golden, framework='identity', batch_size=4,
+                        voxel_offset=(7, 11, 13))
+        assert out.voxel_offset == Cartesian(7, 11, 13)
+        np.testing.assert_allclose(out.numpy().array,
+                                   arrays['e2e_identity_offset_out'],
+                                   rtol=1e-6, atol=1e-7)
+
+    def test_identity_myelin(self, golden):
+        _, arrays = golden
+        out = self._run(golden, framework='identity', batch_size=3,
+                        num_output_channels=4, mask_myelin_threshold=0.3)
+        np.testing.assert_allclose(out.numpy().array,
+                                   arrays['e2e_identity_myelin_out'],
+                                   rtol=1e-6, atol=1e-7)
+
+    def test_pytorch_model_file(self, golden, golden_dir):
+        _, arrays = golden
+        out = self._run(
+            golden, framework='pytorch', batch_size=1,
+            model=os.path.join(golden_dir, 'ref_model.py'),
+            weights=os.path.join(golden_dir, 'ref_model_weights.pt'))
+        np.testing.assert_allclose(out.numpy().array,
+                                   arrays['e2e_pytorch_out'],
+                                   rtol=1e-5, atol=1e-6)
+
+    def test_universal_plugin_contract(self, golden, tmp_path):
+        # the reference universal model-file contract:
+        # PatchInferencer(weight_path, mask).__call__(ndarray) -> masked ndarray
+        plugin = tmp_path / 'universal_identity.py'
+        plugin.write_text(
+            'import numpy as np\n'
+            'class PatchInferencer:\n'
+            '    def __init__(self, weight_path, mask):\n'
+            '        self.mask = mask\n'
+            '    def __call__(self, patch):\n'
+            '        out = np.repeat(patch.astype(np.float32), 3, axis=1)\n'
+            '        return out * self.mask\n')
+        _, arrays = golden
+        out = self._run(golden, framework='universal', batch_size=3,
+                        model=str(plugin))
+        np.testing.assert_allclose(out.numpy().array,
+                                   arrays['e2e_identity_out'],
+                                   rtol=1e-6, atol=1e-7)
+
+    def test_aligned_mode(self, golden):
+        """mask_output_chunk=False: aligned chunk, engine-masked patches,
+        no chunk normalize, margins cropped (vs oracle-style expectation:
+        interior equals input/255 within the mask sum tolerance)."""
+        from chunkflow_amd.inferencer import Inferencer
+        chunk = Chunk.create(size=(18, 56, 56), dtype='uint8', pattern='sin')
+        inf = Inferencer(None, None, (10, 32, 32),
+                         output_patch_overlap=(2, 8, 8),
+                         framework='identity', num_output_channels=1,
+                         batch_size=2, mask_output_chunk=False,
+                         input_size=(18, 56, 56))
+        out = inf(chunk)
+        assert out.shape == (1, 14, 40, 40)
+        assert out.voxel_offset == Cartesian(2, 8, 8)
+        expect = chunk.array.astype(np.float32)[2:-2, 8:-8, 8:-8] / 255.0
+        np.testing.assert_allclose(out.numpy().array[0], expect,
+                                   rtol=1e-2, atol=1e-2)
+
+    def test_all_zero_shortcut(self, golden):
+        from chunkflow_amd.inferencer import Inferencer
+        chunk = Chunk.create(size=(12, 40, 40), dtype='uint8',
+                             pattern='zero')
+        inf = Inferencer(None, None, (10, 32, 32),
+                         output_patch_overlap=(2, 8, 8),
+                         framework='identity', batch_size=2,
+                         mask_output_chunk=True)
+        out = inf(chunk)
+        assert out.numpy().array.max() == 0
