@@ -313,3 +313,32 @@ def test_cli_gpu_pipeline(tmp_path, golden):
     sin = arrays['sin_20x68x72_u8'].astype(np.float32) / 255.0
     np.testing.assert_allclose(got[0], sin[1:-1, 2:-2, 3:-3],
                                rtol=1e-5, atol=1e-5)
+
+
+def test_patch_num_aligned_gpu():
+    from chunkflow_amd.chunk import Chunk
+    chunk = Chunk.create(size=(18, 56, 56), dtype='uint8', pattern='sin')
+    inf = _hip_inferencer(patch_size=(10, 32, 32), overlap=(2, 8, 8),
+                          framework='identity', num_output_channels=2,
+                          batch_size=5, dtype='float16',
+                          mask_output_chunk=False, patch_num=(2, 2, 2))
+    out = inf(chunk)
+    assert out.shape == (2, 14, 40, 40)
+    expect = chunk.array.astype(np.float32)[2:-2, 8:-8, 8:-8] / 255.0
+    np.testing.assert_allclose(out.numpy().array[0].astype(np.float32),
+                               expect, rtol=1e-3, atol=2.0 / 255.0)
+
+
+def test_nonaligned_vs_oracle_gpu():
+    from chunkflow_amd.chunk import Chunk
+    from oracle import oracle_inference
+    rng = np.random.RandomState(11)
+    arr = rng.randint(0, 256, size=(23, 71, 66), dtype=np.uint8)
+    inf = _hip_inferencer(patch_size=(12, 32, 32), overlap=(4, 8, 8),
+                          framework='identity', num_output_channels=2,
+                          batch_size=7, mask_output_chunk=True)
+    out = inf(Chunk(arr.copy(), voxel_offset=(1, 2, 3)))
+    ref = oracle_inference(arr, (12, 32, 32), (4, 8, 8),
+                           num_output_channels=2, batch_size=7,
+                           offset=(1, 2, 3))
+    np.testing.assert_allclose(out.numpy().array, ref, rtol=1e-5, atol=1e-6)

@@ -210,3 +210,40 @@ class TestInferencerCPU:
                          mask_output_chunk=True)
         out = inf(chunk)
         assert out.numpy().array.max() == 0
+
+
+@pytest.mark.skipif(GPU, reason='CPU plumbing path is refused on a GPU box')
+def test_patch_num_aligned_mode():
+    """patch_num mode (reference test_aligned_patch_num shape): aligned
+    inference with --patch-num instead of input size, f16 output dtype."""
+    from chunkflow_amd.inferencer import Inferencer
+    chunk = Chunk.create(size=(18, 56, 56), dtype='uint8', pattern='sin')
+    inf = Inferencer(None, None, (10, 32, 32),
+                     output_patch_overlap=(2, 8, 8), framework='identity',
+                     num_output_channels=2, batch_size=5, dtype='float16',
+                     mask_output_chunk=False, patch_num=(2, 2, 2))
+    assert tuple(inf.input_size) == (18, 56, 56)
+    out = inf(chunk)
+    assert out.numpy().array.dtype == np.float16
+    assert out.shape == (2, 14, 40, 40)
+    expect = chunk.array.astype(np.float32)[2:-2, 8:-8, 8:-8] / 255.0
+    np.testing.assert_allclose(out.numpy().array[0].astype(np.float32),
+                               expect, rtol=1e-3, atol=2.0 / 255.0)
+
+
+@pytest.mark.skipif(GPU, reason='CPU plumbing path is refused on a GPU box')
+def test_nonaligned_vs_oracle_random_geometry():
+    """A fresh odd-size case: product CPU path vs the oracle directly."""
+    from chunkflow_amd.inferencer import Inferencer
+    from oracle import oracle_inference
+    rng = np.random.RandomState(11)
+    arr = rng.randint(0, 256, size=(23, 71, 66), dtype=np.uint8)
+    inf = Inferencer(None, None, (12, 32, 32),
+                     output_patch_overlap=(4, 8, 8), framework='identity',
+                     num_output_channels=2, batch_size=7,
+                     mask_output_chunk=True)
+    out = inf(Chunk(arr.copy(), voxel_offset=(1, 2, 3)))
+    ref = oracle_inference(arr, (12, 32, 32), (4, 8, 8),
+                           num_output_channels=2, batch_size=7,
+                           offset=(1, 2, 3))
+    np.testing.assert_allclose(out.numpy().array, ref, rtol=1e-6, atol=1e-7)
