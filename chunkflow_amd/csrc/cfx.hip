@@ -230,36 +230,57 @@ struct ExtractArgs {
     int starts[64 * 3];    // chunk-local (z, y, x) starts
 };
 
-// one wave per x-line, 4 lines per block (see k_blend note)
+// one wave per x-line, G=4 lines batched per wave iteration (k_blend note)
 template <bool VEC>
 __global__ void k_extract(const float* __restrict__ chunk, int C, int D,
                           int H, int W, ExtractArgs args, int pz, int py,
                           int px, float* __restrict__ out) {
+    constexpr int G = 4;
     long long n_lines = (long long)args.n * C * pz * py;
+    long long n_groups = (n_lines + G - 1) / G;
     long long stride = (long long)gridDim.x * blockDim.y;
-    for (long long line = (long long)blockIdx.x * blockDim.y + threadIdx.y;
-         line < n_lines; line += stride) {
-        int y = (int)(line % py);
-        long long t = line / py;
-        int z = (int)(t % pz);
-        t /= pz;
-        int c = (int)(t % C);
-        int p = (int)(t / C);
-        int z0 = args.starts[p * 3 + 0];
-        int y0 = args.starts[p * 3 + 1];
-        int x0 = args.starts[p * 3 + 2];
-        const float* src =
-            chunk + (((long long)c * D + z0 + z) * H + y0 + y) * W + x0;
-        float* dst =
-            out + ((((long long)p * C + c) * pz + z) * py + y) * px;
+    const int px4 = px / 4;
+    for (long long g = (long long)blockIdx.x * blockDim.y + threadIdx.y;
+         g < n_groups; g += stride) {
+        long long line0 = g * G;
+        int nl = (int)(n_lines - line0 < G ? n_lines - line0 : G);
+        const float* src[G];
+        float* dst[G];
+#pragma unroll
+        for (int j = 0; j < G; ++j) {
+            long long line = line0 + (j < nl ? j : 0);
+            int y = (int)(line % py);
+            long long t = line / py;
+            int z = (int)(t % pz);
+            t /= pz;
+            int c = (int)(t % C);
+            int p = (int)(t / C);
+            int z0 = args.starts[p * 3 + 0];
+            int y0 = args.starts[p * 3 + 1];
+            int x0 = args.starts[p * 3 + 2];
+            src[j] = chunk +
+                (((long long)c * D + z0 + z) * H + y0 + y) * W + x0;
+            dst[j] = out +
+                ((((long long)p * C + c) * pz + z) * py + y) * px;
+        }
         if (VEC) {
-            const float4* s4 = reinterpret_cast<const float4*>(src);
-            float4* d4 = reinterpret_cast<float4*>(dst);
-            for (int x = threadIdx.x; x < px / 4; x += 64)
-                d4[x] = s4[x];
+            for (int x = threadIdx.x; x < px4; x += 64) {
+                float4 v[G];
+#pragma unroll
+                for (int j = 0; j < G; ++j)
+                    if (j < nl)
+                        v[j] = reinterpret_cast<const float4*>(src[j])[x];
+#pragma unroll
+                for (int j = 0; j < G; ++j)
+                    if (j < nl)
+                        reinterpret_cast<float4*>(dst[j])[x] = v[j];
+            }
         } else {
-            for (int x = threadIdx.x; x < px; x += 64)
-                dst[x] = src[x];
+            for (int x = threadIdx.x; x < px; x += 64) {
+#pragma unroll
+                for (int j = 0; j < G; ++j)
+                    if (j < nl) dst[j][x] = src[j][x];
+            }
         }
     }
 }
@@ -285,7 +306,8 @@ extern "C" int cfx_extract_patches(cfx_ctx* ctx, const float* chunk,
         hipEvent_t e0;
         if (prof_begin(ctx, &e0)) return -1;
         long long n_lines = (long long)n * channels * pz * py;
-        int blocks = (int)std::min<long long>((n_lines + 3) / 4, 8192);
+        long long n_groups = (n_lines + 3) / 4;
+        int blocks = (int)std::min<long long>((n_groups + 3) / 4, 8192);
         float* dst = out + (long long)base * channels * pvox;
         if (vec)
             hipLaunchKernelGGL(k_extract<true>, dim3(blocks), dim3(64, 4),
@@ -310,10 +332,12 @@ extern "C" int cfx_extract_patches(cfx_ctx* ctx, const float* chunk,
 // atomics-free: within a launch each output voxel is written exactly once;
 // patches overlap only ACROSS launches, which the stream serializes in the
 // reference's own blend order). Lines along x, float4 when aligned.
-// One WAVE (64 lanes) per x-line, 4 waves per 256-thread block
-// (threadIdx.y = line slot): at rx = 256 each lane moves exactly one float4
-// per line, so every lane of the block is active — the flat one-line-per-
-// block form left 3/4 of the lanes idle and measured only ~35% of HBM peak.
+// One WAVE (64 lanes) per x-line, 4 waves per {64,4} block, and each wave
+// iteration batches G=4 lines: the lane issues all 4 lines' loads (up to 12
+// independent float4 loads in flight) before any accumulate, which is what
+// hides HBM latency here — a single float4 per lane per line measured only
+// ~44% of peak (latency-bound), the flat 256-thread-per-line form ~35%
+// (3/4 idle lanes).
 template <bool VEC, bool MASKED>
 __global__ void k_blend(float* __restrict__ out, int OD, int OH, int OW,
                         const float* __restrict__ patch, int PD, int PH,
@@ -321,45 +345,70 @@ __global__ void k_blend(float* __restrict__ out, int OD, int OH, int OW,
                         int dz0, int dy0, int dx0,   // region start in out
                         int pz0, int py0, int px0,   // region start in patch
                         int rz, int ry, int rx) {
+    constexpr int G = 4;  // lines batched per wave iteration
     long long n_lines = (long long)C * rz * ry;
+    long long n_groups = (n_lines + G - 1) / G;
     long long stride = (long long)gridDim.x * blockDim.y;
-    for (long long line = (long long)blockIdx.x * blockDim.y + threadIdx.y;
-         line < n_lines; line += stride) {
-        int y = (int)(line % ry);
-        long long t = line / ry;
-        int z = (int)(t % rz);
-        int c = (int)(t / rz);
-        float* o =
-            out + (((long long)c * OD + dz0 + z) * OH + dy0 + y) * OW + dx0;
-        const float* p =
-            patch + (((long long)c * PD + pz0 + z) * PH + py0 + y) * PW + px0;
-        const float* m = MASKED
-            ? mask + (((long long)(pz0 + z)) * PH + py0 + y) * PW + px0
-            : nullptr;
+    const int rx4 = rx / 4;
+    for (long long g = (long long)blockIdx.x * blockDim.y + threadIdx.y;
+         g < n_groups; g += stride) {
+        long long line0 = g * G;
+        int nl = (int)(n_lines - line0 < G ? n_lines - line0 : G);
+        float* o[G];
+        const float* p[G];
+        const float* m[G];
+#pragma unroll
+        for (int j = 0; j < G; ++j) {
+            long long line = line0 + (j < nl ? j : 0);
+            int y = (int)(line % ry);
+            long long t = line / ry;
+            int z = (int)(t % rz);
+            int c = (int)(t / rz);
+            o[j] = out +
+                (((long long)c * OD + dz0 + z) * OH + dy0 + y) * OW + dx0;
+            p[j] = patch +
+                (((long long)c * PD + pz0 + z) * PH + py0 + y) * PW + px0;
+            if (MASKED)
+                m[j] = mask +
+                    (((long long)(pz0 + z)) * PH + py0 + y) * PW + px0;
+        }
         if (VEC) {
-            float4* o4 = reinterpret_cast<float4*>(o);
-            const float4* p4 = reinterpret_cast<const float4*>(p);
-            const float4* m4 = reinterpret_cast<const float4*>(m);
-            for (int x = threadIdx.x; x < rx / 4; x += 64) {
-                float4 ov = o4[x];
-                float4 pv = p4[x];
-                if (MASKED) {
-                    float4 mv = m4[x];
-                    ov.x += pv.x * mv.x;
-                    ov.y += pv.y * mv.y;
-                    ov.z += pv.z * mv.z;
-                    ov.w += pv.w * mv.w;
-                } else {
-                    ov.x += pv.x;
-                    ov.y += pv.y;
-                    ov.z += pv.z;
-                    ov.w += pv.w;
+            for (int x = threadIdx.x; x < rx4; x += 64) {
+                float4 ov[G], pv[G], mv[G];
+#pragma unroll
+                for (int j = 0; j < G; ++j) {
+                    if (j < nl) {
+                        ov[j] = reinterpret_cast<float4*>(o[j])[x];
+                        pv[j] = reinterpret_cast<const float4*>(p[j])[x];
+                        if (MASKED)
+                            mv[j] = reinterpret_cast<const float4*>(m[j])[x];
+                    }
                 }
-                o4[x] = ov;
+#pragma unroll
+                for (int j = 0; j < G; ++j) {
+                    if (j < nl) {
+                        if (MASKED) {
+                            ov[j].x += pv[j].x * mv[j].x;
+                            ov[j].y += pv[j].y * mv[j].y;
+                            ov[j].z += pv[j].z * mv[j].z;
+                            ov[j].w += pv[j].w * mv[j].w;
+                        } else {
+                            ov[j].x += pv[j].x;
+                            ov[j].y += pv[j].y;
+                            ov[j].z += pv[j].z;
+                            ov[j].w += pv[j].w;
+                        }
+                        reinterpret_cast<float4*>(o[j])[x] = ov[j];
+                    }
+                }
             }
         } else {
-            for (int x = threadIdx.x; x < rx; x += 64)
-                o[x] += MASKED ? p[x] * m[x] : p[x];
+            for (int x = threadIdx.x; x < rx; x += 64) {
+#pragma unroll
+                for (int j = 0; j < G; ++j)
+                    if (j < nl)
+                        o[j][x] += MASKED ? p[j][x] * m[j][x] : p[j][x];
+            }
         }
     }
 }
@@ -389,7 +438,8 @@ static int blend_one(cfx_ctx* ctx, float* out, int C, const int out_dims[3],
     bool vec = (out_dims[2] % 4 == 0) && (patch_dims[2] % 4 == 0) &&
                (d0[2] % 4 == 0) && (p0[2] % 4 == 0) && (r[2] % 4 == 0);
     long long n_lines = (long long)C * r[0] * r[1];
-    int blocks = (int)std::min<long long>((n_lines + 3) / 4, 8192);
+    long long n_groups = (n_lines + 3) / 4;          // G=4 lines per group
+    int blocks = (int)std::min<long long>((n_groups + 3) / 4, 8192);
     hipEvent_t e0;
     if (prof_begin(ctx, &e0)) return -1;
 #define CFX_LAUNCH_BLEND(V, M)                                              \
