@@ -342,3 +342,45 @@ def test_nonaligned_vs_oracle_gpu():
                            num_output_channels=2, batch_size=7,
                            offset=(1, 2, 3))
     np.testing.assert_allclose(out.numpy().array, ref, rtol=1e-5, atol=1e-6)
+
+
+def test_cli_gpu_config4_chain(tmp_path):
+    """BASELINE config 4 shape at reduced size: create-chunk ->
+    normalize-intensity (HIP) -> inference (HIP) -> crop-margin (HIP) ->
+    connected-components -> save-npy, all chained through the CLI.
+    normalize-intensity feeds [-1,1] floats into inference (the reference
+    chain does the same: float input skips the /255 normalize)."""
+    from click.testing import CliRunner
+    from chunkflow_amd.flow import main
+    out = tmp_path / 'seg.npy'
+    r = CliRunner().invoke(main, [
+        'create-chunk', '--size', '24', '80', '96', '--dtype', 'uint8',
+        '--pattern', 'sin',
+        'normalize-intensity',
+        'inference', '-s', '12', '32', '32',
+        '--output-patch-overlap', '4', '8', '8',
+        '--framework', 'identity', '--batch-size', '6',
+        '--num-output-channels', '3', '--mask-output-chunk',
+        'crop-margin', '-m', '2', '4', '4', '2', '4', '4',
+        'connected-components', '-t', '0.0', '-c', '6',
+        'save-npy', '-f', str(out)], catch_exceptions=False)
+    assert r.exit_code == 0, r.output
+    seg = np.load(out)
+    assert seg.shape == (20, 72, 88)
+    assert seg.dtype == np.uint32
+    assert seg.max() >= 1
+
+    # the same chain against the oracle: normalize-intensity then identity
+    # inference == normalized input, cropped
+    from oracle import oracle_inference
+    from chunkflow_amd.chunk import Chunk
+    sin = Chunk.create(size=(24, 80, 96), dtype='uint8',
+                       pattern='sin').array
+    norm = sin.astype(np.float32) / 127.5 - 1.0
+    ref = oracle_inference(norm, (12, 32, 32), (4, 8, 8),
+                           num_output_channels=3, batch_size=6)
+    ref_crop = ref[:, 2:-2, 4:-4, 4:-4]
+    from scipy import ndimage
+    labels, _ = ndimage.label(
+        ref_crop[0] > 0.0, structure=ndimage.generate_binary_structure(3, 1))
+    assert seg.max() == labels.max()
