@@ -413,6 +413,107 @@ __global__ void k_blend(float* __restrict__ out, int OD, int OH, int OW,
     }
 }
 
+// ---------------------------------------------------------------------------
+// batched blend: many DISJOINT patches in one launch
+// ---------------------------------------------------------------------------
+// The per-patch launch (~34-52 MB) pays a ~1.5-2 us dependent-kernel
+// boundary per patch and leaves each wave a single line-group of work.
+// Overlapping patches must stay ordered (f32 accumulation order = the
+// reference's sequential blend up to f32 commutativity), and patches whose
+// CLIPPED output regions are disjoint share one launch: the host groups
+// patches with a first-fit coloring (chunkflow_amd/grouping.py), so each
+// launch writes every output voxel at most once — no atomics; numerics note
+// in grouping.py (reordering bounded by f32 associativity ulps).
+struct BlendBatchArgs {
+    int n;                  // patches in this launch (<= 32)
+    int d0[32 * 3];         // clipped region start in out
+    int p0[32 * 3];         // clipped region start in patch
+    int r[32 * 3];          // clipped region size
+    int pidx[32];           // batch index into the patch buffer
+    int line_start[33];     // prefix sum of C*rz*ry per patch
+};
+
+template <bool VEC, bool MASKED>
+__global__ void k_blend_batch(float* __restrict__ out, int OD, int OH,
+                              int OW, const float* __restrict__ patch,
+                              int PD, int PH, int PW,
+                              const float* __restrict__ mask, int C,
+                              BlendBatchArgs a, int rxmax) {
+    constexpr int G = 4;
+    long long n_lines = a.line_start[a.n];
+    long long n_groups = (n_lines + G - 1) / G;
+    long long stride = (long long)gridDim.x * blockDim.y;
+    for (long long g = (long long)blockIdx.x * blockDim.y + threadIdx.y;
+         g < n_groups; g += stride) {
+        long long line0 = g * G;
+        int nl = (int)(n_lines - line0 < G ? n_lines - line0 : G);
+        float* o[G];
+        const float* p[G];
+        const float* m[G];
+        int rx[G];
+#pragma unroll
+        for (int j = 0; j < G; ++j) {
+            int line = (int)(line0 + (j < nl ? j : 0));
+            int k = 0;
+            while (line >= a.line_start[k + 1]) ++k;  // n <= 32, wave-uniform
+            int local = line - a.line_start[k];
+            int ry = a.r[k * 3 + 1];
+            int rz = a.r[k * 3 + 0];
+            int y = local % ry;
+            int t = local / ry;
+            int z = t % rz;
+            int c = t / rz;
+            rx[j] = a.r[k * 3 + 2];
+            o[j] = out + (((long long)c * OD + a.d0[k * 3] + z) * OH +
+                          a.d0[k * 3 + 1] + y) * OW + a.d0[k * 3 + 2];
+            p[j] = patch + ((((long long)a.pidx[k] * C + c) * PD +
+                             a.p0[k * 3] + z) * PH +
+                            a.p0[k * 3 + 1] + y) * PW + a.p0[k * 3 + 2];
+            if (MASKED)
+                m[j] = mask + (((long long)(a.p0[k * 3] + z)) * PH +
+                               a.p0[k * 3 + 1] + y) * PW + a.p0[k * 3 + 2];
+        }
+        if (VEC) {
+            for (int x = threadIdx.x; x < rxmax / 4; x += 64) {
+                float4 ov[G], pv[G], mv[G];
+#pragma unroll
+                for (int j = 0; j < G; ++j) {
+                    if (j < nl && x < rx[j] / 4) {
+                        ov[j] = reinterpret_cast<float4*>(o[j])[x];
+                        pv[j] = reinterpret_cast<const float4*>(p[j])[x];
+                        if (MASKED)
+                            mv[j] = reinterpret_cast<const float4*>(m[j])[x];
+                    }
+                }
+#pragma unroll
+                for (int j = 0; j < G; ++j) {
+                    if (j < nl && x < rx[j] / 4) {
+                        if (MASKED) {
+                            ov[j].x += pv[j].x * mv[j].x;
+                            ov[j].y += pv[j].y * mv[j].y;
+                            ov[j].z += pv[j].z * mv[j].z;
+                            ov[j].w += pv[j].w * mv[j].w;
+                        } else {
+                            ov[j].x += pv[j].x;
+                            ov[j].y += pv[j].y;
+                            ov[j].z += pv[j].z;
+                            ov[j].w += pv[j].w;
+                        }
+                        reinterpret_cast<float4*>(o[j])[x] = ov[j];
+                    }
+                }
+            }
+        } else {
+            for (int x = threadIdx.x; x < rxmax; x += 64) {
+#pragma unroll
+                for (int j = 0; j < G; ++j)
+                    if (j < nl && x < rx[j])
+                        o[j][x] += MASKED ? p[j][x] * m[j][x] : p[j][x];
+            }
+        }
+    }
+}
+
 // clip the patch span against the output bounds (Chunk.blend semantics,
 // chunk/base.py:796-807); returns false when the intersection is empty
 static bool clip_region(const int out_dims[3], const int patch_dims[3],
@@ -468,6 +569,64 @@ extern "C" int cfx_blend_accumulate(cfx_ctx* ctx, float* out, int channels,
                                     const float* mask) {
     return blend_one(ctx, out, channels, out_dims, patch, patch_dims,
                      offset_zyx, mask);
+}
+
+// items: n * 4 ints (batch_index, oz, oy, ox). The CALLER guarantees the
+// clipped output regions are pairwise disjoint (first-fit grouping in the
+// host, chunkflow_amd/grouping.py); patches clipped to nothing are skipped.
+extern "C" int cfx_blend_batch(cfx_ctx* ctx, float* out, int channels,
+                               const int out_dims[3], const float* patch,
+                               const int patch_dims[3], const int* items,
+                               int n, const float* mask) {
+    for (int base = 0; base < n; base += 32) {
+        int nb = std::min(32, n - base);
+        BlendBatchArgs a;
+        a.n = 0;
+        a.line_start[0] = 0;
+        bool vec = (out_dims[2] % 4 == 0) && (patch_dims[2] % 4 == 0);
+        int rxmax = 0;
+        double bytes = 0.0;
+        for (int i = 0; i < nb; ++i) {
+            const int* it = items + (base + i) * 4;
+            int off[3] = {it[1], it[2], it[3]};
+            int d0[3], p0[3], r[3];
+            if (!clip_region(out_dims, patch_dims, off, d0, p0, r)) continue;
+            int k = a.n;
+            for (int ax = 0; ax < 3; ++ax) {
+                a.d0[k * 3 + ax] = d0[ax];
+                a.p0[k * 3 + ax] = p0[ax];
+                a.r[k * 3 + ax] = r[ax];
+            }
+            a.pidx[k] = it[0];
+            a.line_start[k + 1] =
+                a.line_start[k] + channels * r[0] * r[1];
+            vec = vec && (d0[2] % 4 == 0) && (p0[2] % 4 == 0) &&
+                  (r[2] % 4 == 0);
+            rxmax = std::max(rxmax, r[2]);
+            double rv = (double)r[0] * r[1] * r[2];
+            bytes += rv * channels * 12.0 + (mask ? rv * 4.0 : 0.0);
+            a.n = k + 1;
+        }
+        if (a.n == 0) continue;
+        long long n_lines = a.line_start[a.n];
+        long long n_groups = (n_lines + 3) / 4;
+        int blocks = (int)std::min<long long>((n_groups + 3) / 4, 8192);
+        hipEvent_t e0;
+        if (prof_begin(ctx, &e0)) return -1;
+#define CFX_LAUNCH_BB(V, M)                                                  \
+    hipLaunchKernelGGL((k_blend_batch<V, M>), dim3(blocks), dim3(64, 4), 0,  \
+                       ctx->stream, out, out_dims[0], out_dims[1],           \
+                       out_dims[2], patch, patch_dims[0], patch_dims[1],     \
+                       patch_dims[2], mask, channels, a, rxmax)
+        if (vec && mask) CFX_LAUNCH_BB(true, true);
+        else if (vec) CFX_LAUNCH_BB(true, false);
+        else if (mask) CFX_LAUNCH_BB(false, true);
+        else CFX_LAUNCH_BB(false, false);
+#undef CFX_LAUNCH_BB
+        CFX_CHECK(hipGetLastError());
+        if (prof_end(ctx, e0, CFX_K_BLEND, bytes)) return -1;
+    }
+    return 0;
 }
 
 // ---------------------------------------------------------------------------

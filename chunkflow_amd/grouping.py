@@ -1,0 +1,59 @@
+"""Disjoint grouping of patch blend targets (first-fit coloring).
+
+Patches whose CLIPPED output regions are disjoint share one kernel launch:
+each output voxel is written at most once per launch, so the accumulate
+stays atomics-free and deterministic. Patch i goes to the lowest-indexed
+group none of whose members overlap it (first-fit), giving ~overlap-depth
+groups (8 for the BASELINE geometries, 27 worst case for a 2x-overlapped
+grid) instead of one launch per patch.
+
+Numerics: within a launch nothing interacts; across launches a pair of
+overlapping patches may accumulate in either order. f32 addition is
+commutative (a+b == b+a bitwise), so only voxels covered by >= 3 patches
+can differ from the reference's sequential order, by f32 associativity
+only — <= a few ulps over the <= 8 overlapping bump-weighted terms, orders
+of magnitude inside the 1e-5/1e-4 parity gates (the exact-order
+alternative degenerates to one group per patch along overlap chains).
+"""
+from typing import List, Tuple
+
+import numpy as np
+
+
+def clip_regions(offsets: np.ndarray, patch_dims, out_dims):
+    """Clipped [lo, hi) boxes of each patch against the output bounds.
+    Returns (lo, hi) int arrays of shape (n, 3); empty regions have
+    hi <= lo on some axis."""
+    offsets = np.asarray(offsets, dtype=np.int64)
+    pd = np.asarray(patch_dims, dtype=np.int64)
+    od = np.asarray(out_dims, dtype=np.int64)
+    lo = np.maximum(offsets, 0)
+    hi = np.minimum(offsets + pd, od)
+    return lo, hi
+
+
+def disjoint_groups(offsets: np.ndarray, patch_dims,
+                    out_dims) -> List[np.ndarray]:
+    """Partition patch indices into disjoint, order-preserving groups.
+
+    offsets: (n, 3) output-local starts. Returns a list of index arrays;
+    concatenated in order they enumerate 0..n-1 with every pair of
+    overlapping patches in strictly increasing groups.
+    """
+    lo, hi = clip_regions(offsets, patch_dims, out_dims)
+    n = lo.shape[0]
+    empty = (hi <= lo).any(axis=1)
+    group_of = np.full(n, -1, dtype=np.int64)
+    for i in range(n):
+        if empty[i]:
+            continue
+        g = 0
+        if i:
+            ov = ((lo[i] < hi[:i]) & (lo[:i] < hi[i])).all(axis=1)
+            ov &= ~empty[:i]
+            taken = set(group_of[:i][ov].tolist())
+            while g in taken:
+                g += 1
+        group_of[i] = g
+    ngroups = int(group_of.max()) + 1 if n else 0
+    return [np.nonzero(group_of == g)[0] for g in range(ngroups)]

@@ -18,7 +18,8 @@ SYMBOLS = [
     'cfx_init', 'cfx_destroy', 'cfx_last_error', 'cfx_version',
     'cfx_set_stream', 'cfx_sync', 'cfx_make_patch_mask',
     'cfx_normalize_intensity', 'cfx_cast_u8_f32_div', 'cfx_extract_patches',
-    'cfx_blend_accumulate', 'cfx_build_chunk_mask', 'cfx_reciprocal',
+    'cfx_blend_accumulate', 'cfx_blend_batch', 'cfx_build_chunk_mask',
+    'cfx_reciprocal',
     'cfx_multiply_mask', 'cfx_max', 'cfx_crop_margin',
     'cfx_mask_using_last_channel', 'cfx_profile_enable', 'cfx_profile_reset',
     'cfx_profile_get',
@@ -147,6 +148,21 @@ class CfxContext:
             _i3(patch_dims), _i3(offset),
             ctypes.c_void_p(mask_ptr) if mask_ptr else None),
             'cfx_blend_accumulate')
+
+    def blend_batch(self, out_ptr, channels, out_dims, patch_ptr,
+                    patch_dims, items, mask_ptr=None):
+        """items: (n, 4) int32 array of (batch_index, oz, oy, ox); clipped
+        regions must be pairwise disjoint."""
+        items = np.ascontiguousarray(items, dtype=np.int32)
+        n = items.shape[0]
+        self._chk(self.lib.cfx_blend_batch(
+            ctypes.c_void_p(self.ctx), ctypes.c_void_p(out_ptr),
+            ctypes.c_int(channels), _i3(out_dims), ctypes.c_void_p(patch_ptr),
+            _i3(patch_dims),
+            items.ctypes.data_as(ctypes.POINTER(ctypes.c_int)),
+            ctypes.c_int(n),
+            ctypes.c_void_p(mask_ptr) if mask_ptr else None),
+            'cfx_blend_batch')
 
     def build_chunk_mask(self, mask_out_ptr, out_dims, patch_mask_ptr,
                          patch_dims, offsets):

@@ -27,6 +27,7 @@ import torch
 from .cartesian import Cartesian, to_cartesian
 from .chunk import Chunk
 from .engines import create_engine
+from .grouping import disjoint_groups
 from .ops import HipOps, TorchOps
 from .patch_mask import make_patch_mask
 from .transform import TransformSequences
@@ -140,6 +141,7 @@ class Inferencer:
             self.engine = None
         self.transform_sequences = TransformSequences() if augment else None
         self._batch_buffers = {}
+        self._group_cache = {}
 
     # ------------------------------------------------------------------
     @property
@@ -184,6 +186,15 @@ class Inferencer:
                 for x in starts[2]:
                     out.append((z, y, x))
         return np.array(out, dtype=np.int32)
+
+    def _groups(self, out_starts: np.ndarray, out_size3):
+        """Disjoint order-preserving launch groups (grouping.py) for a set
+        of output-local patch starts; pure geometry, cached per start set."""
+        key = (out_starts.tobytes(), tuple(out_size3))
+        if key not in self._group_cache:
+            self._group_cache[key] = disjoint_groups(
+                out_starts, tuple(self.output_patch_size), out_size3)
+        return self._group_cache[key]
 
     def _get_batch_buffer(self, n):
         key = n
@@ -251,9 +262,11 @@ class Inferencer:
         patch_mask_t = self.engine.patch_mask
         recip_mask = None
         if self.mask_output_chunk:
-            # rebuilt per input chunk, like the reference (:294-333)
-            recip_mask = self.ops.build_chunk_mask(out_size3, patch_mask_t,
-                                                   out_starts)
+            # mask VALUES rebuilt per input chunk, like the reference
+            # (:294-333); the launch grouping is cached geometry
+            recip_mask = self.ops.build_chunk_mask(
+                out_size3, patch_mask_t, out_starts,
+                groups=self._groups(out_starts, out_size3))
 
         n = in_starts.shape[0]
         fuse_mask = patch_mask_t if not self.engine.pre_masked else None
@@ -270,10 +283,13 @@ class Inferencer:
                 results = self.transform_sequences.backward(results)
                 out_patch = sum(results) / len(results)
                 out_patch = out_patch.contiguous()
-            for b in range(bs):
-                self.ops.blend(output, out_patch, b,
-                               tuple(int(v) for v in out_starts[i + b]),
-                               mask=fuse_mask)
+            bstarts = out_starts[i:i + bs]
+            for idx in self._groups(bstarts, out_size3):
+                items = np.concatenate(
+                    [idx[:, None].astype(np.int32),
+                     bstarts[idx].astype(np.int32)], axis=1)
+                self.ops.blend_batch(output, out_patch, items,
+                                     mask=fuse_mask)
 
         if self.mask_output_chunk:
             self.ops.multiply_mask(output, recip_mask)

@@ -56,13 +56,38 @@ class HipOps:
             out.data_ptr(), channels, out.shape[-3:], patch_ptr, pdims,
             offset, mask.data_ptr() if mask is not None else None)
 
+    def blend_batch(self, out: torch.Tensor, patch_batch: torch.Tensor,
+                    items: np.ndarray, mask: torch.Tensor = None):
+        """items: (n, 4) = (batch_index, oz, oy, ox), regions disjoint."""
+        channels = out.shape[0]
+        self.cfx.blend_batch(
+            out.data_ptr(), channels, out.shape[-3:],
+            patch_batch.data_ptr(), patch_batch.shape[-3:], items,
+            mask.data_ptr() if mask is not None else None)
+
     def build_chunk_mask(self, out_dims, patch_mask: torch.Tensor,
-                         offsets: np.ndarray) -> torch.Tensor:
-        mask = torch.empty(tuple(out_dims), dtype=torch.float32,
+                         offsets: np.ndarray,
+                         groups=None) -> torch.Tensor:
+        """Blend the patch mask at every offset then reciprocal. With
+        `groups` (disjoint index groups over `offsets`), uses one launch
+        per group instead of one per offset."""
+        if groups is None:
+            mask = torch.empty(tuple(out_dims), dtype=torch.float32,
+                               device=patch_mask.device)
+            self.cfx.build_chunk_mask(mask.data_ptr(), out_dims,
+                                      patch_mask.data_ptr(),
+                                      patch_mask.shape[-3:], offsets)
+            return mask
+        mask = torch.zeros(tuple(out_dims), dtype=torch.float32,
                            device=patch_mask.device)
-        self.cfx.build_chunk_mask(mask.data_ptr(), out_dims,
-                                  patch_mask.data_ptr(),
-                                  patch_mask.shape[-3:], offsets)
+        mask3 = mask.unsqueeze(0)  # (1, D, H, W): one "channel"
+        pm = patch_mask.unsqueeze(0)  # batch of one patch
+        for idx in groups:
+            items = np.concatenate(
+                [np.zeros((len(idx), 1), dtype=np.int32),
+                 np.asarray(offsets, dtype=np.int32)[idx]], axis=1)
+            self.blend_batch(mask3, pm, items)
+        self.cfx.reciprocal(mask.data_ptr(), mask.numel())
         return mask
 
     def multiply_mask(self, out: torch.Tensor, mask: torch.Tensor):
@@ -131,7 +156,13 @@ class TorchOps:
         out[..., dst[0], dst[1], dst[2]] += patch[..., src[0], src[1],
                                                   src[2]]
 
-    def build_chunk_mask(self, out_dims, patch_mask, offsets):
+    def blend_batch(self, out, patch_batch, items, mask=None):
+        # CPU plumbing: sequential (order already preserved by grouping)
+        for b, oz, oy, ox in items:
+            self.blend(out, patch_batch, int(b), (int(oz), int(oy), int(ox)),
+                       mask=mask)
+
+    def build_chunk_mask(self, out_dims, patch_mask, offsets, groups=None):
         mask = torch.zeros(tuple(out_dims), dtype=torch.float32)
         pm = patch_mask.unsqueeze(0)  # (1,pz,py,px): one 3-D "patch"
         for off in offsets:

@@ -69,6 +69,15 @@ int cfx_blend_accumulate(cfx_ctx* ctx, float* out, int channels,
                          const int out_dims[3], const float* patch,
                          const int patch_dims[3], const int offset_zyx[3],
                          const float* mask);
+/* many-patch blend in ONE launch: items is a HOST array of n*4 ints
+ * (batch_index, oz, oy, ox). The caller must guarantee the clipped output
+ * regions are pairwise DISJOINT (see the first-fit grouping in
+ * chunkflow_amd/grouping.py) so the accumulate stays atomics-free and
+ * bit-ordered; >32 items are split across launches. */
+int cfx_blend_batch(cfx_ctx* ctx, float* out, int channels,
+                    const int out_dims[3], const float* patch,
+                    const int patch_dims[3], const int* items, int n,
+                    const float* mask);
 /* zero mask_out (out_dims), blend patch_mask at each of n offsets (HOST
  * array, n*3 ints, relative to the output origin), then reciprocal */
 int cfx_build_chunk_mask(cfx_ctx* ctx, float* mask_out, const int out_dims[3],

@@ -247,3 +247,33 @@ def test_nonaligned_vs_oracle_random_geometry():
                            num_output_channels=2, batch_size=7,
                            offset=(1, 2, 3))
     np.testing.assert_allclose(out.numpy().array, ref, rtol=1e-6, atol=1e-7)
+
+
+def test_disjoint_groups_properties():
+    from chunkflow_amd.grouping import disjoint_groups, clip_regions
+    rng = np.random.RandomState(3)
+    out_dims = (30, 40, 50)
+    pd = (8, 12, 16)
+    offsets = np.stack([rng.randint(-4, 28, 60), rng.randint(-4, 36, 60),
+                        rng.randint(-4, 44, 60)], axis=1)
+    groups = disjoint_groups(offsets, pd, out_dims)
+    lo, hi = clip_regions(offsets, pd, out_dims)
+    empty = (hi <= lo).any(axis=1)
+    seen = np.concatenate(groups) if groups else np.array([], dtype=int)
+    assert sorted(seen.tolist()) == [i for i in range(60) if not empty[i]]
+    gi = {}
+    for g, idx in enumerate(groups):
+        for i in idx:
+            gi[int(i)] = g
+    for gidx in groups:
+        for a in range(len(gidx)):          # pairwise disjoint within group
+            for b in range(a + 1, len(gidx)):
+                i, j = gidx[a], gidx[b]
+                assert not ((lo[i] < hi[j]) & (lo[j] < hi[i])).all()
+    # the config-2 grid: 288 patches group into ~the overlap depth
+    from oracle.inference import patch_slices_list
+    starts = np.array([o for _, o in patch_slices_list(
+        (512, 512, 512), (20, 256, 256), (4, 64, 64))], dtype=np.int64)
+    g = disjoint_groups(starts, (20, 256, 256), (512, 512, 512))
+    assert 8 <= len(g) <= 27
+    assert sum(len(x) for x in g) == 288
