@@ -15,6 +15,7 @@
 
 #include <cmath>
 #include <cstdio>
+#include <cstdlib>
 #include <cstring>
 #include <string>
 #include <vector>
@@ -433,13 +434,12 @@ struct BlendBatchArgs {
     int line_start[33];     // prefix sum of C*rz*ry per patch
 };
 
-template <bool VEC, bool MASKED>
+template <int G, bool VEC, bool MASKED, bool NT>
 __global__ void k_blend_batch(float* __restrict__ out, int OD, int OH,
                               int OW, const float* __restrict__ patch,
                               int PD, int PH, int PW,
                               const float* __restrict__ mask, int C,
                               BlendBatchArgs a, int rxmax) {
-    constexpr int G = 4;
     long long n_lines = a.line_start[a.n];
     long long n_groups = (n_lines + G - 1) / G;
     long long stride = (long long)gridDim.x * blockDim.y;
@@ -480,7 +480,16 @@ __global__ void k_blend_batch(float* __restrict__ out, int OD, int OH,
                 for (int j = 0; j < G; ++j) {
                     if (j < nl && x < rx[j] / 4) {
                         ov[j] = reinterpret_cast<float4*>(o[j])[x];
-                        pv[j] = reinterpret_cast<const float4*>(p[j])[x];
+                        if (NT) {
+                            typedef float f4v
+                                __attribute__((ext_vector_type(4)));
+                            f4v t = __builtin_nontemporal_load(
+                                reinterpret_cast<const f4v*>(p[j]) + x);
+                            pv[j] = *reinterpret_cast<float4*>(&t);
+                        } else {
+                            pv[j] =
+                                reinterpret_cast<const float4*>(p[j])[x];
+                        }
                         if (MASKED)
                             mv[j] = reinterpret_cast<const float4*>(m[j])[x];
                     }
@@ -608,20 +617,42 @@ extern "C" int cfx_blend_batch(cfx_ctx* ctx, float* out, int channels,
             a.n = k + 1;
         }
         if (a.n == 0) continue;
+        // tuning knobs (measured via tools/blend_tune.py):
+        //   CFX_BLEND_G  = lines batched per wave iteration (2/4/8)
+        //   CFX_BLEND_NT = nontemporal loads for the read-once patch stream
+        static int env_g = [] {
+            const char* s = getenv("CFX_BLEND_G");
+            int v = s ? atoi(s) : 4;
+            return (v == 2 || v == 4 || v == 8) ? v : 4;
+        }();
+        static bool env_nt = [] {
+            const char* s = getenv("CFX_BLEND_NT");
+            return s && atoi(s) != 0;
+        }();
         long long n_lines = a.line_start[a.n];
-        long long n_groups = (n_lines + 3) / 4;
+        long long n_groups = (n_lines + env_g - 1) / env_g;
         int blocks = (int)std::min<long long>((n_groups + 3) / 4, 8192);
         hipEvent_t e0;
         if (prof_begin(ctx, &e0)) return -1;
-#define CFX_LAUNCH_BB(V, M)                                                  \
-    hipLaunchKernelGGL((k_blend_batch<V, M>), dim3(blocks), dim3(64, 4), 0,  \
-                       ctx->stream, out, out_dims[0], out_dims[1],           \
-                       out_dims[2], patch, patch_dims[0], patch_dims[1],     \
-                       patch_dims[2], mask, channels, a, rxmax)
-        if (vec && mask) CFX_LAUNCH_BB(true, true);
-        else if (vec) CFX_LAUNCH_BB(true, false);
-        else if (mask) CFX_LAUNCH_BB(false, true);
-        else CFX_LAUNCH_BB(false, false);
+#define CFX_LAUNCH_BB(G, V, M, NT)                                           \
+    hipLaunchKernelGGL((k_blend_batch<G, V, M, NT>), dim3(blocks),           \
+                       dim3(64, 4), 0, ctx->stream, out, out_dims[0],        \
+                       out_dims[1], out_dims[2], patch, patch_dims[0],       \
+                       patch_dims[1], patch_dims[2], mask, channels, a,      \
+                       rxmax)
+#define CFX_DISPATCH_G(G)                                                    \
+    do {                                                                     \
+        if (vec && mask && env_nt) CFX_LAUNCH_BB(G, true, true, true);       \
+        else if (vec && mask) CFX_LAUNCH_BB(G, true, true, false);           \
+        else if (vec && env_nt) CFX_LAUNCH_BB(G, true, false, true);         \
+        else if (vec) CFX_LAUNCH_BB(G, true, false, false);                  \
+        else if (mask) CFX_LAUNCH_BB(G, false, true, false);                 \
+        else CFX_LAUNCH_BB(G, false, false, false);                          \
+    } while (0)
+        if (env_g == 2) CFX_DISPATCH_G(2);
+        else if (env_g == 8) CFX_DISPATCH_G(8);
+        else CFX_DISPATCH_G(4);
+#undef CFX_DISPATCH_G
 #undef CFX_LAUNCH_BB
         CFX_CHECK(hipGetLastError());
         if (prof_end(ctx, e0, CFX_K_BLEND, bytes)) return -1;
@@ -670,49 +701,128 @@ extern "C" int cfx_build_chunk_mask(cfx_ctx* ctx, float* mask_out,
 // ---------------------------------------------------------------------------
 // mask-normalize multiply: out[c, i] *= mask[i]
 // ---------------------------------------------------------------------------
-template <bool VEC>
+// order-preserving bit transform lives below (f32_ord); forward-declare
+__host__ __device__ inline unsigned int f32_ord(float f);
+
+// mask-normalize: out[c, i] *= mask[i] for every channel, mask read ONCE
+// (channels in the inner loop — a per-channel grid re-reads the 0.54 GB
+// chunk mask from HBM, it does not fit the 256 MB L3). U=2 float4s per
+// thread per iteration -> 2 mask + 2*C out loads in flight. Optionally
+// fuses the reference's "< 1.0001" sanity scan (inferencer.py:463-466)
+// via a wave-reduced atomic max, saving a full read pass of the output.
+template <bool WITHMAX>
 __global__ void k_maskmul(float* __restrict__ out,
-                          const float* __restrict__ mask, long long n) {
-    int c = blockIdx.y;
+                          const float* __restrict__ mask, long long n4,
+                          int C, long long cstride4,
+                          unsigned int* __restrict__ result) {
+    constexpr int U = 2;
+    long long i0 = ((long long)blockIdx.x * blockDim.x + threadIdx.x) * U;
+    long long stride = (long long)gridDim.x * blockDim.x * U;
+    const float4* m4 = reinterpret_cast<const float4*>(mask);
+    float4* o4 = reinterpret_cast<float4*>(out);
+    float vmax = -INFINITY;
+    for (long long i = i0; i < n4; i += stride) {
+        float4 mv[U];
+        int nu = (int)(n4 - i < U ? n4 - i : U);
+#pragma unroll
+        for (int u = 0; u < U; ++u)
+            if (u < nu) mv[u] = m4[i + u];
+        for (int c = 0; c < C; ++c) {
+            float4 ov[U];
+#pragma unroll
+            for (int u = 0; u < U; ++u)
+                if (u < nu) ov[u] = o4[c * cstride4 + i + u];
+#pragma unroll
+            for (int u = 0; u < U; ++u) {
+                if (u < nu) {
+                    ov[u].x *= mv[u].x;
+                    ov[u].y *= mv[u].y;
+                    ov[u].z *= mv[u].z;
+                    ov[u].w *= mv[u].w;
+                    o4[c * cstride4 + i + u] = ov[u];
+                    if (WITHMAX) {
+                        vmax = fmaxf(vmax, fmaxf(fmaxf(ov[u].x, ov[u].y),
+                                                 fmaxf(ov[u].z, ov[u].w)));
+                    }
+                }
+            }
+        }
+    }
+    if (WITHMAX) {
+        for (int off = 32; off > 0; off >>= 1)
+            vmax = fmaxf(vmax, __shfl_down(vmax, off, 64));
+        if ((threadIdx.x & 63) == 0) atomicMax(result, f32_ord(vmax));
+    }
+}
+
+__global__ void k_maskmul_scalar(float* __restrict__ out,
+                                 const float* __restrict__ mask,
+                                 long long n, int C) {
     long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
     long long stride = gridDim.x * (long long)blockDim.x;
-    if (VEC) {
-        float4* o4 = reinterpret_cast<float4*>(out + c * n * 4);
-        const float4* m4 = reinterpret_cast<const float4*>(mask);
-        for (; i < n; i += stride) {
-            float4 o = o4[i];
-            float4 m = m4[i];
-            o.x *= m.x;
-            o.y *= m.y;
-            o.z *= m.z;
-            o.w *= m.w;
-            o4[i] = o;
-        }
+    for (; i < n; i += stride)
+        for (int c = 0; c < C; ++c) out[c * n + i] *= mask[i];
+}
+
+static int multiply_mask_impl(cfx_ctx* ctx, float* out, const float* mask,
+                              int channels, long long n_voxels,
+                              bool with_max) {
+    hipEvent_t e0;
+    if (prof_begin(ctx, &e0)) return -1;
+    int threads = 256;
+    if (n_voxels % 4 == 0) {
+        long long n4 = n_voxels / 4;
+        long long want = (n4 + threads * 2 - 1) / (threads * 2);
+        int blocks = (int)std::min<long long>(want, 8192);
+        if (with_max)
+            hipLaunchKernelGGL(k_maskmul<true>, dim3(blocks), dim3(threads),
+                               0, ctx->stream, out, mask, n4, channels, n4,
+                               ctx->dev_max);
+        else
+            hipLaunchKernelGGL(k_maskmul<false>, dim3(blocks), dim3(threads),
+                               0, ctx->stream, out, mask, n4, channels, n4,
+                               nullptr);
     } else {
-        float* o = out + c * n;
-        for (; i < n; i += stride) o[i] *= mask[i];
+        long long want = (n_voxels + threads - 1) / threads;
+        int blocks = (int)std::min<long long>(want, 8192);
+        hipLaunchKernelGGL(k_maskmul_scalar, dim3(blocks), dim3(threads), 0,
+                           ctx->stream, out, mask, n_voxels, channels);
+        with_max = false;  // caller falls back to cfx_max
     }
+    CFX_CHECK(hipGetLastError());
+    double bytes = (double)n_voxels * (channels * 8.0 + 4.0);
+    if (prof_end(ctx, e0, CFX_K_MASKMUL, bytes)) return -1;
+    return 0;
 }
 
 extern "C" int cfx_multiply_mask(cfx_ctx* ctx, float* out, const float* mask,
                                  int channels, long long n_voxels) {
-    hipEvent_t e0;
-    if (prof_begin(ctx, &e0)) return -1;
-    int threads = 256;
-    bool vec = (n_voxels % 4 == 0);
-    long long n = vec ? n_voxels / 4 : n_voxels;
-    long long want = (n + threads - 1) / threads;
-    int blocks = (int)std::min<long long>(want, 4096);
-    dim3 grid(blocks, channels);
-    if (vec)
-        hipLaunchKernelGGL(k_maskmul<true>, grid, dim3(threads), 0,
-                           ctx->stream, out, mask, n);
-    else
-        hipLaunchKernelGGL(k_maskmul<false>, grid, dim3(threads), 0,
-                           ctx->stream, out, mask, n);
-    CFX_CHECK(hipGetLastError());
-    double bytes = (double)n_voxels * (channels * 8.0 + 4.0);
-    if (prof_end(ctx, e0, CFX_K_MASKMUL, bytes)) return -1;
+    return multiply_mask_impl(ctx, out, mask, channels, n_voxels, false);
+}
+
+/* fused mask-normalize + max scan; returns -2 when the fused max is
+ * unavailable (unaligned size) and the caller must cfx_max separately */
+extern "C" int cfx_multiply_mask_max(cfx_ctx* ctx, float* out,
+                                     const float* mask, int channels,
+                                     long long n_voxels, float* host_max) {
+    if (n_voxels % 4 != 0) {
+        int rc = multiply_mask_impl(ctx, out, mask, channels, n_voxels,
+                                    false);
+        return rc != 0 ? rc : -2;
+    }
+    unsigned int init = f32_ord(-INFINITY);
+    CFX_CHECK(hipMemcpyAsync(ctx->dev_max, &init, sizeof(init),
+                             hipMemcpyHostToDevice, ctx->stream));
+    int rc = multiply_mask_impl(ctx, out, mask, channels, n_voxels, true);
+    if (rc) return rc;
+    unsigned int outv;
+    CFX_CHECK(hipMemcpyAsync(&outv, ctx->dev_max, sizeof(outv),
+                             hipMemcpyDeviceToHost, ctx->stream));
+    CFX_CHECK(hipStreamSynchronize(ctx->stream));
+    unsigned int u = (outv & 0x80000000u) ? (outv & 0x7fffffffu) : ~outv;
+    float f;
+    __builtin_memcpy(&f, &u, sizeof(f));
+    *host_max = f;
     return 0;
 }
 

@@ -20,7 +20,8 @@ SYMBOLS = [
     'cfx_normalize_intensity', 'cfx_cast_u8_f32_div', 'cfx_extract_patches',
     'cfx_blend_accumulate', 'cfx_blend_batch', 'cfx_build_chunk_mask',
     'cfx_reciprocal',
-    'cfx_multiply_mask', 'cfx_max', 'cfx_crop_margin',
+    'cfx_multiply_mask', 'cfx_multiply_mask_max', 'cfx_max',
+    'cfx_crop_margin',
     'cfx_mask_using_last_channel', 'cfx_profile_enable', 'cfx_profile_reset',
     'cfx_profile_get',
 ]
@@ -184,6 +185,19 @@ class CfxContext:
             ctypes.c_void_p(self.ctx), ctypes.c_void_p(out_ptr),
             ctypes.c_void_p(mask_ptr), ctypes.c_int(channels),
             ctypes.c_longlong(n_voxels)), 'cfx_multiply_mask')
+
+    def multiply_mask_max(self, out_ptr, mask_ptr, channels, n_voxels):
+        """Returns the post-multiply max, or None if the fused path was
+        unavailable (caller should use max())."""
+        out = ctypes.c_float(0)
+        rc = self.lib.cfx_multiply_mask_max(
+            ctypes.c_void_p(self.ctx), ctypes.c_void_p(out_ptr),
+            ctypes.c_void_p(mask_ptr), ctypes.c_int(channels),
+            ctypes.c_longlong(n_voxels), ctypes.byref(out))
+        if rc == -2:
+            return None
+        self._chk(rc, 'cfx_multiply_mask_max')
+        return out.value
 
     def max(self, ptr, n) -> float:
         out = ctypes.c_float(0)

@@ -292,10 +292,10 @@ class Inferencer:
                                      mask=fuse_mask)
 
         if self.mask_output_chunk:
-            self.ops.multiply_mask(output, recip_mask)
-
-        # sanity scan (reference inferencer.py:463-466)
-        maxv = self.ops.max(output)
+            # mask-normalize with the <1.0001 scan fused (:460-466)
+            maxv = self.ops.multiply_mask_max(output, recip_mask)
+        else:
+            maxv = self.ops.max(output)
         if not maxv < 1.0001:
             raise AssertionError(
                 f'output buffer should not be greater than 1: max={maxv}')

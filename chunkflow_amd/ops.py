@@ -94,6 +94,15 @@ class HipOps:
         self.cfx.multiply_mask(out.data_ptr(), mask.data_ptr(),
                                out.shape[0], mask.numel())
 
+    def multiply_mask_max(self, out: torch.Tensor,
+                          mask: torch.Tensor) -> float:
+        """Fused mask-normalize + the <1.0001 max scan."""
+        m = self.cfx.multiply_mask_max(out.data_ptr(), mask.data_ptr(),
+                                       out.shape[0], mask.numel())
+        if m is None:
+            m = self.cfx.max(out.data_ptr(), out.numel())
+        return m
+
     def max(self, t: torch.Tensor) -> float:
         return self.cfx.max(t.data_ptr(), t.numel())
 
@@ -171,6 +180,10 @@ class TorchOps:
 
     def multiply_mask(self, out, mask):
         out *= mask
+
+    def multiply_mask_max(self, out, mask):
+        out *= mask
+        return out.max().item()
 
     def max(self, t):
         return t.max().item()

@@ -89,6 +89,12 @@ int cfx_multiply_mask(cfx_ctx* ctx, float* out, const float* mask,
                       int channels, long long n_voxels);
 /* synchronous max over n floats (the <1.0001 sanity assert) */
 int cfx_max(cfx_ctx* ctx, const float* buf, long long n, float* host_max);
+/* fused mask-normalize + max scan (saves a full output read pass);
+ * returns -2 when n_voxels is not float4-aligned — multiply done, call
+ * cfx_max yourself */
+int cfx_multiply_mask_max(cfx_ctx* ctx, float* out, const float* mask,
+                          int channels, long long n_voxels,
+                          float* host_max);
 /* contiguous copy dropping margins[6] = -z,-y,-x,+z,+y,+x */
 int cfx_crop_margin(cfx_ctx* ctx, const float* in, float* out, int channels,
                     const int in_dims[3], const int margins[6]);
