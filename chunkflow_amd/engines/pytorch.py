@@ -15,6 +15,8 @@ MI355X-first differences from the reference engine:
     not numpy arrays.
 The patch-mask multiply (pytorch.py:113) is fused into the blend kernel.
 """
+import os
+
 import torch
 
 from .base import EngineBase
@@ -45,7 +47,6 @@ class PyTorchEngine(EngineBase):
         # 78.6 vs 112 ms per 12-patch RSUNet batch on MI355X); disable with
         # CFX_CHANNELS_LAST=0. benchmark=True matches the reference
         # (pytorch.py:7) and lets MIOpen tune per shape.
-        import os
         self.channels_last = (str(self.device).startswith('cuda') and
                               os.environ.get('CFX_CHANNELS_LAST', '1') != '0')
         if self.channels_last:

@@ -384,3 +384,19 @@ def test_cli_gpu_config4_chain(tmp_path):
     labels, _ = ndimage.label(
         ref_crop[0] > 0.0, structure=ndimage.generate_binary_structure(3, 1))
     assert seg.max() == labels.max()
+
+
+def test_bf16_engine_close_to_f32(golden, golden_dir):
+    """--dtype bfloat16 conv path (config-5 wiring): bf16 MFMA forward +
+    f32 accumulation stays within bf16 tolerance of the f32 golden."""
+    from chunkflow_amd.chunk import Chunk
+    _, arrays = golden
+    inf = _hip_inferencer(
+        model=os.path.join(golden_dir, 'ref_model.py'),
+        weights=os.path.join(golden_dir, 'ref_model_weights.pt'),
+        framework='pytorch', num_output_channels=3, batch_size=1,
+        dtype='bfloat16', mask_output_chunk=True)
+    out = inf(Chunk(arrays['e2e_input_u8'].copy()))
+    np.testing.assert_allclose(out.numpy().array,
+                               arrays['e2e_pytorch_out'],
+                               rtol=0.05, atol=0.02)
