@@ -400,3 +400,38 @@ def test_bf16_engine_close_to_f32(golden, golden_dir):
     np.testing.assert_allclose(out.numpy().array,
                                arrays['e2e_pytorch_out'],
                                rtol=0.05, atol=0.02)
+
+
+def test_scalar_paths_non_x4_geometry():
+    """x dims not divisible by 4 exercise every kernel's scalar fallback
+    (blend_batch, extract, crop, maskmul via cfx_max path)."""
+    from chunkflow_amd.chunk import Chunk
+    from oracle import oracle_inference
+    rng = np.random.RandomState(21)
+    arr = rng.randint(0, 256, size=(17, 45, 51), dtype=np.uint8)
+    inf = _hip_inferencer(patch_size=(9, 21, 27), overlap=(3, 7, 9),
+                          framework='identity', num_output_channels=2,
+                          batch_size=5, mask_output_chunk=True)
+    out = inf(Chunk(arr.copy()))
+    ref = oracle_inference(arr, (9, 21, 27), (3, 7, 9),
+                           num_output_channels=2, batch_size=5)
+    np.testing.assert_allclose(out.numpy().array, ref, rtol=1e-5, atol=1e-6)
+
+
+def test_f16_masked_mode_no_crash():
+    """--dtype float16 + --mask-output-chunk CRASHES the reference (f16
+    reciprocal overflow -> NaN -> assert). This build
+    accumulates in f32 and only casts the output — documented divergence."""
+    from chunkflow_amd.chunk import Chunk
+    chunk = Chunk.create(size=(28, 100, 100), dtype='uint8', pattern='sin')
+    inf = _hip_inferencer(patch_size=(20, 64, 64), overlap=(4, 16, 16),
+                          framework='identity', num_output_channels=1,
+                          batch_size=4, dtype='float16',
+                          mask_output_chunk=True)
+    out = inf(chunk)
+    got = out.numpy().array
+    assert got.dtype == np.float16
+    assert np.isfinite(got).all()
+    expect = chunk.array.astype(np.float32) / 255.0
+    np.testing.assert_allclose(got[0].astype(np.float32), expect,
+                               rtol=2e-3, atol=2.0 / 255.0)
