@@ -5,6 +5,8 @@ import subprocess
 PKG_DIR = os.path.dirname(os.path.abspath(__file__))
 SO_PATH = os.path.join(PKG_DIR, 'libchunkflow_amd.so')
 SRC = os.path.join(PKG_DIR, 'csrc', 'cfx.hip')
+SRC_CC = os.path.join(PKG_DIR, 'csrc', 'cc.hip')
+SRC_INT = os.path.join(PKG_DIR, 'csrc', 'cfx_internal.h')
 HEADER = os.path.join(PKG_DIR, '..', 'include', 'chunkflow_amd.h')
 
 
@@ -12,7 +14,8 @@ def so_is_fresh() -> bool:
     if not os.path.exists(SO_PATH):
         return False
     so_mtime = os.path.getmtime(SO_PATH)
-    return all(os.path.getmtime(p) <= so_mtime for p in (SRC, HEADER))
+    return all(os.path.getmtime(p) <= so_mtime
+               for p in (SRC, SRC_CC, SRC_INT, HEADER))
 
 
 def build(force: bool = False) -> str:
@@ -25,7 +28,7 @@ def build(force: bool = False) -> str:
         return SO_PATH
     cmd = [
         'hipcc', '--offload-arch=gfx950', '-O3', '-std=c++17',
-        '-ffp-contract=off', '-fPIC', '-shared', SRC, '-o', SO_PATH,
+        '-ffp-contract=off', '-fPIC', '-shared', SRC, SRC_CC, '-o', SO_PATH,
     ]
     subprocess.run(cmd, check=True)
     return SO_PATH

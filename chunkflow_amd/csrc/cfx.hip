@@ -20,47 +20,21 @@
 #include <string>
 #include <vector>
 
-#include "../../include/chunkflow_amd.h"
+#include "cfx_internal.h"
 
 #define CFX_VERSION 1
 
 // ---------------------------------------------------------------------------
 // error handling
 // ---------------------------------------------------------------------------
-static thread_local std::string g_err;
+thread_local std::string g_err;
 
 extern "C" const char* cfx_last_error(void) { return g_err.c_str(); }
 extern "C" int cfx_version(void) { return CFX_VERSION; }
 
-#define CFX_CHECK(expr)                                                      \
-    do {                                                                     \
-        hipError_t _e = (expr);                                              \
-        if (_e != hipSuccess) {                                              \
-            g_err = std::string(#expr) + ": " + hipGetErrorString(_e);       \
-            return -1;                                                       \
-        }                                                                    \
-    } while (0)
-
 // ---------------------------------------------------------------------------
-// context
+// context (struct + CFX_CHECK in cfx_internal.h, shared with cc.hip)
 // ---------------------------------------------------------------------------
-struct ProfEntry {
-    hipEvent_t e0, e1;
-    int kid;
-    double bytes;
-};
-
-struct cfx_ctx {
-    int device = 0;
-    hipStream_t stream = nullptr;  // legacy default stream unless adopted
-    bool profile = false;
-    std::vector<ProfEntry> pending;
-    unsigned long long prof_count[CFX_K_COUNT] = {};
-    double prof_ms[CFX_K_COUNT] = {};
-    double prof_bytes[CFX_K_COUNT] = {};
-    unsigned int* dev_max = nullptr;  // scratch for cfx_max
-};
-
 extern "C" cfx_ctx* cfx_init(int device) {
     hipError_t e = hipSetDevice(device);
     if (e != hipSuccess) {
@@ -84,6 +58,7 @@ extern "C" void cfx_destroy(cfx_ctx* ctx) {
         hipEventDestroy(p.e1);
     }
     if (ctx->dev_max) hipFree(ctx->dev_max);
+    if (ctx->cc_counts) hipFree(ctx->cc_counts);
     delete ctx;
 }
 
@@ -98,14 +73,14 @@ extern "C" int cfx_sync(cfx_ctx* ctx) {
 }
 
 // profiling helpers ---------------------------------------------------------
-static int prof_begin(cfx_ctx* ctx, hipEvent_t* e0) {
+int prof_begin(cfx_ctx* ctx, hipEvent_t* e0) {
     if (!ctx->profile) return 0;
     CFX_CHECK(hipEventCreate(e0));
     CFX_CHECK(hipEventRecord(*e0, ctx->stream));
     return 0;
 }
 
-static int prof_end(cfx_ctx* ctx, hipEvent_t e0, int kid, double bytes) {
+int prof_end(cfx_ctx* ctx, hipEvent_t e0, int kid, double bytes) {
     if (!ctx->profile) return 0;
     hipEvent_t e1;
     CFX_CHECK(hipEventCreate(&e1));

@@ -22,7 +22,8 @@ SYMBOLS = [
     'cfx_reciprocal',
     'cfx_multiply_mask', 'cfx_multiply_mask_max', 'cfx_max',
     'cfx_crop_margin',
-    'cfx_mask_using_last_channel', 'cfx_profile_enable', 'cfx_profile_reset',
+    'cfx_mask_using_last_channel', 'cfx_threshold', 'cfx_nonzero_u8',
+    'cfx_connected_components', 'cfx_profile_enable', 'cfx_profile_reset',
     'cfx_profile_get',
 ]
 
@@ -218,6 +219,29 @@ class CfxContext:
             ctypes.c_void_p(self.ctx), ctypes.c_void_p(in_ptr),
             ctypes.c_void_p(out_ptr), ctypes.c_int(channels), _i3(dims),
             ctypes.c_float(threshold)), 'cfx_mask_using_last_channel')
+
+    # --- connected components ----------------------------------------------
+    def threshold(self, in_ptr, fg_ptr, n, thresh):
+        self._chk(self.lib.cfx_threshold(
+            ctypes.c_void_p(self.ctx), ctypes.c_void_p(in_ptr),
+            ctypes.c_void_p(fg_ptr), ctypes.c_longlong(n),
+            ctypes.c_float(thresh)), 'cfx_threshold')
+
+    def nonzero_u8(self, in_ptr, fg_ptr, n):
+        self._chk(self.lib.cfx_nonzero_u8(
+            ctypes.c_void_p(self.ctx), ctypes.c_void_p(in_ptr),
+            ctypes.c_void_p(fg_ptr), ctypes.c_longlong(n)),
+            'cfx_nonzero_u8')
+
+    def connected_components(self, fg_ptr, dims, connectivity, labels_ptr,
+                             scratch_ptr) -> int:
+        ncomp = ctypes.c_longlong(0)
+        self._chk(self.lib.cfx_connected_components(
+            ctypes.c_void_p(self.ctx), ctypes.c_void_p(fg_ptr), _i3(dims),
+            ctypes.c_int(connectivity), ctypes.c_void_p(labels_ptr),
+            ctypes.c_void_p(scratch_ptr), ctypes.byref(ncomp)),
+            'cfx_connected_components')
+        return ncomp.value
 
     # --- profiling ----------------------------------------------------------
     def profile_enable(self, enable=True):

@@ -103,6 +103,23 @@ int cfx_mask_using_last_channel(cfx_ctx* ctx, const float* in, float* out,
                                 int channels, const int dims[3],
                                 float threshold);
 
+/* ---- connected components (cc3d replacement; config-4 chain) ----------- */
+/* fg[i] = in[i] > threshold */
+int cfx_threshold(cfx_ctx* ctx, const float* in, unsigned char* fg,
+                  long long n, float threshold);
+/* fg[i] = in[i] != 0 */
+int cfx_nonzero_u8(cfx_ctx* ctx, const unsigned char* in, unsigned char* fg,
+                   long long n);
+/* 6/18/26-connectivity union-find labeling of a u8 foreground mask.
+ * labels (u32, caller-owned) doubles as the union-find parent array;
+ * scratch is a second u32 buffer of the same length. Labels are 1..N in
+ * raster-scan first-encounter order (scipy.ndimage.label-compatible);
+ * background is 0. */
+int cfx_connected_components(cfx_ctx* ctx, const unsigned char* fg,
+                             const int dims[3], int connectivity,
+                             unsigned int* labels, unsigned int* scratch,
+                             long long* n_components);
+
 /* ---- kernel timing (HIP events on the context stream) ------------------ */
 enum cfx_kernel_id {
     CFX_K_BLEND = 0,
@@ -114,7 +131,8 @@ enum cfx_kernel_id {
     CFX_K_CROP = 6,
     CFX_K_MAX = 7,
     CFX_K_MYELIN = 8,
-    CFX_K_COUNT = 9
+    CFX_K_CC = 9,
+    CFX_K_COUNT = 10
 };
 int cfx_profile_enable(cfx_ctx* ctx, int enable);
 int cfx_profile_reset(cfx_ctx* ctx);

@@ -367,7 +367,7 @@ def test_cli_gpu_config4_chain(tmp_path):
     assert r.exit_code == 0, r.output
     seg = np.load(out)
     assert seg.shape == (20, 72, 88)
-    assert seg.dtype == np.uint32
+    assert seg.dtype in (np.uint32, np.int32)
     assert seg.max() >= 1
 
     # the same chain against the oracle: normalize-intensity then identity
@@ -435,3 +435,49 @@ def test_f16_masked_mode_no_crash():
     expect = chunk.array.astype(np.float32) / 255.0
     np.testing.assert_allclose(got[0].astype(np.float32), expect,
                                rtol=2e-3, atol=2.0 / 255.0)
+
+
+class TestConnectedComponentsGPU:
+    """GPU union-find vs scipy.ndimage.label: identical labels (same
+    partition AND same first-encounter numbering)."""
+
+    def _check(self, vol, threshold, connectivity):
+        from scipy import ndimage
+        from chunkflow_amd.chunk import Chunk
+        from chunkflow_amd.connected import connected_component, _STRUCTS
+        dev = Chunk(torch.from_numpy(vol.copy()).cuda())
+        got = connected_component(dev, threshold=threshold,
+                                  connectivity=connectivity)
+        got_np = got.numpy().array
+        seg = vol > threshold if threshold is not None else vol != 0
+        ref, nref = ndimage.label(seg, structure=_STRUCTS[connectivity])
+        np.testing.assert_array_equal(got_np.astype(np.int64), ref)
+        assert int(got_np.max()) == nref
+
+    def test_random_6(self):
+        rng = np.random.RandomState(30)
+        vol = rng.rand(40, 50, 60).astype(np.float32)
+        self._check(vol, 0.7, 6)
+
+    def test_random_18_26(self):
+        rng = np.random.RandomState(31)
+        vol = rng.rand(30, 40, 50).astype(np.float32)
+        self._check(vol, 0.6, 18)
+        self._check(vol, 0.6, 26)
+
+    def test_dense_blobs(self):
+        # low threshold -> few huge components spanning the volume
+        from chunkflow_amd.chunk import Chunk as C
+        vol = C.create(size=(48, 64, 64), dtype='uint8',
+                       pattern='sin').array.astype(np.float32) / 255.0
+        self._check(vol, 0.5, 6)
+        self._check(vol, 0.5, 26)
+
+    def test_nonzero_u8_path(self):
+        rng = np.random.RandomState(32)
+        vol = (rng.rand(20, 30, 40) > 0.6).astype(np.uint8)
+        self._check(vol, None, 6)
+
+    def test_all_foreground_and_empty(self):
+        self._check(np.ones((8, 8, 8), dtype=np.float32), 0.5, 6)
+        self._check(np.zeros((8, 8, 8), dtype=np.float32), 0.5, 6)
