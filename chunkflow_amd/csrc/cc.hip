@@ -34,14 +34,29 @@ __device__ __constant__ int BWD[13][3] = {
     {-1, -1, -1}, {-1, -1, 1}, {-1, 1, -1}, {-1, 1, 1},    // 26 (corners)
 };
 
+// Every parent access is an agent-scope relaxed atomic (lowered to an
+// sc1 load/store that bypasses the per-CU L1): a mid-chain node j can only
+// ever READ as "p[j]==j" from a stale init-era L1 line — plain loads
+// produced exactly that (~0.7% of voxels compressed to a non-root on real
+// volumes; MI355X_MICROARCH.md §inter-workgroup visibility).
+__device__ inline unsigned int cc_ld(const unsigned int* p, unsigned int x) {
+    return __hip_atomic_load(&p[x], __ATOMIC_RELAXED,
+                             __HIP_MEMORY_SCOPE_AGENT);
+}
+
+__device__ inline void cc_st(unsigned int* p, unsigned int x,
+                             unsigned int v) {
+    __hip_atomic_store(&p[x], v, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+}
+
 __device__ inline unsigned int cc_find(unsigned int* __restrict__ p,
                                        unsigned int x) {
-    unsigned int px = p[x];
+    unsigned int px = cc_ld(p, x);
     while (px != x) {
-        unsigned int ppx = p[px];
-        p[x] = ppx;  // path halving (benign race)
+        unsigned int ppx = cc_ld(p, px);
+        cc_st(p, x, ppx);  // path halving (ancestor store, benign race)
         x = ppx;
-        px = p[x];
+        px = cc_ld(p, x);
     }
     return x;
 }
@@ -66,7 +81,8 @@ __global__ void k_cc_init(const unsigned char* __restrict__ fg,
     long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
     long long stride = gridDim.x * (long long)blockDim.x;
     for (; i < n; i += stride)
-        parent[i] = fg[i] ? (unsigned int)i : 0xFFFFFFFFu;
+        cc_st(parent, (unsigned int)i,
+              fg[i] ? (unsigned int)i : 0xFFFFFFFFu);
 }
 
 __global__ void k_cc_merge(const unsigned char* __restrict__ fg,
@@ -98,7 +114,9 @@ __global__ void k_cc_compress(const unsigned char* __restrict__ fg,
     long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
     long long stride = gridDim.x * (long long)blockDim.x;
     for (; i < n; i += stride)
-        if (fg[i]) parent[i] = cc_find(parent, (unsigned int)i);
+        if (fg[i])
+            cc_st(parent, (unsigned int)i,
+                  cc_find(parent, (unsigned int)i));
 }
 
 // one 64-lane wave per SCAN_CHUNK: count roots (parent[i] == i)
@@ -110,10 +128,12 @@ __global__ void k_cc_count(const unsigned int* __restrict__ parent,
     long long c1 = c0 + SCAN_CHUNK < n ? c0 + SCAN_CHUNK : n;
     unsigned int cnt = 0;
     for (long long i = c0 + threadIdx.x; i < c1; i += 64)
-        if (parent[i] == (unsigned int)i) ++cnt;
+        if (cc_ld(parent, (unsigned int)i) == (unsigned int)i) ++cnt;
     for (int off = 32; off > 0; off >>= 1)
         cnt += __shfl_down(cnt, off, 64);
-    if (threadIdx.x == 0) counts[blockIdx.x] = cnt;
+    if (threadIdx.x == 0)
+        __hip_atomic_store(&counts[blockIdx.x], cnt, __ATOMIC_RELAXED,
+                           __HIP_MEMORY_SCOPE_AGENT);
 }
 
 // assign newlabel[root] = chunk_offset + local_rank + 1 via wave ballots
@@ -124,15 +144,19 @@ __global__ void k_cc_rank(const unsigned int* __restrict__ parent,
     long long c0 = (long long)blockIdx.x * SCAN_CHUNK;
     if (c0 >= n) return;
     long long c1 = c0 + SCAN_CHUNK < n ? c0 + SCAN_CHUNK : n;
-    unsigned int base = offsets[blockIdx.x];
+    unsigned int base = __hip_atomic_load(
+        &offsets[blockIdx.x], __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
     for (long long i0 = c0; i0 < c1; i0 += 64) {
         long long i = i0 + threadIdx.x;
-        bool is_root = i < c1 && parent[i] == (unsigned int)i;
+        bool is_root =
+            i < c1 &&
+            cc_ld(parent, (unsigned int)i) == (unsigned int)i;
         unsigned long long ballot = __ballot(is_root);
         if (is_root) {
             unsigned int before = (unsigned int)__popcll(
                 ballot & ((1ull << threadIdx.x) - 1ull));
-            newlabel[i] = base + before + 1;
+            __hip_atomic_store(&newlabel[i], base + before + 1,
+                               __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
         }
         base += (unsigned int)__popcll(ballot);
     }
@@ -145,7 +169,8 @@ __global__ void k_cc_final(const unsigned char* __restrict__ fg,
     long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
     long long stride = gridDim.x * (long long)blockDim.x;
     for (; i < n; i += stride)
-        labels[i] = fg[i] ? newlabel[parent[i]] : 0u;
+        labels[i] =
+            fg[i] ? cc_ld(newlabel, cc_ld(parent, (unsigned int)i)) : 0u;
 }
 
 __global__ void k_threshold(const float* __restrict__ in,
