@@ -49,6 +49,21 @@ __device__ inline void cc_st(unsigned int* p, unsigned int x,
     __hip_atomic_store(&p[x], v, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
 }
 
+// read-only find: used where a halving store could overwrite another
+// thread's already-compressed root (k_cc_compress: thread T2 shortcutting
+// THROUGH node i with a stale-read mid-chain ancestor would clobber T1's
+// final parent[i]=root store — observed as ~1e-4 of voxels labeled from
+// unwritten scratch)
+__device__ inline unsigned int cc_find_ro(const unsigned int* p,
+                                          unsigned int x) {
+    unsigned int px = cc_ld(p, x);
+    while (px != x) {
+        x = px;
+        px = cc_ld(p, x);
+    }
+    return x;
+}
+
 __device__ inline unsigned int cc_find(unsigned int* __restrict__ p,
                                        unsigned int x) {
     unsigned int px = cc_ld(p, x);
@@ -116,7 +131,7 @@ __global__ void k_cc_compress(const unsigned char* __restrict__ fg,
     for (; i < n; i += stride)
         if (fg[i])
             cc_st(parent, (unsigned int)i,
-                  cc_find(parent, (unsigned int)i));
+                  cc_find_ro(parent, (unsigned int)i));
 }
 
 // one 64-lane wave per SCAN_CHUNK: count roots (parent[i] == i)

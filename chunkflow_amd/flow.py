@@ -153,6 +153,46 @@ def normalize_intensity(tasks, name, input_chunk_name, output_chunk_name):
         yield task
 
 
+@main.command('normalize-contrast')
+@click.option('--name', type=str, default='normalize-contrast-nkem',
+              help='name of operator.')
+@click.option('--input-chunk-name', '-i', type=str,
+              default=DEFAULT_CHUNK_NAME, help='input chunk name')
+@click.option('--output-chunk-name', '-o', type=str,
+              default=DEFAULT_CHUNK_NAME, help='output chunk name')
+@click.option('--lower-clip-fraction', '-l', type=click.FLOAT, default=0.01,
+              help='lower intensity fraction to clip out.')
+@click.option('--upper-clip-fraction', '-u', type=click.FLOAT, default=0.01,
+              help='upper intensity fraction to clip out.')
+@click.option('--minval', type=click.INT, default=1,
+              help='the minimum intensity of transformed chunk.')
+@click.option('--maxval', type=click.INT, default=255,
+              help='the maximum intensity of transformed chunk.')
+@click.option('--per-section/--whole', default=True,
+              help='per section normalization or normalize the whole chunk.')
+@operator
+def normalize_contrast(tasks, name, input_chunk_name, output_chunk_name,
+                       lower_clip_fraction, upper_clip_fraction, minval,
+                       maxval, per_section):
+    """Normalize the section contrast using precomputed histograms."""
+    import torch
+    from .contrast import normalize_contrast as _nc
+    for task in tasks:
+        if task is not None:
+            start = time()
+            chunk = task[input_chunk_name]
+            if torch.cuda.is_available():
+                chunk = chunk.to_device().clone()
+            else:
+                chunk = chunk.clone()
+            task[output_chunk_name] = _nc(
+                chunk, lower_clip_fraction=lower_clip_fraction,
+                upper_clip_fraction=upper_clip_fraction, minval=minval,
+                maxval=maxval, per_section=per_section)
+            task['log']['timer'][name] = time() - start
+        yield task
+
+
 @main.command('inference')
 @click.option('--name', type=str, default='inference',
               help='name of this operator')

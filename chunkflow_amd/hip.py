@@ -23,7 +23,8 @@ SYMBOLS = [
     'cfx_multiply_mask', 'cfx_multiply_mask_max', 'cfx_max',
     'cfx_crop_margin',
     'cfx_mask_using_last_channel', 'cfx_threshold', 'cfx_nonzero_u8',
-    'cfx_connected_components', 'cfx_profile_enable', 'cfx_profile_reset',
+    'cfx_connected_components', 'cfx_hist_u8', 'cfx_lut_apply_u8',
+    'cfx_profile_enable', 'cfx_profile_reset',
     'cfx_profile_get',
 ]
 
@@ -242,6 +243,19 @@ class CfxContext:
             ctypes.c_void_p(scratch_ptr), ctypes.byref(ncomp)),
             'cfx_connected_components')
         return ncomp.value
+
+    # --- image normalization -------------------------------------------------
+    def hist_u8(self, in_ptr, n_per_sec, nsec, hist_ptr):
+        self._chk(self.lib.cfx_hist_u8(
+            ctypes.c_void_p(self.ctx), ctypes.c_void_p(in_ptr),
+            ctypes.c_longlong(n_per_sec), ctypes.c_int(nsec),
+            ctypes.c_void_p(hist_ptr)), 'cfx_hist_u8')
+
+    def lut_apply_u8(self, buf_ptr, n_per_sec, nsec, lut_ptr):
+        self._chk(self.lib.cfx_lut_apply_u8(
+            ctypes.c_void_p(self.ctx), ctypes.c_void_p(buf_ptr),
+            ctypes.c_longlong(n_per_sec), ctypes.c_int(nsec),
+            ctypes.c_void_p(lut_ptr)), 'cfx_lut_apply_u8')
 
     # --- profiling ----------------------------------------------------------
     def profile_enable(self, enable=True):

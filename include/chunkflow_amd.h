@@ -120,6 +120,15 @@ int cfx_connected_components(cfx_ctx* ctx, const unsigned char* fg,
                              unsigned int* labels, unsigned int* scratch,
                              long long* n_components);
 
+/* ---- image normalization (normalize-contrast operator) ----------------- */
+/* zero + accumulate nsec 256-bin u32 histograms, one per contiguous
+ * n_per_sec-voxel section of a u8 volume */
+int cfx_hist_u8(cfx_ctx* ctx, const unsigned char* in, long long n_per_sec,
+                int nsec, unsigned int* hist);
+/* in-place per-section LUT gather: buf[i] = lut[sec][buf[i]] */
+int cfx_lut_apply_u8(cfx_ctx* ctx, unsigned char* buf, long long n_per_sec,
+                     int nsec, const unsigned char* lut);
+
 /* ---- kernel timing (HIP events on the context stream) ------------------ */
 enum cfx_kernel_id {
     CFX_K_BLEND = 0,

@@ -277,3 +277,18 @@ def test_disjoint_groups_properties():
     g = disjoint_groups(starts, (20, 256, 256), (512, 512, 512))
     assert 8 <= len(g) <= 27
     assert sum(len(x) for x in g) == 288
+
+
+def test_normalize_contrast_product_vs_oracle():
+    from chunkflow_amd.contrast import normalize_contrast
+    from oracle.contrast import oracle_normalize_contrast
+    rng = np.random.RandomState(9)
+    for arr in ((rng.rand(6, 40, 50) * 256).astype(np.uint8),
+                (rng.rand(4, 30, 30) * 80 + 100).astype(np.uint8),
+                np.zeros((3, 10, 10), dtype=np.uint8)):
+        ref = oracle_normalize_contrast(arr.copy())
+        got = normalize_contrast(Chunk(arr.copy()))
+        np.testing.assert_array_equal(got.numpy().array, ref)
+        # --whole is a no-op, like the reference
+        same = normalize_contrast(Chunk(arr.copy()), per_section=False)
+        np.testing.assert_array_equal(same.numpy().array, arr)
