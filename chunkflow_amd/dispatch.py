@@ -109,3 +109,12 @@ def stitch_to_rank0(bboxes: List[BoundingBox], local_outputs: dict,
         for r in reqs:
             r.wait()
         return None
+
+
+def all_reduce_hook(group=None):
+    """pre_normalize_hook for intra-chunk sharding: SUM the per-rank
+    partial blends (one RCCL all-reduce per chunk — the only data-path
+    collective the sharded path needs)."""
+    def hook(output_tensor):
+        dist.all_reduce(output_tensor, op=dist.ReduceOp.SUM, group=group)
+    return hook

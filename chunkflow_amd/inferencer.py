@@ -53,7 +53,16 @@ class Inferencer:
                  mask_myelin_threshold=None,
                  augment: bool = False,
                  dry_run: bool = False,
-                 compute_device: str = None):
+                 compute_device: str = None,
+                 patch_shard: tuple = None,
+                 pre_normalize_hook=None):
+        """patch_shard=(rank, world): intra-chunk multi-GPU (SURVEY.md §8f
+        row 4) — this rank blends only patches with index % world == rank
+        into its own zero buffer; pre_normalize_hook(output_tensor) runs
+        before the mask normalize (dispatch.py installs an all-reduce SUM
+        there, which reconstitutes the full blend: the blend is a sum over
+        patches, so sharding + summing is the same accumulation up to f32
+        reordering)."""
         assert input_size is None or patch_num is None
         input_patch_size = to_cartesian(input_patch_size)
         output_patch_size = to_cartesian(output_patch_size)
@@ -142,6 +151,8 @@ class Inferencer:
         self.transform_sequences = TransformSequences() if augment else None
         self._batch_buffers = {}
         self._group_cache = {}
+        self.patch_shard = patch_shard
+        self.pre_normalize_hook = pre_normalize_hook
 
     # ------------------------------------------------------------------
     @property
@@ -268,6 +279,11 @@ class Inferencer:
                 out_size3, patch_mask_t, out_starts,
                 groups=self._groups(out_starts, out_size3))
 
+        if self.patch_shard is not None:
+            shard_rank, shard_world = self.patch_shard
+            keep = np.arange(in_starts.shape[0]) % shard_world == shard_rank
+            in_starts = in_starts[keep]
+            out_starts = out_starts[keep]
         n = in_starts.shape[0]
         fuse_mask = patch_mask_t if not self.engine.pre_masked else None
         for i in range(0, n, self.batch_size):
@@ -290,6 +306,9 @@ class Inferencer:
                      bstarts[idx].astype(np.int32)], axis=1)
                 self.ops.blend_batch(output, out_patch, items,
                                      mask=fuse_mask)
+
+        if self.pre_normalize_hook is not None:
+            self.pre_normalize_hook(output)
 
         if self.mask_output_chunk:
             # mask-normalize with the <1.0001 scan fused (:460-466)

@@ -60,3 +60,45 @@ def test_shard_tasks_partition():
     parts = [shard_tasks(tasks, r, 4) for r in range(4)]
     assert sorted(sum(parts, [])) == tasks
     assert parts[0] == [0, 4, 8]
+
+
+def _intra_worker(rank, world, port, result_path):
+    os.environ['MASTER_ADDR'] = '127.0.0.1'
+    os.environ['MASTER_PORT'] = str(port)
+    os.environ['RANK'] = str(rank)
+    os.environ['WORLD_SIZE'] = str(world)
+    import numpy as _np
+    import torch.distributed as dist
+    from chunkflow_amd.chunk import Chunk
+    from chunkflow_amd.dispatch import init_distributed, all_reduce_hook
+    from chunkflow_amd.inferencer import Inferencer
+    init_distributed(backend='gloo')
+    arr = _np.load(result_path + '.in.npy')
+    inf = Inferencer(None, None, (10, 32, 32),
+                     output_patch_overlap=(2, 8, 8), framework='identity',
+                     num_output_channels=2, batch_size=3,
+                     mask_output_chunk=True,
+                     patch_shard=(rank, world),
+                     pre_normalize_hook=all_reduce_hook())
+    out = inf(Chunk(arr))
+    if rank == 0:
+        _np.save(result_path, out.numpy().array)
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(180)
+def test_intra_chunk_shard_world2(tmp_path):
+    """Intra-chunk patch-grid sharding: 2 ranks
+    share one chunk's patches; the all-reduced result equals the
+    single-rank oracle at 1e-5."""
+    from oracle import oracle_inference
+    rng = np.random.RandomState(17)
+    arr = rng.randint(0, 256, size=(20, 68, 72), dtype=np.uint8)
+    result = str(tmp_path / 'intra.npy')
+    np.save(result + '.in.npy', arr)
+    mp.spawn(_intra_worker, args=(2, 29533, result), nprocs=2, join=True)
+    got = np.load(result)
+    ref = oracle_inference(arr, (10, 32, 32), (2, 8, 8),
+                           num_output_channels=2, batch_size=3)
+    np.testing.assert_allclose(got, ref, rtol=1e-5, atol=1e-6)
