@@ -92,10 +92,18 @@ class BoundingBox:
         return tuple(slice(a, b) for a, b in zip(self.start, self.stop))
 
     def adjust(self, margin) -> 'BoundingBox':
-        """Grow (positive margin) or shrink the box symmetrically."""
-        m = Cartesian.from_collection(margin[:3]) \
-            if not isinstance(margin, Cartesian) else margin
+        """Grow (positive margin) or shrink the box symmetrically; margin
+        may be a scalar or a zyx triple."""
+        if isinstance(margin, int):
+            m = Cartesian(margin, margin, margin)
+        elif isinstance(margin, Cartesian):
+            m = margin
+        else:
+            m = Cartesian.from_collection(margin[:3])
         return BoundingBox(self.start - m, self.stop + m)
+
+    def clone(self) -> 'BoundingBox':
+        return BoundingBox(self.start, self.stop)
 
     def clamp(self, other: 'BoundingBox') -> 'BoundingBox':
         return BoundingBox(

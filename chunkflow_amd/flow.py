@@ -455,3 +455,142 @@ def log_summary(log_dir, output_size):
         print(f'speed: {voxels / total_mean / 1e3:.1f} kv/s '
               f'({voxels / total_mean / 1e6:.2f} mv/s)')
     yield get_initial_task()
+
+
+@main.command('copy-var')
+@click.option('--from-name', '-f', type=str, default='chunk',
+              help='Variable to be copied')
+@click.option('--to-name', '-t', type=str, default='chunk',
+              help='New variable name')
+@click.option('--deep-copy/--shallow-copy', type=bool, default=True,
+              help='really copy data or just create a new name or reference.')
+@operator
+def copy_var(tasks, from_name, to_name, deep_copy):
+    """Deep or shallow copy a variable."""
+    from copy import deepcopy
+    for task in tasks:
+        if task is not None:
+            if deep_copy:
+                v = task[from_name]
+                task[to_name] = v.clone() if isinstance(v, Chunk) \
+                    else deepcopy(v)
+            else:
+                task[to_name] = task[from_name]
+        yield task
+
+
+@main.command('delete-var')
+@click.option('--var-names', '-v', type=str, required=True,
+              help='the variable names to be deleted')
+@operator
+def delete_var(tasks, var_names):
+    """Delete a Chunk in task to release RAM (or HBM)."""
+    for task in tasks:
+        if task is not None:
+            for var_name in var_names.split(','):
+                del task[var_name]
+        yield task
+
+
+@main.command('skip-task-by-file')
+@click.option('--prefix', '-p', required=True, type=str,
+              help='the pre part of result file path')
+@click.option('--suffix', '-s', default='', type=str,
+              help='the post part of result file path.')
+@click.option('--mode', '-m',
+              type=click.Choice(['missing', 'empty', 'exist']),
+              default='exist',
+              help='skip if the corresponding file is missing/empty/exists')
+@click.option('--adjust-size', '-a', default=None, type=click.INT,
+              callback=default_none, help='expand or shrink the bounding box.')
+@operator
+def skip_task_by_file(tasks, prefix, suffix, mode, adjust_size):
+    """if a result file already exists, skip this task (the reference's
+    at-least-once idempotency family — SURVEY.md §5)."""
+    for task in tasks:
+        if task is not None:
+            bbox = task['bbox']
+            if adjust_size is not None:
+                bbox = bbox.adjust(adjust_size)
+            fname = prefix + bbox.string + suffix
+            if mode == 'empty':
+                if not os.path.exists(fname) or os.path.getsize(fname) == 0:
+                    task = None
+            elif mode == 'missing':
+                if not os.path.exists(fname):
+                    task = None
+            elif mode == 'exist':
+                if os.path.exists(fname):
+                    task = None
+        yield task
+
+
+@main.command('mark-complete')
+@click.option('--prefix', '-p', type=str, required=True,
+              help='pre-path of a file.')
+@click.option('--suffix', '-s', type=str, default='',
+              help='suffix of the flag file.')
+@operator
+def mark_complete(tasks, prefix, suffix):
+    """mark completion of a task as an empty file."""
+    from pathlib import Path
+    for task in tasks:
+        if task is not None:
+            Path(f"{prefix}{task['bbox'].string}{suffix}").touch()
+        yield task
+
+
+@main.command('skip-all-zero')
+@click.option('--input-chunk-name', '-i', type=str,
+              default=DEFAULT_CHUNK_NAME, help='input chunk name')
+@click.option('--prefix', '-p', type=str, default=None,
+              help='pre-path of a trace file.')
+@click.option('--suffix', '-s', type=str, default='',
+              help='post-path of a trace file.')
+@click.option('--adjust-size', '-a', type=click.INT, default=None,
+              help='change the bounding box of the trace file name.')
+@click.option('--chunk-bbox/--task-bbox', default=True,
+              help='use the chunk bounding box or the task one.')
+@operator
+def skip_all_zero(tasks, input_chunk_name, prefix, suffix, adjust_size,
+                  chunk_bbox):
+    """if chunk has all zero, skip this task."""
+    import torch
+    from pathlib import Path
+    for task in tasks:
+        if task is not None:
+            chunk = task[input_chunk_name]
+            arr = chunk.array
+            any_nonzero = bool((arr != 0).any().item()) \
+                if isinstance(arr, torch.Tensor) else bool(np.any(arr))
+            if not any_nonzero:
+                if prefix is not None:
+                    bbox = chunk.bbox if chunk_bbox else task['bbox']
+                    if adjust_size is not None:
+                        bbox = bbox.adjust(adjust_size)
+                    fname = f'{prefix}{bbox.string}{suffix}'
+                    if not os.path.exists(fname):
+                        Path(fname).touch()
+                task = None
+        yield task
+
+
+@main.command('skip-none')
+@click.option('--input-name', '-i', type=str, default=DEFAULT_CHUNK_NAME,
+              help='input name')
+@click.option('--touch/--no-touch', default=True,
+              help='touch an empty file or not')
+@click.option('--prefix', '-p', default=None, help='prefix of output file.')
+@click.option('--suffix', '-s', default=None, help='suffix of output file.')
+@operator
+def skip_none(tasks, input_name, touch, prefix, suffix):
+    """If item is None, skip this task."""
+    from pathlib import Path
+    for task in tasks:
+        if task is not None:
+            if task[input_name] is None:
+                if touch:
+                    assert prefix is not None and suffix is not None
+                    Path(f"{prefix}{task['bbox'].string}{suffix}").touch()
+                task = None
+        yield task

@@ -93,3 +93,35 @@ def test_log_summary(tmp_path):
                  '-s', '512', '512', '512'])
     assert 'inference: mean=3.000s' in r.output
     assert re.search(r'speed: .* kv/s', r.output)
+
+
+def test_skip_family_and_var_ops(tmp_path):
+    """The reference's at-least-once idempotency/resume family
+    (SURVEY.md §5): skip-task-by-file, mark-complete, skip-all-zero,
+    copy-var, delete-var."""
+    mark = str(tmp_path / 'done_')
+    # first run marks completion for both tasks
+    run_cli(['generate-tasks', '--roi-size', '8', '8', '16',
+             '--chunk-size', '8', '8', '8',
+             'create-chunk', '--dtype', 'uint8', '--pattern', 'sin',
+             'copy-var', '-f', 'chunk', '-t', 'backup',
+             'delete-var', '-v', 'backup',
+             'mark-complete', '-p', mark])
+    import glob
+    assert len(glob.glob(mark + '*')) == 2
+    # second run skips both tasks before create-chunk (skip-task-by-file)
+    r = run_cli(['generate-tasks', '--roi-size', '8', '8', '16',
+                 '--chunk-size', '8', '8', '8',
+                 'skip-task-by-file', '-p', mark, '-m', 'exist',
+                 'create-chunk', '--dtype', 'uint8'])
+    assert 'creating chunk' in r.output  # op ran, but tasks were None
+
+    # skip-all-zero drops zero chunks and touches a trace file
+    trace = str(tmp_path / 'zero_')
+    run_cli(['generate-tasks', '--roi-size', '8', '8', '8',
+             '--chunk-size', '8', '8', '8',
+             'create-chunk', '--dtype', 'uint8', '--pattern', 'zero',
+             'skip-all-zero', '-p', trace,
+             'mark-complete', '-p', str(tmp_path / 'never_')])
+    assert len(glob.glob(trace + '*')) == 1
+    assert len(glob.glob(str(tmp_path / 'never_*'))) == 0
