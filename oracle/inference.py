@@ -158,3 +158,52 @@ def oracle_inference(chunk, patch_size, overlap, num_output_channels=3,
         myelin_mask = out[-1] < mask_myelin_threshold
         out = out[:-1] * myelin_mask
     return out
+
+
+def oracle_inference_aligned(chunk, patch_size, overlap,
+                             num_output_channels=3, batch_size=1,
+                             engine=None, dtype='float32',
+                             output_crop_margin=None, patch_mask=None):
+    """Aligned-mode inference (mask_output_chunk=False): the output buffer
+    is input_size - 2*crop_margin (crop_margin defaults to the overlap,
+    inferencer.py:98-107), patches blend bump-masked WITHOUT the chunk-mask
+    normalize, and the buffer clipping drops the margins
+    (inferencer.py:124-139, 369-370; blend clipping chunk/base.py:796-807).
+    """
+    chunk = np.asarray(chunk)
+    input_size = chunk.shape[-3:]
+    if output_crop_margin is None:
+        output_crop_margin = overlap
+    if patch_mask is None:
+        patch_mask = make_patch_mask(patch_size, overlap, dtype=dtype)
+    stride = tuple(p - o for p, o in zip(patch_size, overlap))
+    for i, s, o in zip(input_size, stride, overlap):
+        assert (i - o) % s == 0, 'aligned mode needs aligned input'
+
+    slices = patch_slices_list(input_size, patch_size, overlap)
+    out_size3 = tuple(i - 2 * m for i, m in
+                      zip(input_size, output_crop_margin))
+    out = np.zeros((num_output_channels,) + out_size3, dtype=dtype)
+
+    if np.issubdtype(chunk.dtype, np.integer):
+        dtype_max = np.iinfo(chunk.dtype).max
+        chunk = chunk.astype(dtype)
+        chunk /= dtype_max
+
+    if engine is None:
+        def engine(batch):
+            return identity_engine(batch, patch_mask, num_output_channels)
+
+    buf = np.zeros((batch_size, 1) + tuple(patch_size), dtype=dtype)
+    for i in range(0, len(slices), batch_size):
+        batch = slices[i:i + batch_size]
+        for bi, ((iz, iy, ix), _) in enumerate(batch):
+            buf[bi, 0] = chunk[iz:iz + patch_size[0],
+                               iy:iy + patch_size[1],
+                               ix:ix + patch_size[2]]
+        out_patch = engine(buf)
+        for bi, (_, (oz, oy, ox)) in enumerate(batch):
+            blend_into(out, output_crop_margin, out_patch[bi],
+                       (oz, oy, ox))
+    np.testing.assert_array_less(out, 1.0001)
+    return out

@@ -512,3 +512,53 @@ def test_normalize_contrast_cli_gpu(tmp_path, golden):
     from chunkflow_amd.chunk import Chunk as C
     sin = C.create(size=(8, 40, 48), dtype='uint8', pattern='sin').array
     np.testing.assert_array_equal(got, oracle_normalize_contrast(sin))
+
+
+def test_fuzz_geometries_vs_oracle():
+    """Randomized geometry fuzz: odd sizes, tail clamps, duplicate
+    positions, varying batch/channels/offsets — HIP path vs oracle at
+    1e-5 on six cases."""
+    from chunkflow_amd.chunk import Chunk
+    from oracle import oracle_inference
+    rng = np.random.RandomState(99)
+    for case in range(6):
+        pz = int(rng.randint(6, 14))
+        py = int(rng.randint(16, 40))
+        px = int(rng.randint(16, 40))
+        ov = (max(1, pz // 4), max(2, py // 4), max(2, px // 4))
+        size = (pz + int(rng.randint(0, 2 * pz)),
+                py + int(rng.randint(0, 2 * py)),
+                px + int(rng.randint(0, 2 * px)))
+        nc = int(rng.randint(1, 4))
+        bs = int(rng.randint(1, 7))
+        off = tuple(int(v) for v in rng.randint(0, 20, 3))
+        arr = rng.randint(0, 256, size=size).astype(np.uint8)
+        inf = _hip_inferencer(patch_size=(pz, py, px), overlap=ov,
+                              framework='identity',
+                              num_output_channels=nc, batch_size=bs,
+                              mask_output_chunk=True)
+        got = inf(Chunk(arr.copy(), voxel_offset=off)).numpy().array
+        ref = oracle_inference(arr, (pz, py, px), ov,
+                               num_output_channels=nc, batch_size=bs,
+                               offset=off)
+        np.testing.assert_allclose(
+            got, ref, rtol=1e-5, atol=1e-6,
+            err_msg=f'case {case}: size={size} patch={(pz,py,px)} '
+                    f'ov={ov} nc={nc} bs={bs} off={off}')
+
+
+def test_aligned_mode_vs_oracle():
+    """Aligned mode (no chunk mask) vs the aligned-mode oracle."""
+    from chunkflow_amd.chunk import Chunk
+    from oracle.inference import oracle_inference_aligned
+    rng = np.random.RandomState(41)
+    ps, ov = (8, 24, 24), (2, 8, 8)
+    size = (8 + 2 * 6, 24 + 2 * 16, 24 + 16)  # aligned: (i-o)%s==0
+    arr = rng.randint(0, 256, size=size).astype(np.uint8)
+    inf = _hip_inferencer(patch_size=ps, overlap=ov, framework='identity',
+                          num_output_channels=2, batch_size=3,
+                          mask_output_chunk=False, input_size=size)
+    got = inf(Chunk(arr.copy())).numpy().array
+    ref = oracle_inference_aligned(arr, ps, ov, num_output_channels=2,
+                                   batch_size=3)
+    np.testing.assert_allclose(got, ref, rtol=1e-5, atol=1e-6)
