@@ -24,14 +24,21 @@ def run_one():
     from chunkflow_amd.ops import HipOps
     from chunkflow_amd.grouping import disjoint_groups
     from chunkflow_amd.patch_mask import make_patch_mask
-    from oracle.inference import patch_slices_list
 
     size = (512, 512, 512)
     ps, ov = (20, 256, 256), (4, 64, 64)
     C = 3
     ops = HipOps(0)
-    slices = patch_slices_list(size, ps, ov)
-    starts = np.array([o for _, o in slices], dtype=np.int64)
+    # the config-2 tail-clamped patch grid (same loop as the product tiler)
+    stride = tuple(p - o for p, o in zip(ps, ov))
+    axes = []
+    for d in range(3):
+        ax = []
+        for i in range(0, size[d] - ov[d], stride[d]):
+            ax.append(min(i, size[d] - ps[d]))
+        axes.append(ax)
+    starts = np.array([(z, y, x) for z in axes[0] for y in axes[1]
+                       for x in axes[2]], dtype=np.int64)
     groups = disjoint_groups(starts, ps, size)
     mask = torch.from_numpy(make_patch_mask(ps, ov).copy()).cuda()
     out = torch.zeros((C,) + size, dtype=torch.float32, device='cuda')
