@@ -1,0 +1,238 @@
+// Hand-written MFMA 3x3x3 convolution for gfx950 — the RSUNet ResBlock
+// conv (C == K in {28, 36, 48, 64}, stride 1, pad 1, NDHWC f32).
+//
+// MIOpen's ck-xdlops kernels reach ~58 TF/s f32 on these shapes; the small
+// channel depth (N-dim 28..64) starves generic implicit GEMM. This kernel
+// exploits the structure directly:
+//   * conv = sum over the 27 taps of GEMM( input-shifted [M x C],
+//     W_tap [C x K] ) — computed per workgroup from ONE LDS-resident input
+//     slab (output tile + 3x3x3 halo), so each input value is read from
+//     HBM once per tile (x-halo 2/34, y-halo 2/TY+2, z-halo shared);
+//   * v_mfma_f32_16x16x4_f32 (exact f32, 155 TF ceiling): D tile = 16
+//     x-positions x 16 output channels, reduction 4 input channels per
+//     instruction; A operand = one LDS dword per lane (bank-conflict-free
+//     by construction: the lane address stride C mod 32 is kept out of
+//     {0, 16} by padding the slab's x-stride), B operand = one weight
+//     dword per lane, staged through LDS per tap and reused by every
+//     M-tile of the wave;
+//   * fused bias + optional ELU + optional residual add in the epilogue.
+//
+// Geometry per workgroup (template TY): output tile TZ=2 x TY x TX=32
+// voxels, 4 waves; TY=4 for C<=36 (slab <= 118 KB), TY=2 for C in
+// {48, 64} (slab <= 148 KB incl. padding). Weights are (27, C, K) f32
+// (prepared host-side from torch's (K, C, 3, 3, 3)).
+#include <hip/hip_runtime.h>
+
+#include "cfx_internal.h"
+
+namespace {
+
+typedef float f32x4 __attribute__((ext_vector_type(4)));
+
+constexpr int TX = 32;  // output x per workgroup
+constexpr int TZ = 2;   // output z per workgroup
+
+constexpr int padc(int C) {  // slab voxel stride: keep (stride mod 32)
+    return (C % 32 == 0 || C % 32 == 16) ? C + 4 : C;  // out of {0, 16}
+}
+
+// one wave handles TY==4 ? one y row (2 z x 32 x) : (TY==2) one (z, y)
+// pair of 1 z x 32 x
+template <int C, int K, int TY>
+__global__ __launch_bounds__(256, 1) void k_conv3(
+    const float* __restrict__ in,    // (N, D, H, W, C) channels-last
+    const float* __restrict__ wgt,   // (27, C, K)
+    const float* __restrict__ bias,  // (K)
+    const float* __restrict__ res,   // optional residual, same layout as out
+    float* __restrict__ out,         // (N, D, H, W, K)
+    int N, int D, int H, int W, int do_elu) {
+    constexpr int PC = padc(C);
+    constexpr int SX = TX + 2;            // slab x extent (halo)
+    constexpr int SY = TY + 2;
+    constexpr int SZ = TZ + 2;
+    constexpr int KK = C / 4;             // reduction steps per tap
+    constexpr int NT = (K + 15) / 16;     // 16-wide output-channel tiles
+    // wave covers TZ*TX voxels at one y (TY==4) or TX at one (z,y)
+    // (TY==2); each m-tile is 16 consecutive x positions
+    constexpr int M_TILES = TY == 4 ? (TZ * TX) / 16 : TX / 16;
+
+    __shared__ float slab[SZ * SY * SX * PC];
+    __shared__ float wtile[C * 16];  // one tap x one 16-wide K tile
+
+    const int bx = blockIdx.x;                  // x block
+    const int by = blockIdx.y;                  // y block
+    const int bzn = blockIdx.z;                 // fused (n, z-block)
+    const int zblocks = (D + TZ - 1) / TZ;
+    const int n = bzn / zblocks;
+    const int z0 = (bzn % zblocks) * TZ;
+    const int y0 = by * TY;
+    const int x0 = bx * TX;
+
+    const int tid = threadIdx.x;                // 0..255
+    const int wave = tid >> 6;
+    const int lane = tid & 63;
+
+    // ---- stage the input slab (zero-padded at volume borders) ----------
+    // each thread copies float4 runs along C; C % 4 == 0 for all widths
+    {
+        const int c4n = C / 4;
+        const int vox = SZ * SY * SX;
+        for (int idx = tid; idx < vox * c4n; idx += 256) {
+            const int c4 = idx % c4n;
+            const int v = idx / c4n;
+            const int sx = v % SX;
+            const int sy = (v / SX) % SY;
+            const int sz = v / (SX * SY);
+            const int gz = z0 + sz - 1;
+            const int gy = y0 + sy - 1;
+            const int gx = x0 + sx - 1;
+            f32x4 val = {0.f, 0.f, 0.f, 0.f};
+            if (gz >= 0 && gz < D && gy >= 0 && gy < H && gx >= 0 &&
+                gx < W) {
+                const float* src = in +
+                    ((((long long)n * D + gz) * H + gy) * W + gx) * C +
+                    c4 * 4;
+                val = *reinterpret_cast<const f32x4*>(src);
+            }
+            float* dst = &slab[((sz * SY + sy) * SX + sx) * PC + c4 * 4];
+            dst[0] = val.x;
+            dst[1] = val.y;
+            dst[2] = val.z;
+            dst[3] = val.w;
+        }
+    }
+    __syncthreads();
+
+    // ---- wave tile origin ----------------------------------------------
+    // TY==4: wave w -> y=w, covers z0..z0+1 x 32x  (M_TILES = 4)
+    // TY==2: wave w -> (z = w>>1, y = w&1), covers 32x (M_TILES = 2)
+    const int wy = TY == 4 ? wave : (wave & 1);
+    const int wz0 = TY == 4 ? 0 : (wave >> 1);
+
+    f32x4 acc[M_TILES][NT];
+#pragma unroll
+    for (int m = 0; m < M_TILES; ++m)
+#pragma unroll
+        for (int t = 0; t < NT; ++t)
+            acc[m][t] = {0.f, 0.f, 0.f, 0.f};
+
+    const int a_row = lane & 15;   // x within the m-tile (A row, B col)
+    const int a_k = lane >> 4;     // reduction sub-index (0..3)
+    const int col16 = lane & 15;
+
+    // ---- 27 taps ---------------------------------------------------------
+    for (int tap = 0; tap < 27; ++tap) {
+        const int dz = tap / 9 - 1;
+        const int dy = (tap / 3) % 3 - 1;
+        const int dx = tap % 3 - 1;
+#pragma unroll
+        for (int nt = 0; nt < NT; ++nt) {
+            // stage W[tap][:, nt*16 : nt*16+16] into LDS (C x 16)
+            {
+                const int j0 = nt * 16;
+                const int jn = (K - j0) < 16 ? (K - j0) : 16;
+                for (int idx = tid; idx < C * 16; idx += 256) {
+                    const int j = idx & 15;
+                    const int c = idx >> 4;
+                    wtile[c * 16 + j] =
+                        j < jn ? wgt[((long long)tap * C + c) * K + j0 + j]
+                               : 0.f;
+                }
+            }
+            __syncthreads();
+            // kk outer / m inner: the M_TILES accumulator chains
+            // interleave, hiding the 40-cycle dependent-MFMA latency (the
+            // 16x16x4 issue interval is 32) at one wave per SIMD
+            const float* arow[M_TILES];
+#pragma unroll
+            for (int m = 0; m < M_TILES; ++m) {
+                const int mz = TY == 4 ? (m >> 1) : wz0;
+                const int mx = TY == 4 ? (m & 1) * 16 : m * 16;
+                arow[m] = &slab[(((1 + mz + dz) * SY +
+                                  (1 + wy + dy)) * SX +
+                                 (1 + mx + dx)) * PC + a_row * PC + a_k];
+            }
+#pragma unroll
+            for (int kk = 0; kk < KK; ++kk) {
+                // A[i = l&15][k = l>>4], B[k = l>>4][j = l&15]
+                const float b = wtile[(kk * 4 + a_k) * 16 + col16];
+#pragma unroll
+                for (int m = 0; m < M_TILES; ++m) {
+                    const float a = arow[m][kk * 4];
+                    acc[m][nt] = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                        a, b, acc[m][nt], 0, 0, 0);
+                }
+            }
+            __syncthreads();
+        }
+    }
+
+    // ---- epilogue: D[row = (lane>>4)*4 + reg][col = lane&15] -------------
+    const int col = lane & 15;
+    const int rbase = (lane >> 4) * 4;
+#pragma unroll
+    for (int m = 0; m < M_TILES; ++m) {
+        const int mz = TY == 4 ? (m >> 1) : wz0;
+        const int mx = TY == 4 ? (m & 1) * 16 : m * 16;
+        const int gz = z0 + mz;
+        const int gy = y0 + wy;
+        if (gz >= D || gy >= H) continue;
+#pragma unroll
+        for (int t = 0; t < NT; ++t) {
+            const int j = t * 16 + col;
+            if (j >= K) continue;
+            const float bj = bias ? bias[j] : 0.f;
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                const int gx = x0 + mx + rbase + r;
+                if (gx >= W) continue;
+                long long o =
+                    ((((long long)n * D + gz) * H + gy) * W + gx) * K + j;
+                float v = acc[m][t][r] + bj;
+                if (res) v += res[o];
+                if (do_elu) v = v > 0.f ? v : expm1f(v);
+                out[o] = v;
+            }
+        }
+    }
+}
+
+}  // namespace
+
+extern "C" int cfx_conv3_ndhwc(cfx_ctx* ctx, const float* in,
+                               const float* wgt, const float* bias,
+                               const float* residual, float* out, int N,
+                               int D, int H, int W, int C, int K,
+                               int do_elu) {
+    if (C != K) {
+        g_err = "cfx_conv3_ndhwc: only C == K widths are instantiated";
+        return -1;
+    }
+    const int zblocks = (D + TZ - 1) / TZ;
+    dim3 block(256);
+#define CFX_CONV_CASE(CW, TYV)                                               \
+    case CW: {                                                               \
+        dim3 grid((W + TX - 1) / TX, (H + TYV - 1) / TYV,                    \
+                  (unsigned)(N * zblocks));                                  \
+        hipEvent_t e0;                                                       \
+        if (prof_begin(ctx, &e0)) return -1;                                 \
+        hipLaunchKernelGGL((k_conv3<CW, CW, TYV>), grid, block, 0,           \
+                           ctx->stream, in, wgt, bias, residual, out, N, D,  \
+                           H, W, do_elu);                                    \
+        CFX_CHECK(hipGetLastError());                                        \
+        double flops = 2.0 * 27.0 * CW * CW * (double)N * D * H * W;         \
+        if (prof_end(ctx, e0, CFX_K_CONV, flops)) return -1;                 \
+        break;                                                               \
+    }
+    switch (C) {
+        CFX_CONV_CASE(28, 4)
+        CFX_CONV_CASE(36, 4)
+        CFX_CONV_CASE(48, 2)
+        CFX_CONV_CASE(64, 2)
+        default:
+            g_err = "cfx_conv3_ndhwc: unsupported channel width";
+            return -1;
+    }
+#undef CFX_CONV_CASE
+    return 0;
+}

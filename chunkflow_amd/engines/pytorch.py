@@ -51,6 +51,13 @@ class PyTorchEngine(EngineBase):
                               os.environ.get('CFX_CHANNELS_LAST', '1') != '0')
         if self.channels_last:
             self.model = self.model.to(memory_format=torch.channels_last_3d)
+            # swap eligible 3x3x3 ResBlock convs for the hand-written MFMA
+            # kernel (fastconv.py); CFX_FASTCONV=0 keeps MIOpen everywhere
+            if (self.dtype != 'bfloat16'
+                    and os.environ.get('CFX_FASTCONV', '1') != '0'):
+                from ..fastconv import maybe_accelerate
+                idx = int(str(self.device).split(':')[-1])                     if ':' in str(self.device) else 0
+                self.fastconv_count = maybe_accelerate(self.model, idx)
         torch.backends.cudnn.benchmark = True
         self.pre_process = getattr(net_source, 'pre_process', None)
         self.post_process = getattr(net_source, 'post_process', None)

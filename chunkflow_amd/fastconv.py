@@ -1,0 +1,79 @@
+"""Graph surgery: swap eligible nn.Conv3d modules for the hand-written
+gfx950 MFMA conv kernel (csrc/conv.hip).
+
+Eligible: kernel (3,3,3), stride 1, padding 1, dilation 1, groups 1,
+in_channels == out_channels in {28, 36, 48, 64} — the RSUNet ResBlock
+convs, ~85% of the affinity UNet's FLOPs. Everything else stays on
+MIOpen. Enabled in the pytorch engine via CFX_FASTCONV (see engine);
+weights are repacked once to the kernel's (27, C, K) tap-major layout.
+"""
+import torch
+import torch.nn as nn
+
+_CTX = {}
+
+
+def get_cfx(device_index: int = 0):
+    if device_index not in _CTX:
+        from .hip import CfxContext
+        ctx = CfxContext(device_index)
+        ctx.adopt_torch_stream()
+        _CTX[device_index] = ctx
+    return _CTX[device_index]
+
+
+ELIGIBLE_WIDTHS = (28, 36, 48, 64)
+
+
+def _eligible(m: nn.Module) -> bool:
+    return (isinstance(m, nn.Conv3d)
+            and m.kernel_size == (3, 3, 3)
+            and m.stride == (1, 1, 1)
+            and m.padding == (1, 1, 1)
+            and m.dilation == (1, 1, 1)
+            and m.groups == 1
+            and m.in_channels == m.out_channels
+            and m.in_channels in ELIGIBLE_WIDTHS)
+
+
+class CfxConv3d(nn.Module):
+    """Drop-in for an eligible nn.Conv3d; consumes/produces channels-last
+    (NDHWC) tensors, exactly the memory format the engine runs in."""
+
+    def __init__(self, conv: nn.Conv3d, device_index: int = 0):
+        super().__init__()
+        self.C = conv.in_channels
+        self.K = conv.out_channels
+        self.device_index = device_index
+        w = conv.weight.detach()  # (K, C, 3, 3, 3)
+        wtap = w.permute(2, 3, 4, 1, 0).reshape(27, self.C, self.K)
+        self.register_buffer('wtap', wtap.contiguous().float())
+        if conv.bias is not None:
+            self.register_buffer('bias', conv.bias.detach().float())
+        else:
+            self.bias = None
+
+    def forward(self, x):
+        assert x.dtype == torch.float32, 'fastconv is the f32 path'
+        x = x.contiguous(memory_format=torch.channels_last_3d)
+        n, c, d, h, w = x.shape
+        out = torch.empty((n, self.K, d, h, w), dtype=torch.float32,
+                          device=x.device,
+                          memory_format=torch.channels_last_3d)
+        get_cfx(self.device_index).conv3_ndhwc(
+            x.data_ptr(), self.wtap.data_ptr(),
+            self.bias.data_ptr() if self.bias is not None else None,
+            None, out.data_ptr(), n, d, h, w, self.C, self.K)
+        return out
+
+
+def maybe_accelerate(model: nn.Module, device_index: int = 0) -> int:
+    """Replace eligible convs in-place; returns the replacement count."""
+    count = 0
+    for parent in model.modules():
+        for name, child in list(parent.named_children()):
+            if _eligible(child):
+                setattr(parent, name, CfxConv3d(child, device_index)
+                        .to(f'cuda:{device_index}'))
+                count += 1
+    return count

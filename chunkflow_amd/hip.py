@@ -24,13 +24,14 @@ SYMBOLS = [
     'cfx_crop_margin',
     'cfx_mask_using_last_channel', 'cfx_threshold', 'cfx_nonzero_u8',
     'cfx_connected_components', 'cfx_hist_u8', 'cfx_lut_apply_u8',
+    'cfx_conv3_ndhwc',
     'cfx_profile_enable', 'cfx_profile_reset',
     'cfx_profile_get',
 ]
 
 KERNEL_IDS = {
     'blend': 0, 'extract': 1, 'normalize': 2, 'cast': 3, 'reciprocal': 4,
-    'maskmul': 5, 'crop': 6, 'max': 7, 'myelin': 8, 'cc': 9,
+    'maskmul': 5, 'crop': 6, 'max': 7, 'myelin': 8, 'cc': 9, 'conv': 10,
 }
 
 _lib = None
@@ -256,6 +257,18 @@ class CfxContext:
             ctypes.c_void_p(self.ctx), ctypes.c_void_p(buf_ptr),
             ctypes.c_longlong(n_per_sec), ctypes.c_int(nsec),
             ctypes.c_void_p(lut_ptr)), 'cfx_lut_apply_u8')
+
+    def conv3_ndhwc(self, in_ptr, wgt_ptr, bias_ptr, residual_ptr, out_ptr,
+                    n, d, h, w, c, k, do_elu=False):
+        self._chk(self.lib.cfx_conv3_ndhwc(
+            ctypes.c_void_p(self.ctx), ctypes.c_void_p(in_ptr),
+            ctypes.c_void_p(wgt_ptr),
+            ctypes.c_void_p(bias_ptr) if bias_ptr else None,
+            ctypes.c_void_p(residual_ptr) if residual_ptr else None,
+            ctypes.c_void_p(out_ptr), ctypes.c_int(n), ctypes.c_int(d),
+            ctypes.c_int(h), ctypes.c_int(w), ctypes.c_int(c),
+            ctypes.c_int(k), ctypes.c_int(1 if do_elu else 0)),
+            'cfx_conv3_ndhwc')
 
     # --- profiling ----------------------------------------------------------
     def profile_enable(self, enable=True):

@@ -129,6 +129,15 @@ int cfx_hist_u8(cfx_ctx* ctx, const unsigned char* in, long long n_per_sec,
 int cfx_lut_apply_u8(cfx_ctx* ctx, unsigned char* buf, long long n_per_sec,
                      int nsec, const unsigned char* lut);
 
+/* ---- hand-written MFMA convolution (RSUNet ResBlock 3x3x3) -------------- */
+/* NDHWC f32, stride 1, pad 1, C == K in {28, 36, 48, 64}; wgt layout
+ * (27, C, K) with tap = ((dz+1)*3 + (dy+1))*3 + (dx+1); bias may be NULL;
+ * residual (same layout as out) may be NULL; do_elu applies ELU(alpha=1)
+ * after bias/residual. profile bytes field carries FLOPs for this id. */
+int cfx_conv3_ndhwc(cfx_ctx* ctx, const float* in, const float* wgt,
+                    const float* bias, const float* residual, float* out,
+                    int N, int D, int H, int W, int C, int K, int do_elu);
+
 /* ---- kernel timing (HIP events on the context stream) ------------------ */
 enum cfx_kernel_id {
     CFX_K_BLEND = 0,
@@ -141,7 +150,8 @@ enum cfx_kernel_id {
     CFX_K_MAX = 7,
     CFX_K_MYELIN = 8,
     CFX_K_CC = 9,
-    CFX_K_COUNT = 10
+    CFX_K_CONV = 10,
+    CFX_K_COUNT = 11
 };
 int cfx_profile_enable(cfx_ctx* ctx, int enable);
 int cfx_profile_reset(cfx_ctx* ctx);

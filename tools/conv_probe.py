@@ -1,0 +1,61 @@
+"""A/B the hand-written MFMA conv vs torch/MIOpen per RSUNet shape, and
+check numerics vs a float64 CPU reference."""
+import os, sys, time
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+import numpy as np
+import torch
+import torch.nn.functional as F
+
+from chunkflow_amd.fastconv import get_cfx
+
+torch.backends.cudnn.benchmark = True
+cl = torch.channels_last_3d
+
+def probe(C, D, H, W, N=12, iters=10):
+    torch.manual_seed(0)
+    x = torch.randn(N, C, D, H, W, device='cuda').contiguous(memory_format=cl)
+    wt = torch.randn(C, C, 3, 3, 3, device='cuda') * (1.0 / (27 * C) ** 0.5)
+    bias = torch.randn(C, device='cuda') * 0.1
+    wtap = wt.permute(2, 3, 4, 1, 0).reshape(27, C, C).contiguous()
+    out = torch.empty_like(x)
+    cfx = get_cfx(0)
+
+    # numerics vs fp64 CPU
+    ref = F.conv3d(x.double().cpu(), wt.double().cpu(), bias.double().cpu(),
+                   padding=1)[:2]
+    cfx.conv3_ndhwc(x.data_ptr(), wtap.data_ptr(), bias.data_ptr(), None,
+                    out.data_ptr(), N, D, H, W, C, C)
+    torch.cuda.synchronize()
+    err = (out[:2].double().cpu() - ref).abs().max().item()
+    scale = ref.abs().max().item()
+
+    def t_mine():
+        t0 = time.perf_counter()
+        for _ in range(iters):
+            cfx.conv3_ndhwc(x.data_ptr(), wtap.data_ptr(), bias.data_ptr(),
+                            None, out.data_ptr(), N, D, H, W, C, C)
+        torch.cuda.synchronize()
+        return (time.perf_counter() - t0) / iters
+
+    wm = wt.contiguous(memory_format=cl)
+    def t_torch():
+        t0 = time.perf_counter()
+        for _ in range(iters):
+            F.conv3d(x, wm, bias, padding=1)
+        torch.cuda.synchronize()
+        return (time.perf_counter() - t0) / iters
+
+    for _ in range(3):
+        t_mine(); t_torch()
+    tm, tt = t_mine(), t_torch()
+    flops = 2.0 * 27 * C * C * N * D * H * W
+    print({'C': C, 'shape': (N, D, H, W), 'err_vs_fp64': err,
+           'rel': err / scale,
+           'mine_ms': tm * 1e3, 'torch_ms': tt * 1e3,
+           'mine_TF': flops / tm / 1e12, 'torch_TF': flops / tt / 1e12,
+           'speedup': tt / tm}, flush=True)
+
+probe(28, 20, 256, 256)
+probe(36, 20, 128, 128)
+probe(48, 20, 64, 64)
+probe(64, 20, 32, 32)
