@@ -29,16 +29,17 @@ namespace {
 
 typedef float f32x4 __attribute__((ext_vector_type(4)));
 
-constexpr int TX = 32;  // output x per workgroup
 constexpr int TZ = 2;   // output z per workgroup
 
 constexpr int padc(int C) {  // slab voxel stride: keep (stride mod 32)
     return (C % 32 == 0 || C % 32 == 16) ? C + 4 : C;  // out of {0, 16}
 }
 
-// one wave handles TY==4 ? one y row (2 z x 32 x) : (TY==2) one (z, y)
-// pair of 1 z x 32 x
-template <int C, int K, int TY>
+// one wave handles TY==4 ? one y row (2 z x TX x) : (TY==2) one (z, y)
+// pair of 1 z x TX x. TAPG = taps whose weights are staged per barrier
+// pair (the dominant v1 cost was 108 stage+barrier episodes per
+// workgroup; TAPG=9 cuts it to 6 for C=28).
+template <int C, int K, int TY, int TX, int TAPG>
 __global__ __launch_bounds__(256, 1) void k_conv3(
     const float* __restrict__ in,    // (N, D, H, W, C) channels-last
     const float* __restrict__ wgt,   // (27, C, K)
@@ -57,7 +58,9 @@ __global__ __launch_bounds__(256, 1) void k_conv3(
     constexpr int M_TILES = TY == 4 ? (TZ * TX) / 16 : TX / 16;
 
     __shared__ float slab[SZ * SY * SX * PC];
-    __shared__ float wtile[C * 16];  // one tap x one 16-wide K tile
+    // TAPG taps x NT 16-wide K tiles, each a [C][16] block (the 16-dword
+    // row stride keeps the B-fragment reads bank-conflict-free)
+    __shared__ float wtile[TAPG * NT * C * 16];
 
     const int bx = blockIdx.x;                  // x block
     const int by = blockIdx.y;                  // y block
@@ -120,29 +123,30 @@ __global__ __launch_bounds__(256, 1) void k_conv3(
     const int a_k = lane >> 4;     // reduction sub-index (0..3)
     const int col16 = lane & 15;
 
-    // ---- 27 taps ---------------------------------------------------------
-    for (int tap = 0; tap < 27; ++tap) {
-        const int dz = tap / 9 - 1;
-        const int dy = (tap / 3) % 3 - 1;
-        const int dx = tap % 3 - 1;
+    // ---- 27 taps in groups of TAPG -------------------------------------
+    static_assert(27 % TAPG == 0, "TAPG must divide 27");
+    for (int g = 0; g < 27 / TAPG; ++g) {
+        // stage the group's weights: TAPG x NT blocks of [C][16]
+        for (int idx = tid; idx < TAPG * NT * C * 16; idx += 256) {
+            const int j = idx & 15;
+            const int c = (idx >> 4) % C;
+            const int nt = (idx >> 4) / C % NT;
+            const int tl = (idx >> 4) / C / NT;
+            const int jg = nt * 16 + j;
+            wtile[idx] = jg < K
+                ? wgt[((long long)(g * TAPG + tl) * C + c) * K + jg]
+                : 0.f;
+        }
+        __syncthreads();
 #pragma unroll
-        for (int nt = 0; nt < NT; ++nt) {
-            // stage W[tap][:, nt*16 : nt*16+16] into LDS (C x 16)
-            {
-                const int j0 = nt * 16;
-                const int jn = (K - j0) < 16 ? (K - j0) : 16;
-                for (int idx = tid; idx < C * 16; idx += 256) {
-                    const int j = idx & 15;
-                    const int c = idx >> 4;
-                    wtile[c * 16 + j] =
-                        j < jn ? wgt[((long long)tap * C + c) * K + j0 + j]
-                               : 0.f;
-                }
-            }
-            __syncthreads();
-            // kk outer / m inner: the M_TILES accumulator chains
-            // interleave, hiding the 40-cycle dependent-MFMA latency (the
-            // 16x16x4 issue interval is 32) at one wave per SIMD
+        for (int tl = 0; tl < TAPG; ++tl) {
+            const int tap = g * TAPG + tl;
+            const int dz = tap / 9 - 1;
+            const int dy = (tap / 3) % 3 - 1;
+            const int dx = tap % 3 - 1;
+            // kk outer / m inner: the accumulator chains interleave,
+            // hiding the 40-cycle dependent-MFMA latency (the 16x16x4
+            // issue interval is 32) at one wave per SIMD
             const float* arow[M_TILES];
 #pragma unroll
             for (int m = 0; m < M_TILES; ++m) {
@@ -152,19 +156,24 @@ __global__ __launch_bounds__(256, 1) void k_conv3(
                                   (1 + wy + dy)) * SX +
                                  (1 + mx + dx)) * PC + a_row * PC + a_k];
             }
+            const float* wblk = &wtile[tl * NT * C * 16];
 #pragma unroll
-            for (int kk = 0; kk < KK; ++kk) {
-                // A[i = l&15][k = l>>4], B[k = l>>4][j = l&15]
-                const float b = wtile[(kk * 4 + a_k) * 16 + col16];
+            for (int nt = 0; nt < NT; ++nt) {
 #pragma unroll
-                for (int m = 0; m < M_TILES; ++m) {
-                    const float a = arow[m][kk * 4];
-                    acc[m][nt] = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                        a, b, acc[m][nt], 0, 0, 0);
+                for (int kk = 0; kk < KK; ++kk) {
+                    // A[i = l&15][k = l>>4], B[k = l>>4][j = l&15]
+                    const float b =
+                        wblk[(nt * C + kk * 4 + a_k) * 16 + col16];
+#pragma unroll
+                    for (int m = 0; m < M_TILES; ++m) {
+                        const float a = arow[m][kk * 4];
+                        acc[m][nt] = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                            a, b, acc[m][nt], 0, 0, 0);
+                    }
                 }
             }
-            __syncthreads();
         }
+        __syncthreads();
     }
 
     // ---- epilogue: D[row = (lane>>4)*4 + reg][col = lane&15] -------------
@@ -210,25 +219,30 @@ extern "C" int cfx_conv3_ndhwc(cfx_ctx* ctx, const float* in,
     }
     const int zblocks = (D + TZ - 1) / TZ;
     dim3 block(256);
-#define CFX_CONV_CASE(CW, TYV)                                               \
+    // LDS budget per width (slab + wtile <= 160 KiB):
+    //   28: 91.4 + 9 taps * 32.3/9 ... = 91.4 + 28.7 = 120 KiB (TAPG 9)
+    //   36: 117.5 + 3 * 6.9 = 138 KiB (TAPG 3)
+    //   48: 113.2 + 3 * 9.2 = 141 KiB (TAPG 3, TY 2)
+    //   64: TX=16 slab 78.3 + 16.4 = 95 KiB (TAPG 1, TY 2)
+#define CFX_CONV_CASE(CW, TYV, TXV, TAPGV)                                   \
     case CW: {                                                               \
-        dim3 grid((W + TX - 1) / TX, (H + TYV - 1) / TYV,                    \
+        dim3 grid((W + TXV - 1) / TXV, (H + TYV - 1) / TYV,                  \
                   (unsigned)(N * zblocks));                                  \
         hipEvent_t e0;                                                       \
         if (prof_begin(ctx, &e0)) return -1;                                 \
-        hipLaunchKernelGGL((k_conv3<CW, CW, TYV>), grid, block, 0,           \
-                           ctx->stream, in, wgt, bias, residual, out, N, D,  \
-                           H, W, do_elu);                                    \
+        hipLaunchKernelGGL((k_conv3<CW, CW, TYV, TXV, TAPGV>), grid, block,  \
+                           0, ctx->stream, in, wgt, bias, residual, out, N,  \
+                           D, H, W, do_elu);                                 \
         CFX_CHECK(hipGetLastError());                                        \
         double flops = 2.0 * 27.0 * CW * CW * (double)N * D * H * W;         \
         if (prof_end(ctx, e0, CFX_K_CONV, flops)) return -1;                 \
         break;                                                               \
     }
     switch (C) {
-        CFX_CONV_CASE(28, 4)
-        CFX_CONV_CASE(36, 4)
-        CFX_CONV_CASE(48, 2)
-        CFX_CONV_CASE(64, 2)
+        CFX_CONV_CASE(28, 4, 32, 9)
+        CFX_CONV_CASE(36, 4, 32, 3)
+        CFX_CONV_CASE(48, 2, 32, 3)
+        CFX_CONV_CASE(64, 2, 16, 1)
         default:
             g_err = "cfx_conv3_ndhwc: unsupported channel width";
             return -1;
