@@ -76,32 +76,55 @@ __global__ __launch_bounds__(256, 1) void k_conv3(
     const int lane = tid & 63;
 
     // ---- stage the input slab (zero-padded at volume borders) ----------
-    // each thread copies float4 runs along C; C % 4 == 0 for all widths
+    // float4 runs along C (C % 4 == 0 for all widths). Loads are issued
+    // UNCONDITIONALLY from clamped addresses and zeroed afterwards — a
+    // per-element guarded load makes hipcc branch around each load and
+    // drain vmcnt per element (cdna_hip_programming.md §5 trap (c)), which
+    // was the dominant cost of the first version at 1 workgroup/CU.
     {
         const int c4n = C / 4;
         const int vox = SZ * SY * SX;
-        for (int idx = tid; idx < vox * c4n; idx += 256) {
-            const int c4 = idx % c4n;
-            const int v = idx / c4n;
-            const int sx = v % SX;
-            const int sy = (v / SX) % SY;
-            const int sz = v / (SX * SY);
-            const int gz = z0 + sz - 1;
-            const int gy = y0 + sy - 1;
-            const int gx = x0 + sx - 1;
-            f32x4 val = {0.f, 0.f, 0.f, 0.f};
-            if (gz >= 0 && gz < D && gy >= 0 && gy < H && gx >= 0 &&
-                gx < W) {
-                const float* src = in +
-                    ((((long long)n * D + gz) * H + gy) * W + gx) * C +
-                    c4 * 4;
-                val = *reinterpret_cast<const f32x4*>(src);
+        const bool interior = z0 >= 1 && z0 + TZ + 1 <= D && y0 >= 1 &&
+                              y0 + TY + 1 <= H && x0 >= 1 &&
+                              x0 + TX + 1 <= W;
+        const float* base =
+            in + ((((long long)n * D + z0 - 1) * H + y0 - 1) * W + x0 - 1)
+                 * C;
+        if (interior) {
+            for (int idx = tid; idx < vox * c4n; idx += 256) {
+                const int c4 = idx % c4n;
+                const int v = idx / c4n;
+                const int sx = v % SX;
+                const int sy = (v / SX) % SY;
+                const int sz = v / (SX * SY);
+                const f32x4 val = *reinterpret_cast<const f32x4*>(
+                    base + (((long long)sz * H + sy) * W + sx) * C +
+                    c4 * 4);
+                *reinterpret_cast<f32x4*>(
+                    &slab[((sz * SY + sy) * SX + sx) * PC + c4 * 4]) = val;
             }
-            float* dst = &slab[((sz * SY + sy) * SX + sx) * PC + c4 * 4];
-            dst[0] = val.x;
-            dst[1] = val.y;
-            dst[2] = val.z;
-            dst[3] = val.w;
+        } else {
+            for (int idx = tid; idx < vox * c4n; idx += 256) {
+                const int c4 = idx % c4n;
+                const int v = idx / c4n;
+                const int sx = v % SX;
+                const int sy = (v / SX) % SY;
+                const int sz = v / (SX * SY);
+                const int gz = z0 + sz - 1;
+                const int gy = y0 + sy - 1;
+                const int gx = x0 + sx - 1;
+                const bool ok = gz >= 0 && gz < D && gy >= 0 && gy < H &&
+                                gx >= 0 && gx < W;
+                const int cz = ok ? gz : 0;
+                const int cy = ok ? gy : 0;
+                const int cx = ok ? gx : 0;
+                f32x4 val = *reinterpret_cast<const f32x4*>(
+                    in + ((((long long)n * D + cz) * H + cy) * W + cx) * C +
+                    c4 * 4);
+                if (!ok) val = {0.f, 0.f, 0.f, 0.f};
+                *reinterpret_cast<f32x4*>(
+                    &slab[((sz * SY + sy) * SX + sx) * PC + c4 * 4]) = val;
+            }
         }
     }
     __syncthreads();
