@@ -29,18 +29,17 @@ namespace {
 
 typedef float f32x4 __attribute__((ext_vector_type(4)));
 
-constexpr int TZ = 2;   // output z per workgroup
-
 constexpr int padc(int C) {  // slab voxel stride: keep (stride mod 32)
     return (C % 32 == 0 || C % 32 == 16) ? C + 4 : C;  // out of {0, 16}
 }
 
-// one wave handles TY==4 ? one y row (2 z x TX x) : (TY==2) one (z, y)
-// pair of 1 z x TX x. TAPG = taps whose weights are staged per barrier
-// pair (the dominant v1 cost was 108 stage+barrier episodes per
-// workgroup; TAPG=9 cuts it to 6 for C=28).
-template <int C, int K, int TY, int TX, int TAPG>
-__global__ __launch_bounds__(256, 1) void k_conv3(
+// Generic wave->tile mapping: the workgroup's TZ x TY x (TX/16) 16-wide
+// m-tiles are dealt round-robin to NW = NTHREADS/64 waves; NTHREADS=512
+// puts 2 waves on each SIMD (the co-resident wave hides LDS latency the
+// single wave of the 256-thread shape cannot). TAPG = taps whose weights
+// are staged per barrier pair.
+template <int C, int K, int TZv, int TY, int TX, int TAPG, int NTHREADS>
+__global__ __launch_bounds__(NTHREADS, 1) void k_conv3(
     const float* __restrict__ in,    // (N, D, H, W, C) channels-last
     const float* __restrict__ wgt,   // (27, C, K)
     const float* __restrict__ bias,  // (K)
@@ -50,12 +49,13 @@ __global__ __launch_bounds__(256, 1) void k_conv3(
     constexpr int PC = padc(C);
     constexpr int SX = TX + 2;            // slab x extent (halo)
     constexpr int SY = TY + 2;
-    constexpr int SZ = TZ + 2;
+    constexpr int SZ = TZv + 2;
     constexpr int KK = C / 4;             // reduction steps per tap
     constexpr int NT = (K + 15) / 16;     // 16-wide output-channel tiles
-    // wave covers TZ*TX voxels at one y (TY==4) or TX at one (z,y)
-    // (TY==2); each m-tile is 16 consecutive x positions
-    constexpr int M_TILES = TY == 4 ? (TZ * TX) / 16 : TX / 16;
+    constexpr int NW = NTHREADS / 64;     // waves per workgroup
+    constexpr int XT = TX / 16;           // 16-wide x tiles
+    constexpr int M_TILES = (TZv * TY * XT) / NW;  // m-tiles per wave
+    static_assert((TZv * TY * XT) % NW == 0, "tiles must split evenly");
 
     __shared__ float slab[SZ * SY * SX * PC];
     // TAPG taps x NT 16-wide K tiles, each a [C][16] block (the 16-dword
@@ -65,13 +65,13 @@ __global__ __launch_bounds__(256, 1) void k_conv3(
     const int bx = blockIdx.x;                  // x block
     const int by = blockIdx.y;                  // y block
     const int bzn = blockIdx.z;                 // fused (n, z-block)
-    const int zblocks = (D + TZ - 1) / TZ;
+    const int zblocks = (D + TZv - 1) / TZv;
     const int n = bzn / zblocks;
-    const int z0 = (bzn % zblocks) * TZ;
+    const int z0 = (bzn % zblocks) * TZv;
     const int y0 = by * TY;
     const int x0 = bx * TX;
 
-    const int tid = threadIdx.x;                // 0..255
+    const int tid = threadIdx.x;
     const int wave = tid >> 6;
     const int lane = tid & 63;
 
@@ -84,14 +84,14 @@ __global__ __launch_bounds__(256, 1) void k_conv3(
     {
         const int c4n = C / 4;
         const int vox = SZ * SY * SX;
-        const bool interior = z0 >= 1 && z0 + TZ + 1 <= D && y0 >= 1 &&
+        const bool interior = z0 >= 1 && z0 + TZv + 1 <= D && y0 >= 1 &&
                               y0 + TY + 1 <= H && x0 >= 1 &&
                               x0 + TX + 1 <= W;
         const float* base =
             in + ((((long long)n * D + z0 - 1) * H + y0 - 1) * W + x0 - 1)
                  * C;
         if (interior) {
-            for (int idx = tid; idx < vox * c4n; idx += 256) {
+            for (int idx = tid; idx < vox * c4n; idx += NTHREADS) {
                 const int c4 = idx % c4n;
                 const int v = idx / c4n;
                 const int sx = v % SX;
@@ -104,7 +104,7 @@ __global__ __launch_bounds__(256, 1) void k_conv3(
                     &slab[((sz * SY + sy) * SX + sx) * PC + c4 * 4]) = val;
             }
         } else {
-            for (int idx = tid; idx < vox * c4n; idx += 256) {
+            for (int idx = tid; idx < vox * c4n; idx += NTHREADS) {
                 const int c4 = idx % c4n;
                 const int v = idx / c4n;
                 const int sx = v % SX;
@@ -129,11 +129,15 @@ __global__ __launch_bounds__(256, 1) void k_conv3(
     }
     __syncthreads();
 
-    // ---- wave tile origin ----------------------------------------------
-    // TY==4: wave w -> y=w, covers z0..z0+1 x 32x  (M_TILES = 4)
-    // TY==2: wave w -> (z = w>>1, y = w&1), covers 32x (M_TILES = 2)
-    const int wy = TY == 4 ? wave : (wave & 1);
-    const int wz0 = TY == 4 ? 0 : (wave >> 1);
+    // ---- wave tile origins (generic round-robin deal) -------------------
+    int tmz[M_TILES], tmy[M_TILES], tmx[M_TILES];
+#pragma unroll
+    for (int m = 0; m < M_TILES; ++m) {
+        const int g = wave * M_TILES + m;
+        tmx[m] = (g % XT) * 16;
+        tmy[m] = (g / XT) % TY;
+        tmz[m] = g / (XT * TY);
+    }
 
     f32x4 acc[M_TILES][NT];
 #pragma unroll
@@ -150,7 +154,8 @@ __global__ __launch_bounds__(256, 1) void k_conv3(
     static_assert(27 % TAPG == 0, "TAPG must divide 27");
     for (int g = 0; g < 27 / TAPG; ++g) {
         // stage the group's weights: TAPG x NT blocks of [C][16]
-        for (int idx = tid; idx < TAPG * NT * C * 16; idx += 256) {
+        for (int idx = tid; idx < TAPG * NT * C * 16;
+             idx += NTHREADS) {
             const int j = idx & 15;
             const int c = (idx >> 4) % C;
             const int nt = (idx >> 4) / C % NT;
@@ -173,11 +178,9 @@ __global__ __launch_bounds__(256, 1) void k_conv3(
             const float* arow[M_TILES];
 #pragma unroll
             for (int m = 0; m < M_TILES; ++m) {
-                const int mz = TY == 4 ? (m >> 1) : wz0;
-                const int mx = TY == 4 ? (m & 1) * 16 : m * 16;
-                arow[m] = &slab[(((1 + mz + dz) * SY +
-                                  (1 + wy + dy)) * SX +
-                                 (1 + mx + dx)) * PC + a_row * PC + a_k];
+                arow[m] = &slab[(((1 + tmz[m] + dz) * SY +
+                                  (1 + tmy[m] + dy)) * SX +
+                                 (1 + tmx[m] + dx)) * PC + a_row * PC + a_k];
             }
             const float* wblk = &wtile[tl * NT * C * 16];
 #pragma unroll
@@ -204,10 +207,8 @@ __global__ __launch_bounds__(256, 1) void k_conv3(
     const int rbase = (lane >> 4) * 4;
 #pragma unroll
     for (int m = 0; m < M_TILES; ++m) {
-        const int mz = TY == 4 ? (m >> 1) : wz0;
-        const int mx = TY == 4 ? (m & 1) * 16 : m * 16;
-        const int gz = z0 + mz;
-        const int gy = y0 + wy;
+        const int gz = z0 + tmz[m];
+        const int gy = y0 + tmy[m];
         if (gz >= D || gy >= H) continue;
 #pragma unroll
         for (int t = 0; t < NT; ++t) {
@@ -216,7 +217,7 @@ __global__ __launch_bounds__(256, 1) void k_conv3(
             const float bj = bias ? bias[j] : 0.f;
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
-                const int gx = x0 + mx + rbase + r;
+                const int gx = x0 + tmx[m] + rbase + r;
                 if (gx >= W) continue;
                 long long o =
                     ((((long long)n * D + gz) * H + gy) * W + gx) * K + j;
@@ -240,32 +241,34 @@ extern "C" int cfx_conv3_ndhwc(cfx_ctx* ctx, const float* in,
         g_err = "cfx_conv3_ndhwc: only C == K widths are instantiated";
         return -1;
     }
-    const int zblocks = (D + TZ - 1) / TZ;
-    dim3 block(256);
-    // LDS budget per width (slab + wtile <= 160 KiB):
-    //   28: 91.4 + 9 taps * 32.3/9 ... = 91.4 + 28.7 = 120 KiB (TAPG 9)
-    //   36: 117.5 + 3 * 6.9 = 138 KiB (TAPG 3)
-    //   48: 113.2 + 3 * 9.2 = 141 KiB (TAPG 3, TY 2)
-    //   64: TX=16 slab 78.3 + 16.4 = 95 KiB (TAPG 1, TY 2)
-#define CFX_CONV_CASE(CW, TYV, TXV, TAPGV)                                   \
+    // LDS budget per width (slab + wtile <= 160 KiB); NTHREADS=512 puts
+    // 2 waves per SIMD where the slab allows it
+#define CFX_CONV_CASE(CW, TZV, TYV, TXV, TAPGV, NTH)                         \
     case CW: {                                                               \
+        const int zb = (D + TZV - 1) / TZV;                                  \
         dim3 grid((W + TXV - 1) / TXV, (H + TYV - 1) / TYV,                  \
-                  (unsigned)(N * zblocks));                                  \
+                  (unsigned)(N * zb));                                       \
         hipEvent_t e0;                                                       \
         if (prof_begin(ctx, &e0)) return -1;                                 \
-        hipLaunchKernelGGL((k_conv3<CW, CW, TYV, TXV, TAPGV>), grid, block,  \
-                           0, ctx->stream, in, wgt, bias, residual, out, N,  \
-                           D, H, W, do_elu);                                 \
+        hipLaunchKernelGGL((k_conv3<CW, CW, TZV, TYV, TXV, TAPGV, NTH>),     \
+                           grid, dim3(NTH), 0, ctx->stream, in, wgt, bias,   \
+                           residual, out, N, D, H, W, do_elu);               \
         CFX_CHECK(hipGetLastError());                                        \
         double flops = 2.0 * 27.0 * CW * CW * (double)N * D * H * W;         \
         if (prof_end(ctx, e0, CFX_K_CONV, flops)) return -1;                 \
         break;                                                               \
     }
     switch (C) {
-        CFX_CONV_CASE(28, 4, 32, 9)
-        CFX_CONV_CASE(36, 4, 32, 3)
-        CFX_CONV_CASE(48, 2, 32, 3)
-        CFX_CONV_CASE(64, 2, 16, 1)
+        // 28: slab 3*10*34*28*4 = 114 KB + wtile 10.5 KB (TZ1 TY8, 8 waves)
+        CFX_CONV_CASE(28, 1, 8, 32, 3, 512)
+        // 36: slab 3*10*34*36*4 = 147 KB + wtile 4.5 KB (TAPG 1)
+        CFX_CONV_CASE(36, 1, 8, 32, 1, 512)
+        // 48: slab 3*6*34*52*4 = 127 KB + wtile 9.2 KB (TZ1 TY4, 8 waves
+        // -> 1 m-tile per wave)
+        CFX_CONV_CASE(48, 1, 4, 32, 1, 512)
+        // 64: slab 3*6*18*68*4 = 86 KB + wtile 16.4 KB (TZ1 TY4 TX16;
+        // 8 waves x ... 1*4*1 = 4 tiles < 8 waves -> use 256 threads)
+        CFX_CONV_CASE(64, 2, 2, 16, 1, 256)
         default:
             g_err = "cfx_conv3_ndhwc: unsupported channel width";
             return -1;
