@@ -276,3 +276,156 @@ extern "C" int cfx_conv3_ndhwc(cfx_ctx* ctx, const float* in,
 #undef CFX_CONV_CASE
     return 0;
 }
+
+namespace {
+
+// 32x32x2 variant (C == K <= 32): one 32-x-position x 32-channel D tile
+// per wave, single accumulator chain (the 32x32x2 dependent latency equals
+// its 64-cycle issue interval, so one chain sustains the full rate), and
+// one A + one B LDS dword per 4096 flops (3x fewer LDS reads per flop than
+// the 16x16x4 path). Slab layout [z][y][c][x] (x fastest) makes the
+// A-fragment read (lanes 0..31 = consecutive x) trivially conflict-free.
+template <int C, int K, int TY, int TAPG>
+__global__ __launch_bounds__(512, 1) void k_conv3_w32(
+    const float* __restrict__ in, const float* __restrict__ wgt,
+    const float* __restrict__ bias, const float* __restrict__ res,
+    float* __restrict__ out, int N, int D, int H, int W, int do_elu) {
+    constexpr int TX = 32;
+    constexpr int SX = TX + 2;
+    constexpr int SY = TY + 2;
+    constexpr int SZ = 3;                // TZ = 1
+    constexpr int KK = C / 2;            // reduction pairs per tap
+    static_assert(K <= 32 && C % 2 == 0, "");
+
+    __shared__ float slab[SZ * SY * C * SX];
+    __shared__ float wtile[TAPG * C * 32];
+
+    const int zblocks = D;               // TZ = 1
+    const int n = blockIdx.z / zblocks;
+    const int z0 = blockIdx.z % zblocks;
+    const int y0 = blockIdx.y * TY;
+    const int x0 = blockIdx.x * TX;
+    const int tid = threadIdx.x;
+    const int wave = tid >> 6;           // 0..7 -> y row
+    const int lane = tid & 63;
+
+    // ---- stage the slab: read NDHWC float4 runs, scatter to [c][x] ------
+    {
+        const int c4n = C / 4;
+        const bool interior = z0 >= 1 && z0 + 2 <= D && y0 >= 1 &&
+                              y0 + TY + 1 <= H && x0 >= 1 &&
+                              x0 + TX + 1 <= W;
+        for (int idx = tid; idx < SZ * SY * SX * c4n; idx += 512) {
+            const int c4 = idx % c4n;
+            const int v = idx / c4n;
+            const int sx = v % SX;
+            const int sy = (v / SX) % SY;
+            const int sz = v / (SX * SY);
+            const int gz = z0 + sz - 1;
+            const int gy = y0 + sy - 1;
+            const int gx = x0 + sx - 1;
+            f32x4 val;
+            if (interior) {
+                val = *reinterpret_cast<const f32x4*>(
+                    in + ((((long long)n * D + gz) * H + gy) * W + gx) * C +
+                    c4 * 4);
+            } else {
+                const bool ok = gz >= 0 && gz < D && gy >= 0 && gy < H &&
+                                gx >= 0 && gx < W;
+                val = *reinterpret_cast<const f32x4*>(
+                    in + ((((long long)n * D + (ok ? gz : 0)) * H +
+                           (ok ? gy : 0)) * W + (ok ? gx : 0)) * C +
+                    c4 * 4);
+                if (!ok) val = {0.f, 0.f, 0.f, 0.f};
+            }
+            float* base = &slab[((sz * SY + sy) * C + c4 * 4) * SX + sx];
+            base[0 * SX] = val.x;
+            base[1 * SX] = val.y;
+            base[2 * SX] = val.z;
+            base[3 * SX] = val.w;
+        }
+    }
+    __syncthreads();
+
+    typedef float f32x16 __attribute__((ext_vector_type(16)));
+    f32x16 acc = {};
+    const int ax = lane & 31;            // A row (x), B col (j)
+    const int ak = lane >> 5;            // reduction sub-index (0..1)
+
+    static_assert(27 % TAPG == 0, "");
+    for (int g = 0; g < 27 / TAPG; ++g) {
+        for (int idx = tid; idx < TAPG * C * 32; idx += 512) {
+            const int j = idx & 31;
+            const int c = (idx >> 5) % C;
+            const int tl = (idx >> 5) / C;
+            wtile[idx] = j < K
+                ? wgt[((long long)(g * TAPG + tl) * C + c) * K + j]
+                : 0.f;
+        }
+        __syncthreads();
+#pragma unroll
+        for (int tl = 0; tl < TAPG; ++tl) {
+            const int tap = g * TAPG + tl;
+            const int dz = tap / 9 - 1;
+            const int dy = (tap / 3) % 3 - 1;
+            const int dx = tap % 3 - 1;
+            const float* arow =
+                &slab[((1 + dz) * SY + (1 + wave + dy)) * C * SX +
+                      (1 + dx) + ax];
+            const float* wrow = &wtile[tl * C * 32 + ax];
+#pragma unroll
+            for (int kk = 0; kk < KK; ++kk) {
+                const float a = arow[(kk * 2 + ak) * SX];
+                const float b = wrow[(kk * 2 + ak) * 32];
+                acc = __builtin_amdgcn_mfma_f32_32x32x2f32(a, b, acc, 0, 0,
+                                                           0);
+            }
+        }
+        __syncthreads();
+    }
+
+    // ---- epilogue: D[row = (reg&3) + 8*(reg>>2) + 4*(lane>>5)][col] -----
+    const int gz = z0;
+    const int gy = y0 + wave;
+    const int j = lane & 31;
+    if (gy < H && j < K) {
+        const float bj = bias ? bias[j] : 0.f;
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+            const int row = (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
+            const int gx = x0 + row;
+            if (gx >= W) continue;
+            long long o =
+                ((((long long)n * D + gz) * H + gy) * W + gx) * K + j;
+            float v = acc[r] + bj;
+            if (res) v += res[o];
+            if (do_elu) v = v > 0.f ? v : expm1f(v);
+            out[o] = v;
+        }
+    }
+}
+
+}  // namespace
+
+// 32x32x2 path for C == K <= 32 (currently instantiated for 28); falls
+// back to the 16x16x4 path when CFX_CONV_W32=0
+extern "C" int cfx_conv3_ndhwc_w32(cfx_ctx* ctx, const float* in,
+                                   const float* wgt, const float* bias,
+                                   const float* residual, float* out, int N,
+                                   int D, int H, int W, int C, int K,
+                                   int do_elu) {
+    if (C != 28 || K != 28) {
+        g_err = "cfx_conv3_ndhwc_w32: only C == K == 28 instantiated";
+        return -1;
+    }
+    dim3 grid((W + 31) / 32, (H + 7) / 8, (unsigned)(N * D));
+    hipEvent_t e0;
+    if (prof_begin(ctx, &e0)) return -1;
+    hipLaunchKernelGGL((k_conv3_w32<28, 28, 8, 3>), grid, dim3(512), 0,
+                       ctx->stream, in, wgt, bias, residual, out, N, D, H,
+                       W, do_elu);
+    CFX_CHECK(hipGetLastError());
+    double flops = 2.0 * 27.0 * 28 * 28 * (double)N * D * H * W;
+    if (prof_end(ctx, e0, CFX_K_CONV, flops)) return -1;
+    return 0;
+}

@@ -11,7 +11,7 @@ from chunkflow_amd.fastconv import get_cfx
 torch.backends.cudnn.benchmark = True
 cl = torch.channels_last_3d
 
-def probe(C, D, H, W, N=12, iters=10):
+def probe(C, D, H, W, N=12, iters=10, w32=False):
     torch.manual_seed(0)
     x = torch.randn(N, C, D, H, W, device='cuda').contiguous(memory_format=cl)
     wt = torch.randn(C, C, 3, 3, 3, device='cuda') * (1.0 / (27 * C) ** 0.5)
@@ -24,7 +24,7 @@ def probe(C, D, H, W, N=12, iters=10):
     ref = F.conv3d(x.double().cpu(), wt.double().cpu(), bias.double().cpu(),
                    padding=1)[:2]
     cfx.conv3_ndhwc(x.data_ptr(), wtap.data_ptr(), bias.data_ptr(), None,
-                    out.data_ptr(), N, D, H, W, C, C)
+                    out.data_ptr(), N, D, H, W, C, C, w32=w32)
     torch.cuda.synchronize()
     err = (out[:2].double().cpu() - ref).abs().max().item()
     scale = ref.abs().max().item()
@@ -33,7 +33,7 @@ def probe(C, D, H, W, N=12, iters=10):
         t0 = time.perf_counter()
         for _ in range(iters):
             cfx.conv3_ndhwc(x.data_ptr(), wtap.data_ptr(), bias.data_ptr(),
-                            None, out.data_ptr(), N, D, H, W, C, C)
+                            None, out.data_ptr(), N, D, H, W, C, C, w32=w32)
         torch.cuda.synchronize()
         return (time.perf_counter() - t0) / iters
 
@@ -49,13 +49,14 @@ def probe(C, D, H, W, N=12, iters=10):
         t_mine(); t_torch()
     tm, tt = t_mine(), t_torch()
     flops = 2.0 * 27 * C * C * N * D * H * W
-    print({'C': C, 'shape': (N, D, H, W), 'err_vs_fp64': err,
+    print({'C': C, 'w32': w32, 'shape': (N, D, H, W), 'err_vs_fp64': err,
            'rel': err / scale,
            'mine_ms': tm * 1e3, 'torch_ms': tt * 1e3,
            'mine_TF': flops / tm / 1e12, 'torch_TF': flops / tt / 1e12,
            'speedup': tt / tm}, flush=True)
 
 probe(28, 20, 256, 256)
+probe(28, 20, 256, 256, w32=True)
 probe(36, 20, 128, 128)
 probe(48, 20, 64, 64)
 probe(64, 20, 32, 32)
