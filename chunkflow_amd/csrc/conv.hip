@@ -258,6 +258,26 @@ extern "C" int cfx_conv3_ndhwc(cfx_ctx* ctx, const float* in,
         if (prof_end(ctx, e0, CFX_K_CONV, flops)) return -1;                 \
         break;                                                               \
     }
+    // CFX_CONV_V=1 selects the half-slab TX=16 shape for C=28: 71 KB of
+    // LDS admits TWO 512-thread workgroups per CU (4 waves/SIMD), so one
+    // workgroup's slab staging overlaps the other's MFMAs
+    static int conv_v = [] {
+        const char* s = getenv("CFX_CONV_V");
+        return s ? atoi(s) : 0;
+    }();
+    if (C == 28 && conv_v == 1) {
+        // half-slab TX=16 shape: 71 KB LDS -> 2 workgroups per CU
+        dim3 grid((W + 15) / 16, (H + 7) / 8, (unsigned)(N * D));
+        hipEvent_t e0;
+        if (prof_begin(ctx, &e0)) return -1;
+        hipLaunchKernelGGL((k_conv3<28, 28, 1, 8, 16, 3, 512>), grid,
+                           dim3(512), 0, ctx->stream, in, wgt, bias,
+                           residual, out, N, D, H, W, do_elu);
+        CFX_CHECK(hipGetLastError());
+        double flops = 2.0 * 27.0 * 28 * 28 * (double)N * D * H * W;
+        if (prof_end(ctx, e0, CFX_K_CONV, flops)) return -1;
+        return 0;
+    }
     switch (C) {
         // 28: slab 3*10*34*28*4 = 114 KB + wtile 10.5 KB (TZ1 TY8, 8 waves)
         CFX_CONV_CASE(28, 1, 8, 32, 3, 512)
