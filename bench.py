@@ -33,7 +33,7 @@ import torch
 REPO = os.path.dirname(os.path.abspath(__file__))
 sys.path.insert(0, REPO)
 
-from chunkflow_amd.cartesian import BoundingBoxes, Cartesian
+from chunkflow_amd.cartesian import BoundingBoxes
 from chunkflow_amd.chunk import Chunk
 from chunkflow_amd.dispatch import init_distributed, stitch_to_rank0
 from chunkflow_amd.inferencer import Inferencer

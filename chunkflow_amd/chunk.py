@@ -11,9 +11,6 @@ host methods below are the reference semantics at the pipeline boundary.
 """
 from __future__ import annotations
 
-from numbers import Number
-from typing import Union
-
 import numpy as np
 
 from .cartesian import BoundingBox, Cartesian

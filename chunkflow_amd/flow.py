@@ -24,8 +24,8 @@ import numpy as np
 from .cartesian import BoundingBox, BoundingBoxes, Cartesian
 from .chunk import Chunk
 from .plugin import Plugin
-from .runtime import (CartesianParam, DEFAULT_CHUNK_NAME, default_none,
-                      generator, get_initial_task, main, operator, state)
+from .runtime import (DEFAULT_CHUNK_NAME, default_none, generator,
+                      get_initial_task, main, operator, state)
 
 
 @main.command('generate-tasks')

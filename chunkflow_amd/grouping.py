@@ -15,7 +15,7 @@ only — <= a few ulps over the <= 8 overlapping bump-weighted terms, orders
 of magnitude inside the 1e-5/1e-4 parity gates (the exact-order
 alternative degenerates to one group per patch along overlap chains).
 """
-from typing import List, Tuple
+from typing import List
 
 import numpy as np
 

@@ -11,8 +11,7 @@ import zlib
 import numpy as np
 import pytest
 
-from oracle import (make_patch_mask, patch_slices_list, oracle_inference,
-                    identity_engine)
+from oracle import make_patch_mask, oracle_inference, patch_slices_list
 
 
 def crc(arr):

@@ -35,7 +35,6 @@ def run_case(name, fg_np, connectivity=6):
             print(f'   vox ({z},{y},{x}) got={got[z,y,x]} ref={ref[z,y,x]} '
                   f'fg={fg_np[z,y,x]}')
         # partition check ignoring numbering
-        from collections import defaultdict
         m = {}
         consistent = True
         for g, r in zip(got[fg_np > 0].ravel(), ref[fg_np > 0].ravel()):

@@ -2,7 +2,6 @@
 check numerics vs a float64 CPU reference."""
 import os, sys, time
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
-import numpy as np
 import torch
 import torch.nn.functional as F
 
