@@ -265,6 +265,34 @@ extern "C" int cfx_conv3_ndhwc(cfx_ctx* ctx, const float* in,
         const char* s = getenv("CFX_CONV_V");
         return s ? atoi(s) : 0;
     }();
+    if (C == 28 && conv_v == 2) {
+        // 3 workgroups/CU: 47 KB LDS (256 thr, TZ1 TY4 TX16)
+        dim3 grid((W + 15) / 16, (H + 3) / 4, (unsigned)(N * D));
+        hipEvent_t e0;
+        if (prof_begin(ctx, &e0)) return -1;
+        hipLaunchKernelGGL((k_conv3<28, 28, 1, 4, 16, 3, 256>), grid,
+                           dim3(256), 0, ctx->stream, in, wgt, bias,
+                           residual, out, N, D, H, W, do_elu);
+        CFX_CHECK(hipGetLastError());
+        double flops = 2.0 * 27.0 * 28 * 28 * (double)N * D * H * W;
+        if (prof_end(ctx, e0, CFX_K_CONV, flops)) return -1;
+        return 0;
+    }
+    if (C == 28 && conv_v == 3) {
+        // 2 x 384-thread workgroups/CU (3 waves/SIMD), TZ2 TY6 TX16:
+        // slab 4*8*18*28*4 = 64.5 KB + wtile 10.5 KB = 75 KB
+        const int zb = (D + 1) / 2;
+        dim3 grid((W + 15) / 16, (H + 5) / 6, (unsigned)(N * zb));
+        hipEvent_t e0;
+        if (prof_begin(ctx, &e0)) return -1;
+        hipLaunchKernelGGL((k_conv3<28, 28, 2, 6, 16, 3, 384>), grid,
+                           dim3(384), 0, ctx->stream, in, wgt, bias,
+                           residual, out, N, D, H, W, do_elu);
+        CFX_CHECK(hipGetLastError());
+        double flops = 2.0 * 27.0 * 28 * 28 * (double)N * D * H * W;
+        if (prof_end(ctx, e0, CFX_K_CONV, flops)) return -1;
+        return 0;
+    }
     if (C == 28 && conv_v == 1) {
         // half-slab TX=16 shape: 71 KB LDS -> 2 workgroups per CU
         dim3 grid((W + 15) / 16, (H + 7) / 8, (unsigned)(N * D));
