@@ -632,3 +632,17 @@ class TestFastConv:
         monkeypatch.setenv('CFX_FASTCONV', '1')
         fast = run()
         np.testing.assert_allclose(fast, base, rtol=1e-4, atol=1e-4)
+
+
+def test_augment_identity_roundtrip_gpu():
+    """--augment on the HIP path (TTA forward/backward on device)."""
+    from chunkflow_amd.chunk import Chunk
+    rng = np.random.RandomState(77)
+    arr = rng.randint(0, 256, size=(16, 40, 40), dtype=np.uint8)
+    kw = dict(overlap=(2, 8, 8), framework='identity',
+              num_output_channels=2, batch_size=3, mask_output_chunk=True)
+    plain = _hip_inferencer(patch_size=(8, 24, 24), **kw)(Chunk(arr.copy()))
+    aug = _hip_inferencer(patch_size=(8, 24, 24), augment=True,
+                          **kw)(Chunk(arr.copy()))
+    np.testing.assert_allclose(aug.numpy().array, plain.numpy().array,
+                               rtol=1e-5, atol=1e-6)

@@ -292,3 +292,21 @@ def test_normalize_contrast_product_vs_oracle():
         # --whole is a no-op, like the reference
         same = normalize_contrast(Chunk(arr.copy()), per_section=False)
         np.testing.assert_array_equal(same.numpy().array, arr)
+
+
+@pytest.mark.skipif(GPU, reason='CPU plumbing path is refused on a GPU box')
+def test_augment_identity_roundtrip():
+    """--augment with the identity engine (reference
+    test_test_time_augmentation): the 8 TTA variants average back to the
+    un-augmented result (square patches; the reference's flips act on
+    batch/channel axes — transform.py docstring)."""
+    from chunkflow_amd.inferencer import Inferencer
+    rng = np.random.RandomState(77)
+    arr = rng.randint(0, 256, size=(16, 40, 40), dtype=np.uint8)
+    kw = dict(output_patch_overlap=(2, 8, 8), framework='identity',
+              num_output_channels=2, batch_size=3, mask_output_chunk=True)
+    plain = Inferencer(None, None, (8, 24, 24), **kw)(Chunk(arr.copy()))
+    aug = Inferencer(None, None, (8, 24, 24), augment=True,
+                     **kw)(Chunk(arr.copy()))
+    np.testing.assert_allclose(aug.numpy().array, plain.numpy().array,
+                               rtol=1e-5, atol=1e-6)
