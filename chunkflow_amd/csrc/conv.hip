@@ -477,3 +477,193 @@ extern "C" int cfx_conv3_ndhwc_w32(cfx_ctx* ctx, const float* in,
     if (prof_end(ctx, e0, CFX_K_CONV, flops)) return -1;
     return 0;
 }
+
+namespace {
+
+// Persistent-z variant: the workgroup walks ALL z planes of its (y, x)
+// tile. Weights for all 27 taps stay LDS-RESIDENT for the whole workgroup
+// (staged once, amortized over D planes — per-tap-group re-staging was
+// most of the per-z overhead), and the input slab is a 3-plane ring with
+// ONE ~20 KB plane staged per z. LDS: ring 3 x SY x SX x C (60.5 KB at
+// C=28, TY=8, TX=16) + weights 27 x NT x C x 16 (96.8 KB) = 157 KB.
+template <int C, int K, int TY, int TX>
+__global__ __launch_bounds__(512, 1) void k_conv3_zring(
+    const float* __restrict__ in, const float* __restrict__ wgt,
+    const float* __restrict__ bias, const float* __restrict__ res,
+    float* __restrict__ out, int N, int D, int H, int W, int do_elu) {
+    constexpr int PC = padc(C);
+    constexpr int SX = TX + 2;
+    constexpr int SY = TY + 2;
+    constexpr int KK = C / 4;
+    constexpr int NT = (K + 15) / 16;
+    constexpr int NW = 8;                 // 512 threads
+    constexpr int XT = TX / 16;
+    constexpr int M_TILES = (TY * XT) / NW;
+    static_assert((TY * XT) % NW == 0, "");
+
+    __shared__ float ring[3 * SY * SX * PC];
+    __shared__ float wall[27 * NT * C * 16];
+
+    const int n = blockIdx.z;
+    const int y0 = blockIdx.y * TY;
+    const int x0 = blockIdx.x * TX;
+    const int tid = threadIdx.x;
+    const int wave = tid >> 6;
+    const int lane = tid & 63;
+
+    // ---- weights: all taps, staged once -------------------------------
+    for (int idx = tid; idx < 27 * NT * C * 16; idx += 512) {
+        const int j = idx & 15;
+        const int c = (idx >> 4) % C;
+        const int nt = (idx >> 4) / C % NT;
+        const int tap = (idx >> 4) / C / NT;
+        const int jg = nt * 16 + j;
+        wall[idx] =
+            jg < K ? wgt[((long long)tap * C + c) * K + jg] : 0.f;
+    }
+
+    // plane stager: global plane P (may be -1 or D: zero padding) into
+    // ring slot (P + 1) % 3
+    auto stage_plane = [&](int P) {
+        const int slot = ((P + 1) % 3 + 3) % 3;
+        const int c4n = C / 4;
+        const bool zin = P >= 0 && P < D;
+        const bool interior = zin && y0 >= 1 && y0 + TY + 1 <= H &&
+                              x0 >= 1 && x0 + TX + 1 <= W;
+        for (int idx = tid; idx < SY * SX * c4n; idx += 512) {
+            const int c4 = idx % c4n;
+            const int v = idx / c4n;
+            const int sx = v % SX;
+            const int sy = v / SX;
+            const int gy = y0 + sy - 1;
+            const int gx = x0 + sx - 1;
+            f32x4 val;
+            if (interior) {
+                val = *reinterpret_cast<const f32x4*>(
+                    in + ((((long long)n * D + P) * H + gy) * W + gx) * C +
+                    c4 * 4);
+            } else {
+                const bool ok = zin && gy >= 0 && gy < H && gx >= 0 &&
+                                gx < W;
+                val = *reinterpret_cast<const f32x4*>(
+                    in + ((((long long)n * D + (zin ? P : 0)) * H +
+                           (ok ? gy : 0)) * W + (ok ? gx : 0)) * C +
+                    c4 * 4);
+                if (!ok) val = {0.f, 0.f, 0.f, 0.f};
+            }
+            *reinterpret_cast<f32x4*>(
+                &ring[(slot * SY * SX + v) * PC + c4 * 4]) = val;
+        }
+    };
+
+    stage_plane(-1);
+    stage_plane(0);
+
+    int tmy[M_TILES], tmx[M_TILES];
+#pragma unroll
+    for (int m = 0; m < M_TILES; ++m) {
+        const int g = wave * M_TILES + m;
+        tmx[m] = (g % XT) * 16;
+        tmy[m] = g / XT;
+    }
+    const int a_row = lane & 15;
+    const int a_k = lane >> 4;
+    const int col16 = lane & 15;
+    const int colj = lane & 15;
+    const int rbase = (lane >> 4) * 4;
+
+    for (int z = 0; z < D; ++z) {
+        stage_plane(z + 1);
+        __syncthreads();
+
+        f32x4 acc[M_TILES][NT];
+#pragma unroll
+        for (int m = 0; m < M_TILES; ++m)
+#pragma unroll
+            for (int t = 0; t < NT; ++t)
+                acc[m][t] = {0.f, 0.f, 0.f, 0.f};
+
+        for (int dzi = 0; dzi < 3; ++dzi) {
+            const int slot = ((z + dzi) % 3 + 3) % 3;  // plane z + dzi - 1
+            const float* plane = &ring[slot * SY * SX * PC];
+#pragma unroll
+            for (int tl = 0; tl < 9; ++tl) {
+                const int tap = dzi * 9 + tl;
+                const int dy = tl / 3 - 1;
+                const int dx = tl % 3 - 1;
+                const float* arow[M_TILES];
+#pragma unroll
+                for (int m = 0; m < M_TILES; ++m) {
+                    arow[m] = &plane[((1 + tmy[m] + dy) * SX +
+                                      (1 + tmx[m] + dx)) * PC +
+                                     a_row * PC + a_k];
+                }
+                const float* wblk = &wall[tap * NT * C * 16];
+#pragma unroll
+                for (int nt = 0; nt < NT; ++nt) {
+#pragma unroll
+                    for (int kk = 0; kk < KK; ++kk) {
+                        const float b =
+                            wblk[(nt * C + kk * 4 + a_k) * 16 + col16];
+#pragma unroll
+                        for (int m = 0; m < M_TILES; ++m) {
+                            const float a = arow[m][kk * 4];
+                            acc[m][nt] =
+                                __builtin_amdgcn_mfma_f32_16x16x4f32(
+                                    a, b, acc[m][nt], 0, 0, 0);
+                        }
+                    }
+                }
+            }
+        }
+
+        // epilogue for this z
+#pragma unroll
+        for (int m = 0; m < M_TILES; ++m) {
+            const int gy = y0 + tmy[m];
+            if (gy >= H) continue;
+#pragma unroll
+            for (int t = 0; t < NT; ++t) {
+                const int j = t * 16 + colj;
+                if (j >= K) continue;
+                const float bj = bias ? bias[j] : 0.f;
+#pragma unroll
+                for (int r = 0; r < 4; ++r) {
+                    const int gx = x0 + tmx[m] + rbase + r;
+                    if (gx >= W) continue;
+                    long long o =
+                        ((((long long)n * D + z) * H + gy) * W + gx) * K +
+                        j;
+                    float v = acc[m][t][r] + bj;
+                    if (res) v += res[o];
+                    if (do_elu) v = v > 0.f ? v : expm1f(v);
+                    out[o] = v;
+                }
+            }
+        }
+        __syncthreads();  // before the next z overwrites the oldest plane
+    }
+}
+
+}  // namespace
+
+extern "C" int cfx_conv3_ndhwc_zring(cfx_ctx* ctx, const float* in,
+                                     const float* wgt, const float* bias,
+                                     const float* residual, float* out,
+                                     int N, int D, int H, int W, int C,
+                                     int K, int do_elu) {
+    if (C != 28 || K != 28) {
+        g_err = "cfx_conv3_ndhwc_zring: only C == K == 28 instantiated";
+        return -1;
+    }
+    dim3 grid((W + 15) / 16, (H + 7) / 8, (unsigned)N);
+    hipEvent_t e0;
+    if (prof_begin(ctx, &e0)) return -1;
+    hipLaunchKernelGGL((k_conv3_zring<28, 28, 8, 16>), grid, dim3(512), 0,
+                       ctx->stream, in, wgt, bias, residual, out, N, D, H,
+                       W, do_elu);
+    CFX_CHECK(hipGetLastError());
+    double flops = 2.0 * 27.0 * 28 * 28 * (double)N * D * H * W;
+    if (prof_end(ctx, e0, CFX_K_CONV, flops)) return -1;
+    return 0;
+}

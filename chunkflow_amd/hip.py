@@ -25,6 +25,7 @@ SYMBOLS = [
     'cfx_mask_using_last_channel', 'cfx_threshold', 'cfx_nonzero_u8',
     'cfx_connected_components', 'cfx_hist_u8', 'cfx_lut_apply_u8',
     'cfx_conv3_ndhwc', 'cfx_conv3_ndhwc_w32',
+    'cfx_conv3_ndhwc_zring',
     'cfx_profile_enable', 'cfx_profile_reset',
     'cfx_profile_get',
 ]
@@ -259,8 +260,11 @@ class CfxContext:
             ctypes.c_void_p(lut_ptr)), 'cfx_lut_apply_u8')
 
     def conv3_ndhwc(self, in_ptr, wgt_ptr, bias_ptr, residual_ptr, out_ptr,
-                    n, d, h, w, c, k, do_elu=False, w32=False):
-        fn = self.lib.cfx_conv3_ndhwc_w32 if w32 else self.lib.cfx_conv3_ndhwc
+                    n, d, h, w, c, k, do_elu=False, w32=False,
+                    zring=False):
+        fn = (self.lib.cfx_conv3_ndhwc_zring if zring
+              else self.lib.cfx_conv3_ndhwc_w32 if w32
+              else self.lib.cfx_conv3_ndhwc)
         self._chk(fn(
             ctypes.c_void_p(self.ctx), ctypes.c_void_p(in_ptr),
             ctypes.c_void_p(wgt_ptr),

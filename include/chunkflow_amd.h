@@ -137,6 +137,12 @@ int cfx_lut_apply_u8(cfx_ctx* ctx, unsigned char* buf, long long n_per_sec,
 int cfx_conv3_ndhwc(cfx_ctx* ctx, const float* in, const float* wgt,
                     const float* bias, const float* residual, float* out,
                     int N, int D, int H, int W, int C, int K, int do_elu);
+/* the persistent-z ring variant (weights LDS-resident, one input plane
+ * staged per z; C == K == 28 instantiated) */
+int cfx_conv3_ndhwc_zring(cfx_ctx* ctx, const float* in, const float* wgt,
+                          const float* bias, const float* residual,
+                          float* out, int N, int D, int H, int W, int C,
+                          int K, int do_elu);
 /* the 32x32x2-MFMA variant (C == K == 28 instantiated) */
 int cfx_conv3_ndhwc_w32(cfx_ctx* ctx, const float* in, const float* wgt,
                         const float* bias, const float* residual,

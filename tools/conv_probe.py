@@ -10,7 +10,7 @@ from chunkflow_amd.fastconv import get_cfx
 torch.backends.cudnn.benchmark = True
 cl = torch.channels_last_3d
 
-def probe(C, D, H, W, N=12, iters=10, w32=False):
+def probe(C, D, H, W, N=12, iters=10, w32=False, zring=False):
     torch.manual_seed(0)
     x = torch.randn(N, C, D, H, W, device='cuda').contiguous(memory_format=cl)
     wt = torch.randn(C, C, 3, 3, 3, device='cuda') * (1.0 / (27 * C) ** 0.5)
@@ -23,7 +23,7 @@ def probe(C, D, H, W, N=12, iters=10, w32=False):
     ref = F.conv3d(x.double().cpu(), wt.double().cpu(), bias.double().cpu(),
                    padding=1)[:2]
     cfx.conv3_ndhwc(x.data_ptr(), wtap.data_ptr(), bias.data_ptr(), None,
-                    out.data_ptr(), N, D, H, W, C, C, w32=w32)
+                    out.data_ptr(), N, D, H, W, C, C, w32=w32, zring=zring)
     torch.cuda.synchronize()
     err = (out[:2].double().cpu() - ref).abs().max().item()
     scale = ref.abs().max().item()
@@ -32,7 +32,7 @@ def probe(C, D, H, W, N=12, iters=10, w32=False):
         t0 = time.perf_counter()
         for _ in range(iters):
             cfx.conv3_ndhwc(x.data_ptr(), wtap.data_ptr(), bias.data_ptr(),
-                            None, out.data_ptr(), N, D, H, W, C, C, w32=w32)
+                            None, out.data_ptr(), N, D, H, W, C, C, w32=w32, zring=zring)
         torch.cuda.synchronize()
         return (time.perf_counter() - t0) / iters
 
@@ -48,7 +48,7 @@ def probe(C, D, H, W, N=12, iters=10, w32=False):
         t_mine(); t_torch()
     tm, tt = t_mine(), t_torch()
     flops = 2.0 * 27 * C * C * N * D * H * W
-    print({'C': C, 'w32': w32, 'shape': (N, D, H, W), 'err_vs_fp64': err,
+    print({'C': C, 'w32': w32, 'zring': zring, 'shape': (N, D, H, W), 'err_vs_fp64': err,
            'rel': err / scale,
            'mine_ms': tm * 1e3, 'torch_ms': tt * 1e3,
            'mine_TF': flops / tm / 1e12, 'torch_TF': flops / tt / 1e12,
@@ -56,6 +56,7 @@ def probe(C, D, H, W, N=12, iters=10, w32=False):
 
 probe(28, 20, 256, 256)
 probe(28, 20, 256, 256, w32=True)
+probe(28, 20, 256, 256, zring=True)
 probe(36, 20, 128, 128)
 probe(48, 20, 64, 64)
 probe(64, 20, 32, 32)
