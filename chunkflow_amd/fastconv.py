@@ -22,7 +22,11 @@ def get_cfx(device_index: int = 0):
     return _CTX[device_index]
 
 
-ELIGIBLE_WIDTHS = (28, 36, 48, 64)
+# widths where the hand kernel BEATS MIOpen (the persistent-z ring wins
+# C=28 at 88.6 vs 81.9 TF; the generic shapes still trail at 36/48/64, so
+# those stay on MIOpen — DESIGN.md §10 ladder)
+ELIGIBLE_WIDTHS = (28,)
+ZRING_WIDTHS = (28,)
 
 
 def _eligible(m: nn.Module) -> bool:
@@ -53,8 +57,7 @@ class CfxConv3d(nn.Module):
         else:
             self.bias = None
 
-    def forward(self, x):
-        assert x.dtype == torch.float32, 'fastconv is the f32 path'
+    def _run(self, x, residual=None, elu=False):
         x = x.contiguous(memory_format=torch.channels_last_3d)
         n, c, d, h, w = x.shape
         out = torch.empty((n, self.K, d, h, w), dtype=torch.float32,
@@ -63,17 +66,53 @@ class CfxConv3d(nn.Module):
         get_cfx(self.device_index).conv3_ndhwc(
             x.data_ptr(), self.wtap.data_ptr(),
             self.bias.data_ptr() if self.bias is not None else None,
-            None, out.data_ptr(), n, d, h, w, self.C, self.K)
+            residual.data_ptr() if residual is not None else None,
+            out.data_ptr(), n, d, h, w, self.C, self.K, do_elu=elu,
+            zring=self.C in ZRING_WIDTHS)
         return out
+
+    def forward(self, x):
+        assert x.dtype == torch.float32, 'fastconv is the f32 path'
+        return self._run(x)
+
+
+class CfxResBlock(nn.Module):
+    """Fused replacement for the RSUNet ResBlock pattern (conv1 -> ELU ->
+    conv2 -> +x -> ELU): the ELU and residual-add run in the conv
+    epilogues, removing three full-tensor elementwise passes per block."""
+
+    def __init__(self, block: nn.Module, device_index: int = 0):
+        super().__init__()
+        self.c1 = CfxConv3d(block.conv1, device_index)
+        self.c2 = CfxConv3d(block.conv2, device_index)
+
+    def forward(self, x):
+        x = x.contiguous(memory_format=torch.channels_last_3d)
+        h = self.c1._run(x, elu=True)
+        return self.c2._run(h, residual=x, elu=True)
+
+
+def _resblock_like(m: nn.Module) -> bool:
+    return (hasattr(m, 'conv1') and hasattr(m, 'conv2')
+            and isinstance(getattr(m, 'act', None), nn.ELU)
+            and getattr(m.act, 'alpha', None) == 1.0
+            and _eligible(m.conv1) and _eligible(m.conv2)
+            and type(m).forward.__qualname__.startswith('ResBlock'))
 
 
 def maybe_accelerate(model: nn.Module, device_index: int = 0) -> int:
-    """Replace eligible convs in-place; returns the replacement count."""
+    """Replace eligible ResBlocks (fused) and lone convs in-place;
+    returns the replacement count."""
     count = 0
+    dev = f'cuda:{device_index}'
     for parent in model.modules():
         for name, child in list(parent.named_children()):
-            if _eligible(child):
-                setattr(parent, name, CfxConv3d(child, device_index)
-                        .to(f'cuda:{device_index}'))
+            if _resblock_like(child):
+                setattr(parent, name,
+                        CfxResBlock(child, device_index).to(dev))
+                count += 1
+            elif _eligible(child):
+                setattr(parent, name,
+                        CfxConv3d(child, device_index).to(dev))
                 count += 1
     return count

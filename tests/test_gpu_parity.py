@@ -646,3 +646,27 @@ def test_augment_identity_roundtrip_gpu():
                           **kw)(Chunk(arr.copy()))
     np.testing.assert_allclose(aug.numpy().array, plain.numpy().array,
                                rtol=1e-5, atol=1e-6)
+
+
+def test_zring_parity_and_fusion():
+    """Persistent-z ring conv (the variant that beats MIOpen at C=28):
+    parity incl. fused ELU+residual epilogue."""
+    import torch.nn.functional as F
+    from chunkflow_amd.fastconv import get_cfx
+    cl = torch.channels_last_3d
+    torch.manual_seed(5)
+    n, C, d, h, w = 2, 28, 9, 37, 41
+    x = torch.randn(n, C, d, h, w, device='cuda').contiguous(
+        memory_format=cl)
+    wt = torch.randn(C, C, 3, 3, 3, device='cuda') * 0.05
+    bias = torch.randn(C, device='cuda') * 0.1
+    wtap = wt.permute(2, 3, 4, 1, 0).reshape(27, C, C).contiguous()
+    res = torch.randn_like(x).contiguous(memory_format=cl)
+    out = torch.empty_like(x)
+    get_cfx(0).conv3_ndhwc(x.data_ptr(), wtap.data_ptr(), bias.data_ptr(),
+                           res.data_ptr(), out.data_ptr(), n, d, h, w, C,
+                           C, do_elu=True, zring=True)
+    ref = torch.nn.functional.elu(F.conv3d(x, wt, bias, padding=1) + res)
+    torch.cuda.synchronize()
+    np.testing.assert_allclose(out.cpu().numpy(), ref.cpu().numpy(),
+                               rtol=1e-4, atol=5e-5)
