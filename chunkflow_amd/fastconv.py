@@ -96,8 +96,20 @@ def _resblock_like(m: nn.Module) -> bool:
     return (hasattr(m, 'conv1') and hasattr(m, 'conv2')
             and isinstance(getattr(m, 'act', None), nn.ELU)
             and getattr(m.act, 'alpha', None) == 1.0
-            and _eligible(m.conv1) and _eligible(m.conv2)
-            and type(m).forward.__qualname__.startswith('ResBlock'))
+            and not isinstance(m, (CfxConv3d, CfxResBlock))
+            and _eligible(m.conv1) and _eligible(m.conv2))
+
+
+@torch.no_grad()
+def _fusion_matches(block: nn.Module, fused: 'CfxResBlock') -> bool:
+    """Functional probe at surgery time: the fused block must reproduce the
+    original forward on a random input (guards against user modules that
+    merely LOOK like a ResBlock)."""
+    C = fused.c1.C
+    x = torch.randn(1, C, 4, 18, 22, device=fused.c1.wtap.device)         .contiguous(memory_format=torch.channels_last_3d)
+    want = block.to(x.device)(x)
+    got = fused(x)
+    return bool(torch.allclose(got, want, rtol=1e-4, atol=1e-4))
 
 
 def maybe_accelerate(model: nn.Module, device_index: int = 0) -> int:
@@ -108,9 +120,17 @@ def maybe_accelerate(model: nn.Module, device_index: int = 0) -> int:
     for parent in model.modules():
         for name, child in list(parent.named_children()):
             if _resblock_like(child):
-                setattr(parent, name,
-                        CfxResBlock(child, device_index).to(dev))
-                count += 1
+                fused = CfxResBlock(child, device_index).to(dev)
+                if _fusion_matches(child, fused):
+                    setattr(parent, name, fused)
+                    count += 1
+                    continue
+                # forward does something else: fall through to lone convs
+                for cn in ('conv1', 'conv2'):
+                    setattr(child, cn,
+                            CfxConv3d(getattr(child, cn),
+                                      device_index).to(dev))
+                    count += 1
             elif _eligible(child):
                 setattr(parent, name,
                         CfxConv3d(child, device_index).to(dev))
