@@ -54,7 +54,7 @@ class PyTorchEngine(EngineBase):
             # swap eligible 3x3x3 ResBlock convs for the hand-written MFMA
             # kernel (fastconv.py); CFX_FASTCONV=0 keeps MIOpen everywhere
             if (self.dtype != 'bfloat16'
-                    and os.environ.get('CFX_FASTCONV', '0') == '1'):
+                    and os.environ.get('CFX_FASTCONV', '1') != '0'):
                 from ..fastconv import maybe_accelerate
                 idx = int(str(self.device).split(':')[-1])                     if ':' in str(self.device) else 0
                 self.fastconv_count = maybe_accelerate(self.model, idx)
