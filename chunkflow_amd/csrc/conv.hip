@@ -486,16 +486,17 @@ namespace {
 // most of the per-z overhead), and the input slab is a 3-plane ring with
 // ONE ~20 KB plane staged per z. LDS: ring 3 x SY x SX x C (60.5 KB at
 // C=28, TY=8, TX=16) + weights 27 x NT x C x 16 (96.8 KB) = 157 KB.
-template <int C, int K, int TY, int TX>
+template <int C, int K, int TY, int TX, int NTK>
 __global__ __launch_bounds__(512, 1) void k_conv3_zring(
     const float* __restrict__ in, const float* __restrict__ wgt,
     const float* __restrict__ bias, const float* __restrict__ res,
-    float* __restrict__ out, int N, int D, int H, int W, int do_elu) {
+    float* __restrict__ out, int N, int D, int H, int W, int do_elu,
+    int j0) {
     constexpr int PC = padc(C);
     constexpr int SX = TX + 2;
     constexpr int SY = TY + 2;
     constexpr int KK = C / 4;
-    constexpr int NT = (K + 15) / 16;
+    constexpr int NT = NTK;               // K tiles computed this launch
     constexpr int NW = 8;                 // 512 threads
     constexpr int XT = TX / 16;
     constexpr int M_TILES = (TY * XT) / NW;
@@ -517,7 +518,7 @@ __global__ __launch_bounds__(512, 1) void k_conv3_zring(
         const int c = (idx >> 4) % C;
         const int nt = (idx >> 4) / C % NT;
         const int tap = (idx >> 4) / C / NT;
-        const int jg = nt * 16 + j;
+        const int jg = j0 + nt * 16 + j;
         wall[idx] =
             jg < K ? wgt[((long long)tap * C + c) * K + jg] : 0.f;
     }
@@ -624,7 +625,7 @@ __global__ __launch_bounds__(512, 1) void k_conv3_zring(
             if (gy >= H) continue;
 #pragma unroll
             for (int t = 0; t < NT; ++t) {
-                const int j = t * 16 + colj;
+                const int j = j0 + t * 16 + colj;
                 if (j >= K) continue;
                 const float bj = bias ? bias[j] : 0.f;
 #pragma unroll
@@ -652,18 +653,29 @@ extern "C" int cfx_conv3_ndhwc_zring(cfx_ctx* ctx, const float* in,
                                      const float* residual, float* out,
                                      int N, int D, int H, int W, int C,
                                      int K, int do_elu) {
-    if (C != 28 || K != 28) {
-        g_err = "cfx_conv3_ndhwc_zring: only C == K == 28 instantiated";
-        return -1;
-    }
     dim3 grid((W + 15) / 16, (H + 7) / 8, (unsigned)N);
     hipEvent_t e0;
     if (prof_begin(ctx, &e0)) return -1;
-    hipLaunchKernelGGL((k_conv3_zring<28, 28, 8, 16>), grid, dim3(512), 0,
-                       ctx->stream, in, wgt, bias, residual, out, N, D, H,
-                       W, do_elu);
+    if (C == 28 && K == 28) {
+        // wall holds both K tiles (96.8 KB + 60.5 KB ring)
+        hipLaunchKernelGGL((k_conv3_zring<28, 28, 8, 16, 2>), grid,
+                           dim3(512), 0, ctx->stream, in, wgt, bias,
+                           residual, out, N, D, H, W, do_elu, 0);
+    } else if (C == 36 && K == 36) {
+        // one 16-wide K tile per launch (wall 62.2 KB + ring 77.8 KB);
+        // the input re-read costs ~3x HBM traffic of a 113 MB activation
+        // per conv — negligible next to the compute
+        for (int j0 = 0; j0 < 36; j0 += 16) {
+            hipLaunchKernelGGL((k_conv3_zring<36, 36, 8, 16, 1>), grid,
+                               dim3(512), 0, ctx->stream, in, wgt, bias,
+                               residual, out, N, D, H, W, do_elu, j0);
+        }
+    } else {
+        g_err = "cfx_conv3_ndhwc_zring: width not instantiated";
+        return -1;
+    }
     CFX_CHECK(hipGetLastError());
-    double flops = 2.0 * 27.0 * 28 * 28 * (double)N * D * H * W;
+    double flops = 2.0 * 27.0 * C * K * (double)N * D * H * W;
     if (prof_end(ctx, e0, CFX_K_CONV, flops)) return -1;
     return 0;
 }

@@ -57,6 +57,7 @@ def probe(C, D, H, W, N=12, iters=10, w32=False, zring=False):
 probe(28, 20, 256, 256)
 probe(28, 20, 256, 256, w32=True)
 probe(28, 20, 256, 256, zring=True)
+probe(36, 20, 128, 128, zring=True)
 probe(36, 20, 128, 128)
 probe(48, 20, 64, 64)
 probe(64, 20, 32, 32)
