@@ -22,11 +22,12 @@ def get_cfx(device_index: int = 0):
     return _CTX[device_index]
 
 
-# widths where the hand kernel BEATS MIOpen (the persistent-z ring wins
-# C=28 at 88.6 vs 81.9 TF; the generic shapes still trail at 36/48/64, so
-# those stay on MIOpen — DESIGN.md §10 ladder)
-ELIGIBLE_WIDTHS = (28,)
-ZRING_WIDTHS = (28,)
+# widths where the hand kernel BEATS MIOpen (the persistent-z ring:
+# C=28 88.6 vs 81.9 TF, C=36 69.5 vs 64.7; 48/64 stay on MIOpen — the
+# weight wall no longer fits LDS beside the ring at those widths, and
+# their spatial extents are small — DESIGN.md §10 ladder)
+ELIGIBLE_WIDTHS = (28, 36)
+ZRING_WIDTHS = (28, 36)
 
 
 def _eligible(m: nn.Module) -> bool:
