@@ -310,3 +310,28 @@ def test_augment_identity_roundtrip():
                      **kw)(Chunk(arr.copy()))
     np.testing.assert_allclose(aug.numpy().array, plain.numpy().array,
                                rtol=1e-5, atol=1e-6)
+
+
+def test_fastconv_eligibility_rules():
+    """Graph-surgery eligibility (no GPU needed): only stride-1 pad-1 3^3
+    convs at the widths where the hand kernel beats MIOpen."""
+    import torch.nn as nn
+    from chunkflow_amd.fastconv import _eligible, _resblock_like
+    assert _eligible(nn.Conv3d(28, 28, 3, padding=1))
+    assert _eligible(nn.Conv3d(36, 36, 3, padding=1))
+    assert not _eligible(nn.Conv3d(48, 48, 3, padding=1))  # MIOpen faster
+    assert not _eligible(nn.Conv3d(28, 36, 3, padding=1))  # C != K
+    assert not _eligible(nn.Conv3d(28, 28, 3, padding=1, stride=2))
+    assert not _eligible(nn.Conv3d(28, 28, (1, 5, 5), padding=(0, 2, 2)))
+    assert not _eligible(nn.Conv3d(28, 28, 3, padding=1, groups=28))
+
+    import importlib.util
+    spec = importlib.util.spec_from_file_location(
+        'rsunet_t', os.path.join(
+            os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+            'examples', 'nets', 'rsunet.py'))
+    mod = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(mod)
+    blocks = [m for m in mod.InstantiatedModel.modules()
+              if _resblock_like(m)]
+    assert len(blocks) == 4  # the two 28- and two 36-wide ResBlocks
