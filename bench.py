@@ -209,6 +209,17 @@ def main():
         elapsed = float(t.item())
 
     blend = cfx.profile_get('blend')
+    # the hand MFMA conv kernels (fastconv) publish FLOPs in the bytes slot
+    conv_ctx = None
+    if args.engine == 'rsunet' and args.dtype == 'float32':
+        try:
+            from chunkflow_amd.fastconv import _CTX
+            conv_ctx = _CTX.get(local_rank)
+        except Exception:
+            conv_ctx = None
+    conv = None
+    if conv_ctx is not None:
+        conv = conv_ctx.profile_get('conv')
     cfx.profile_enable(False)
 
     if rank != 0:
@@ -220,6 +231,15 @@ def main():
     value = voxels / elapsed
 
     traffic = read_pmc_traffic()
+    conv_roofline = None
+    if conv and conv['count'] > 0 and conv['total_ms'] > 0:
+        tflops = conv['bytes'] / (conv['total_ms'] / 1e3) / 1e12
+        conv_roofline = {
+            'bound': 'mfma', 'achieved_tflops': tflops,
+            'peak_tflops': 157.3,  # f32 MFMA peak (MI355X_MICROARCH.md)
+            'frac': tflops / 157.3, 'launches': conv['count'],
+            'kernel': 'k_conv3_zring (fastconv, widths 28/36)',
+        }
     roofline = None
     if blend['count'] > 0 and blend['total_ms'] > 0:
         achieved = blend['bytes'] / (blend['total_ms'] / 1e3)
@@ -234,6 +254,7 @@ def main():
             'launches': blend['count'],
             'algorithmic_bytes_per_launch': blend['bytes'] / blend['count'],
             'avg_launch_ms': blend['total_ms'] / blend['count'],
+            'conv': conv_roofline,
         }
 
     cpu = None
