@@ -196,6 +196,17 @@ def main():
     cfx = inferencer.ops.cfx
     cfx.profile_reset()
     cfx.profile_enable(True)
+    # the fastconv kernels run on their own context; profile it too
+    conv_ctx = None
+    if args.engine == 'rsunet' and args.dtype == 'float32':
+        try:
+            from chunkflow_amd.fastconv import _CTX
+            conv_ctx = _CTX.get(local_rank)
+        except Exception:
+            conv_ctx = None
+    if conv_ctx is not None:
+        conv_ctx.profile_reset()
+        conv_ctx.profile_enable(True)
 
     barrier_sync()
     t0 = time.perf_counter()
@@ -210,16 +221,10 @@ def main():
 
     blend = cfx.profile_get('blend')
     # the hand MFMA conv kernels (fastconv) publish FLOPs in the bytes slot
-    conv_ctx = None
-    if args.engine == 'rsunet' and args.dtype == 'float32':
-        try:
-            from chunkflow_amd.fastconv import _CTX
-            conv_ctx = _CTX.get(local_rank)
-        except Exception:
-            conv_ctx = None
     conv = None
     if conv_ctx is not None:
         conv = conv_ctx.profile_get('conv')
+        conv_ctx.profile_enable(False)
     cfx.profile_enable(False)
 
     if rank != 0:
