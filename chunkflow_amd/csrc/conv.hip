@@ -679,3 +679,169 @@ extern "C" int cfx_conv3_ndhwc_zring(cfx_ctx* ctx, const float* in,
     if (prof_end(ctx, e0, CFX_K_CONV, flops)) return -1;
     return 0;
 }
+
+namespace {
+
+// bf16 persistent-z ring (config-5 path): v_mfma_f32_32x32x16_bf16 — one
+// 32-x by 32-channel D tile per wave, K-reduction 16 input channels per
+// instruction, fragment layout probe-verified on hardware
+// (tools/mfma_layout_probe.py): A[i=l&31][k=8*(l>>5)+e], B likewise, D as
+// the f32 32x32 map. Activations ride the 3-plane LDS ring in [x][c]
+// order (lane reads its 8-channel A fragment as one 16-byte ds_read; the
+// per-x stride is padded to 40 elements so the 16-lane b128 groups land
+// on distinct banks); the bf16 weight wall (27 taps x 32 j x padded c)
+// stays LDS-resident for the whole workgroup. C is zero-padded to 32.
+typedef __bf16 cfx_bf16;
+typedef cfx_bf16 bf16x8 __attribute__((ext_vector_type(8)));
+typedef float f32x16 __attribute__((ext_vector_type(16)));
+
+template <int C, int K, int TY, int TX>
+__global__ __launch_bounds__(512, 1) void k_conv3_zring_bf16(
+    const cfx_bf16* __restrict__ in,    // (N, D, H, W, C)
+    const cfx_bf16* __restrict__ wgt,   // (27, 32, 32): [tap][j][c], padded
+    const float* __restrict__ bias,     // (K) f32
+    const cfx_bf16* __restrict__ res,   // optional residual (out layout)
+    cfx_bf16* __restrict__ out,         // (N, D, H, W, K)
+    int N, int D, int H, int W, int do_elu) {
+    constexpr int CP = 32;               // padded channel count
+    constexpr int PCB = CP + 8;          // slab per-x stride (bank spread)
+    constexpr int SX = TX + 2;
+    constexpr int SY = TY + 2;
+    constexpr int KK = CP / 16;          // 2 reduction steps per tap
+    static_assert(C <= CP && K <= 32, "");
+
+    __shared__ cfx_bf16 ring[3 * SY * SX * PCB];
+    __shared__ cfx_bf16 wall[27 * 32 * PCB];
+
+    const int n = blockIdx.z;
+    const int y0 = blockIdx.y * TY;
+    const int x0 = blockIdx.x * TX;
+    const int tid = threadIdx.x;
+    const int wave = tid >> 6;           // 0..7 -> y row (TY == 8)
+    const int lane = tid & 63;
+
+    // weight wall, staged once: wall[(tap*32 + j)*PCB + c]
+    for (int idx = tid; idx < 27 * 32 * CP; idx += 512) {
+        const int c = idx % CP;
+        const int j = (idx / CP) % 32;
+        const int tap = idx / (CP * 32);
+        wall[(tap * 32 + j) * PCB + c] = wgt[(tap * 32 + j) * 32 + c];
+    }
+
+    auto stage_plane = [&](int P) {
+        const int slot = ((P + 1) % 3 + 3) % 3;
+        const bool zin = P >= 0 && P < D;
+        const bool interior = zin && y0 >= 1 && y0 + TY + 1 <= H &&
+                              x0 >= 1 && x0 + TX + 1 <= W;
+        // copy C bf16 per voxel as 4-channel chunks (C % 4 == 0), zero the
+        // pad channels once via the c4 == C/4 slot
+        constexpr int C4 = CP / 4;
+        typedef cfx_bf16 bf16x4 __attribute__((ext_vector_type(4)));
+        for (int idx = tid; idx < SY * SX * C4; idx += 512) {
+            const int c4 = idx % C4;
+            const int v = idx / C4;
+            const int sx = v % SX;
+            const int sy = v / SX;
+            const int gy = y0 + sy - 1;
+            const int gx = x0 + sx - 1;
+            bf16x4 val = {};
+            if (c4 * 4 < C) {
+                if (interior) {
+                    val = *reinterpret_cast<const bf16x4*>(
+                        in + ((((long long)n * D + P) * H + gy) * W + gx)
+                                 * C + c4 * 4);
+                } else {
+                    const bool ok = zin && gy >= 0 && gy < H && gx >= 0 &&
+                                    gx < W;
+                    val = *reinterpret_cast<const bf16x4*>(
+                        in + ((((long long)n * D + (zin ? P : 0)) * H +
+                               (ok ? gy : 0)) * W + (ok ? gx : 0)) * C +
+                        c4 * 4);
+                    if (!ok) val = {};
+                }
+            }
+            *reinterpret_cast<bf16x4*>(
+                &ring[((slot * SY + sy) * SX + sx) * PCB + c4 * 4]) = val;
+        }
+    };
+
+    stage_plane(-1);
+    stage_plane(0);
+
+    const int ax = lane & 31;            // A row (x), B col (j)
+    const int khalf = (lane >> 5) * 8;   // this lane's k sub-range base
+
+    for (int z = 0; z < D; ++z) {
+        stage_plane(z + 1);
+        __syncthreads();
+
+        f32x16 acc = {};
+        for (int dzi = 0; dzi < 3; ++dzi) {
+            const int slot = ((z + dzi) % 3 + 3) % 3;
+            const cfx_bf16* plane = &ring[slot * SY * SX * PCB];
+#pragma unroll
+            for (int tl = 0; tl < 9; ++tl) {
+                const int tap = dzi * 9 + tl;
+                const int dy = tl / 3 - 1;
+                const int dx = tl % 3 - 1;
+                const cfx_bf16* arow =
+                    &plane[((1 + wave + dy) * SX + (1 + dx) + ax) * PCB +
+                           khalf];
+                const cfx_bf16* brow = &wall[(tap * 32 + ax) * PCB + khalf];
+#pragma unroll
+                for (int kk = 0; kk < KK; ++kk) {
+                    const bf16x8 a =
+                        *reinterpret_cast<const bf16x8*>(arow + kk * 16);
+                    const bf16x8 b =
+                        *reinterpret_cast<const bf16x8*>(brow + kk * 16);
+                    acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, acc,
+                                                                  0, 0, 0);
+                }
+            }
+        }
+
+        const int gy = y0 + wave;
+        const int j = lane & 31;
+        if (gy < H && j < K) {
+            const float bj = bias ? bias[j] : 0.f;
+#pragma unroll
+            for (int r = 0; r < 16; ++r) {
+                const int row = (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
+                const int gx = x0 + row;
+                if (gx >= W) continue;
+                long long o =
+                    ((((long long)n * D + z) * H + gy) * W + gx) * K + j;
+                float v = acc[r] + bj;
+                if (res) v += (float)res[o];
+                if (do_elu) v = v > 0.f ? v : expm1f(v);
+                out[o] = (cfx_bf16)v;
+            }
+        }
+        __syncthreads();
+    }
+}
+
+}  // namespace
+
+extern "C" int cfx_conv3_ndhwc_bf16(cfx_ctx* ctx, const void* in,
+                                    const void* wgt, const float* bias,
+                                    const void* residual, void* out, int N,
+                                    int D, int H, int W, int C, int K,
+                                    int do_elu) {
+    if (C != 28 || K != 28) {
+        g_err = "cfx_conv3_ndhwc_bf16: only C == K == 28 instantiated";
+        return -1;
+    }
+    dim3 grid((W + 31) / 32, (H + 7) / 8, (unsigned)N);
+    hipEvent_t e0;
+    if (prof_begin(ctx, &e0)) return -1;
+    hipLaunchKernelGGL((k_conv3_zring_bf16<28, 28, 8, 32>), grid, dim3(512),
+                       0, ctx->stream, (const cfx_bf16*)in,
+                       (const cfx_bf16*)wgt, bias,
+                       (const cfx_bf16*)residual, (cfx_bf16*)out, N, D, H,
+                       W, do_elu);
+    CFX_CHECK(hipGetLastError());
+    double flops = 2.0 * 27.0 * 28 * 28 * (double)N * D * H * W;
+    if (prof_end(ctx, e0, CFX_K_CONV, flops)) return -1;
+    return 0;
+}

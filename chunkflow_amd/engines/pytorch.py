@@ -52,12 +52,16 @@ class PyTorchEngine(EngineBase):
         if self.channels_last:
             self.model = self.model.to(memory_format=torch.channels_last_3d)
             # swap eligible 3x3x3 ResBlock convs for the hand-written MFMA
-            # kernel (fastconv.py); CFX_FASTCONV=0 keeps MIOpen everywhere
-            if (self.dtype != 'bfloat16'
-                    and os.environ.get('CFX_FASTCONV', '1') != '0'):
-                from ..fastconv import maybe_accelerate
+            # kernels (fastconv.py); CFX_FASTCONV=0 keeps MIOpen everywhere
+            if os.environ.get('CFX_FASTCONV', '1') != '0':
                 idx = int(str(self.device).split(':')[-1])                     if ':' in str(self.device) else 0
-                self.fastconv_count = maybe_accelerate(self.model, idx)
+                if self.dtype == 'bfloat16':
+                    from ..fastconv import maybe_accelerate_bf16
+                    self.fastconv_count = maybe_accelerate_bf16(
+                        self.model, idx)
+                else:
+                    from ..fastconv import maybe_accelerate
+                    self.fastconv_count = maybe_accelerate(self.model, idx)
         torch.backends.cudnn.benchmark = True
         self.pre_process = getattr(net_source, 'pre_process', None)
         self.post_process = getattr(net_source, 'post_process', None)

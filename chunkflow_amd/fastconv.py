@@ -137,3 +137,105 @@ def maybe_accelerate(model: nn.Module, device_index: int = 0) -> int:
                         CfxConv3d(child, device_index).to(dev))
                 count += 1
     return count
+
+
+BF16_WIDTHS = (28,)
+
+
+class CfxConv3dBF16(nn.Module):
+    """bf16 drop-in for an eligible nn.Conv3d (config-5 path): the bf16
+    persistent-z ring kernel on v_mfma_f32_32x32x16_bf16."""
+
+    def __init__(self, conv: nn.Conv3d, device_index: int = 0):
+        super().__init__()
+        self.C = conv.in_channels
+        self.K = conv.out_channels
+        self.device_index = device_index
+        w = conv.weight.detach().float()  # (K, C, 3, 3, 3)
+        pack = torch.zeros(27, 32, 32)
+        pack[:, :self.K, :self.C] = w.permute(2, 3, 4, 0, 1) \
+            .reshape(27, self.K, self.C)
+        self.register_buffer('wpack', pack.to(torch.bfloat16).contiguous())
+        if conv.bias is not None:
+            self.register_buffer('bias', conv.bias.detach().float())
+        else:
+            self.bias = None
+
+    def _run(self, x, residual=None, elu=False):
+        assert x.dtype == torch.bfloat16
+        x = x.contiguous(memory_format=torch.channels_last_3d)
+        n, c, d, h, w = x.shape
+        out = torch.empty((n, self.K, d, h, w), dtype=torch.bfloat16,
+                          device=x.device,
+                          memory_format=torch.channels_last_3d)
+        get_cfx(self.device_index).conv3_ndhwc_bf16(
+            x.data_ptr(), self.wpack.data_ptr(),
+            self.bias.data_ptr() if self.bias is not None else None,
+            residual.data_ptr() if residual is not None else None,
+            out.data_ptr(), n, d, h, w, self.C, self.K, do_elu=elu)
+        return out
+
+    def forward(self, x):
+        return self._run(x)
+
+
+class CfxResBlockBF16(nn.Module):
+    def __init__(self, block: nn.Module, device_index: int = 0):
+        super().__init__()
+        self.c1 = CfxConv3dBF16(block.conv1, device_index)
+        self.c2 = CfxConv3dBF16(block.conv2, device_index)
+
+    def forward(self, x):
+        x = x.contiguous(memory_format=torch.channels_last_3d)
+        h = self.c1._run(x, elu=True)
+        return self.c2._run(h, residual=x, elu=True)
+
+
+def maybe_accelerate_bf16(model: nn.Module, device_index: int = 0) -> int:
+    """bf16 surgery (widths where the bf16 ring is instantiated); same
+    functional-probe guard as the f32 path."""
+    count = 0
+    dev = f'cuda:{device_index}'
+
+    def elig(m):
+        return (isinstance(m, nn.Conv3d) and m.kernel_size == (3, 3, 3)
+                and m.stride == (1, 1, 1) and m.padding == (1, 1, 1)
+                and m.dilation == (1, 1, 1) and m.groups == 1
+                and m.in_channels == m.out_channels
+                and m.in_channels in BF16_WIDTHS)
+
+    def rb(m):
+        return (hasattr(m, 'conv1') and hasattr(m, 'conv2')
+                and isinstance(getattr(m, 'act', None), nn.ELU)
+                and getattr(m.act, 'alpha', None) == 1.0
+                and not isinstance(m, (CfxConv3dBF16, CfxResBlockBF16))
+                and elig(m.conv1) and elig(m.conv2))
+
+    @torch.no_grad()
+    def matches(block, fused):
+        x = torch.randn(1, fused.c1.C, 4, 18, 22, device=dev) \
+            .to(torch.bfloat16) \
+            .contiguous(memory_format=torch.channels_last_3d)
+        want = block.to(dev)(x)
+        got = fused(x)
+        return bool(torch.allclose(got.float(), want.float(), rtol=0.05,
+                                   atol=0.05))
+
+    for parent in model.modules():
+        for name, child in list(parent.named_children()):
+            if rb(child):
+                fused = CfxResBlockBF16(child, device_index).to(dev)
+                if matches(child, fused):
+                    setattr(parent, name, fused)
+                    count += 1
+                    continue
+                for cn in ('conv1', 'conv2'):
+                    setattr(child, cn,
+                            CfxConv3dBF16(getattr(child, cn),
+                                          device_index).to(dev))
+                    count += 1
+            elif elig(child):
+                setattr(parent, name,
+                        CfxConv3dBF16(child, device_index).to(dev))
+                count += 1
+    return count

@@ -25,7 +25,7 @@ SYMBOLS = [
     'cfx_mask_using_last_channel', 'cfx_threshold', 'cfx_nonzero_u8',
     'cfx_connected_components', 'cfx_hist_u8', 'cfx_lut_apply_u8',
     'cfx_conv3_ndhwc', 'cfx_conv3_ndhwc_w32',
-    'cfx_conv3_ndhwc_zring',
+    'cfx_conv3_ndhwc_zring', 'cfx_conv3_ndhwc_bf16',
     'cfx_profile_enable', 'cfx_profile_reset',
     'cfx_profile_get',
 ]
@@ -245,6 +245,18 @@ class CfxContext:
             ctypes.c_void_p(scratch_ptr), ctypes.byref(ncomp)),
             'cfx_connected_components')
         return ncomp.value
+
+    def conv3_ndhwc_bf16(self, in_ptr, wgt_ptr, bias_ptr, residual_ptr,
+                         out_ptr, n, d, h, w, c, k, do_elu=False):
+        self._chk(self.lib.cfx_conv3_ndhwc_bf16(
+            ctypes.c_void_p(self.ctx), ctypes.c_void_p(in_ptr),
+            ctypes.c_void_p(wgt_ptr),
+            ctypes.c_void_p(bias_ptr) if bias_ptr else None,
+            ctypes.c_void_p(residual_ptr) if residual_ptr else None,
+            ctypes.c_void_p(out_ptr), ctypes.c_int(n), ctypes.c_int(d),
+            ctypes.c_int(h), ctypes.c_int(w), ctypes.c_int(c),
+            ctypes.c_int(k), ctypes.c_int(1 if do_elu else 0)),
+            'cfx_conv3_ndhwc_bf16')
 
     # --- image normalization -------------------------------------------------
     def hist_u8(self, in_ptr, n_per_sec, nsec, hist_ptr):

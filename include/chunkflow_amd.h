@@ -143,6 +143,13 @@ int cfx_conv3_ndhwc_zring(cfx_ctx* ctx, const float* in, const float* wgt,
                           const float* bias, const float* residual,
                           float* out, int N, int D, int H, int W, int C,
                           int K, int do_elu);
+/* bf16 persistent-z ring (v_mfma_f32_32x32x16_bf16; C == K == 28):
+ * in/out/residual are bf16 NDHWC, wgt is a bf16 (27, 32, 32) [tap][j][c]
+ * zero-padded pack, bias stays f32; epilogue math in f32 */
+int cfx_conv3_ndhwc_bf16(cfx_ctx* ctx, const void* in, const void* wgt,
+                         const float* bias, const void* residual,
+                         void* out, int N, int D, int H, int W, int C,
+                         int K, int do_elu);
 /* the 32x32x2-MFMA variant (C == K == 28 instantiated) */
 int cfx_conv3_ndhwc_w32(cfx_ctx* ctx, const float* in, const float* wgt,
                         const float* bias, const float* residual,
