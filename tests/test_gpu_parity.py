@@ -670,3 +670,46 @@ def test_zring_parity_and_fusion():
     torch.cuda.synchronize()
     np.testing.assert_allclose(out.cpu().numpy(), ref.cpu().numpy(),
                                rtol=1e-4, atol=5e-5)
+
+
+def test_bf16_zring_parity():
+    """bf16 z-ring conv vs torch bf16 conv (both accumulate f32): agree
+    within bf16 rounding of the activations."""
+    import torch.nn.functional as F
+    from chunkflow_amd.fastconv import CfxConv3dBF16
+    cl = torch.channels_last_3d
+    torch.manual_seed(7)
+    conv = torch.nn.Conv3d(28, 28, 3, padding=1).cuda()
+    m = CfxConv3dBF16(conv).cuda()
+    x = (torch.randn(2, 28, 7, 33, 41, device='cuda') * 0.5) \
+        .to(torch.bfloat16).contiguous(memory_format=cl)
+    got = m(x).float()
+    ref = F.conv3d(x.float(), conv.weight.float()
+                   .to(torch.bfloat16).float(),
+                   conv.bias.float(), padding=1)
+    np.testing.assert_allclose(got.cpu().numpy(), ref.cpu().numpy(),
+                               rtol=0.05, atol=0.03)
+
+
+def test_bf16_engine_fastconv_close_to_f32(golden_dir):
+    """config-5 wiring: the bf16 engine with bf16 fastconv stays within
+    bf16 tolerance of the f32 MIOpen path on a small RSUNet run."""
+    from chunkflow_amd.chunk import Chunk
+    chunk = Chunk.create(size=(20, 128, 128), dtype='uint8', pattern='sin')
+    model = os.path.join(REPO, 'examples', 'nets', 'rsunet.py')
+
+    def run(dtype, fast):
+        os.environ['CFX_FASTCONV'] = fast
+        try:
+            inf = _hip_inferencer(model=model, framework='pytorch',
+                                  patch_size=(20, 64, 64),
+                                  overlap=(4, 16, 16), dtype=dtype,
+                                  num_output_channels=3, batch_size=2,
+                                  mask_output_chunk=True)
+            return inf(chunk).numpy().array
+        finally:
+            os.environ.pop('CFX_FASTCONV', None)
+
+    base = run('float32', '0')
+    fast_bf16 = run('bfloat16', '1')
+    np.testing.assert_allclose(fast_bf16, base, rtol=0.05, atol=0.03)

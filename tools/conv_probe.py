@@ -61,3 +61,40 @@ probe(36, 20, 128, 128, zring=True)
 probe(36, 20, 128, 128)
 probe(48, 20, 64, 64)
 probe(64, 20, 32, 32)
+
+
+def probe_bf16(C=28, D=20, H=256, W=256, N=24, iters=8):
+    from chunkflow_amd.fastconv import CfxConv3dBF16
+    torch.manual_seed(0)
+    conv = torch.nn.Conv3d(C, C, 3, padding=1).cuda()
+    m = CfxConv3dBF16(conv).cuda()
+    x = (torch.randn(N, C, D, H, W, device='cuda') * 0.3) \
+        .to(torch.bfloat16).contiguous(memory_format=cl)
+    convb = conv.to(torch.bfloat16).to(memory_format=cl)
+    got = m(x).float()
+    ref = torch.nn.functional.conv3d(x, convb.weight, convb.bias,
+                                     padding=1).float()
+    err = (got - ref).abs().max().item()
+
+    def t(f):
+        t0 = time.perf_counter()
+        for _ in range(iters):
+            f()
+        torch.cuda.synchronize()
+        return (time.perf_counter() - t0) / iters
+
+    for _ in range(3):
+        m(x); torch.nn.functional.conv3d(x, convb.weight, convb.bias,
+                                         padding=1)
+    tm = t(lambda: m(x))
+    tt = t(lambda: torch.nn.functional.conv3d(x, convb.weight, convb.bias,
+                                              padding=1))
+    flops = 2.0 * 27 * C * C * N * D * H * W
+    print({'bf16': True, 'C': C, 'shape': (N, D, H, W),
+           'err_vs_torch_bf16': err, 'mine_ms': tm * 1e3,
+           'torch_ms': tt * 1e3, 'mine_TF': flops / tm / 1e12,
+           'torch_TF': flops / tt / 1e12, 'speedup': tt / tm}, flush=True)
+
+
+if '--bf16' in sys.argv:
+    probe_bf16(28, 32, 256, 256, N=24)
