@@ -125,3 +125,29 @@ def test_skip_family_and_var_ops(tmp_path):
              'mark-complete', '-p', str(tmp_path / 'never_')])
     assert len(glob.glob(trace + '*')) == 1
     assert len(glob.glob(str(tmp_path / 'never_*'))) == 0
+
+
+@pytest.mark.skipif(GPU, reason='CPU plumbing pipeline')
+def test_universal_example_file(golden):
+    """The shipped universal example file works through the CLI (the
+    reference's examples/inference/universal_identity.py contract)."""
+    import os
+    _, arrays = golden
+    example = os.path.join(
+        os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+        'examples', 'inference', 'universal_identity.py')
+    import tempfile
+    with tempfile.TemporaryDirectory() as td:
+        out = os.path.join(td, 'u.npy')
+        run_cli(['create-chunk', '--size', '20', '68', '72',
+                 '--dtype', 'uint8', '--pattern', 'sin',
+                 'inference', '-m', example, '-s', '10', '32', '32',
+                 '--output-patch-overlap', '2', '8', '8',
+                 '--framework', 'universal', '--batch-size', '3',
+                 '--num-output-channels', '3', '--mask-output-chunk',
+                 'save-npy', '-f', out])
+        got = np.load(out)
+    from chunkflow_amd.chunk import Chunk as C
+    sin = C.create(size=(20, 68, 72), dtype='uint8', pattern='sin').array
+    np.testing.assert_allclose(got[0], sin.astype(np.float32) / 255.0,
+                               rtol=1e-5, atol=1e-5)
