@@ -118,7 +118,7 @@ def maybe_accelerate(model: nn.Module, device_index: int = 0) -> int:
     returns the replacement count."""
     count = 0
     dev = f'cuda:{device_index}'
-    for parent in model.modules():
+    for parent in list(model.modules()):  # snapshot: we mutate the tree
         for name, child in list(parent.named_children()):
             if _resblock_like(child):
                 fused = CfxResBlock(child, device_index).to(dev)
@@ -221,7 +221,7 @@ def maybe_accelerate_bf16(model: nn.Module, device_index: int = 0) -> int:
         return bool(torch.allclose(got.float(), want.float(), rtol=0.05,
                                    atol=0.05))
 
-    for parent in model.modules():
+    for parent in list(model.modules()):  # snapshot: we mutate the tree
         for name, child in list(parent.named_children()):
             if rb(child):
                 fused = CfxResBlockBF16(child, device_index).to(dev)

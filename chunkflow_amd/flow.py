@@ -11,8 +11,11 @@ operators (reference chunkflow/flow/flow.py):
   crop-margin           :2053-2084
   plugin                :1751-1800
   connected-components  :1803-1829 (host path; GPU kernel is a 'next' row)
+  normalize-contrast    :1672-1710 (device histogram/LUT kernels)
   log-summary           :1633-1647 (voxels/sec report shape)
-plus npy load/save at the pipeline edge.
+plus the at-least-once resume family (skip-task-by-file, mark-complete,
+skip-all-zero, skip-none — SURVEY.md §5), copy-var/delete-var, and npy
+load/save at the pipeline edge.
 """
 import json
 import os

@@ -42,14 +42,7 @@ def main():
     import torch
     here = '/tmp/mfma_probe'
     os.makedirs(here, exist_ok=True)
-    with open(f'{here}/probe.hip', 'w') as f:
-        f.write(SRC)
-    subprocess.run(['hipcc', '--offload-arch=gfx950', '-O2', '-fPIC',
-                    '-shared', f'{here}/probe.hip', '-o',
-                    f'{here}/probe.so'], check=True)
     import ctypes
-    lib = ctypes.CDLL(f'{here}/probe.so')
-
     rng = np.random.RandomState(0)
     A = (rng.randn(32, 16) * 0.5).astype(np.float32)
     B = (rng.randn(16, 32) * 0.5).astype(np.float32)
@@ -57,8 +50,6 @@ def main():
     tB = torch.from_numpy(B).to(torch.bfloat16).cuda()
     tD = torch.zeros(32, 32, dtype=torch.float32, device='cuda')
 
-    # launch via hipModule? simpler: use torch to drive the raw kernel via
-    # ctypes hipLaunchKernel is messy; embed a C launcher instead
     with open(f'{here}/launch.hip', 'w') as f:
         f.write(SRC + r'''
 extern "C" int run(const void* a, const void* b, void* d) {
