@@ -274,9 +274,30 @@ class Chunk:
     def max(self):
         return self.array.max()
 
-    # --- simple file IO at the pipeline edge (npy/npz) ---------------------
+    # --- simple file IO at the pipeline edge (npy/tif) ---------------------
     def to_npy(self, file_name: str):
         np.save(file_name, self.numpy().array)
+
+    def to_tif(self, file_name: str = None, compression: str = 'zlib'):
+        """Save as a (multi-page) TIFF via the in-repo codec, keeping the
+        reference's semantics (chunk/base.py:238-263) including the
+        float32 -> *255 uint8 visual-scaling quirk."""
+        from . import tiffio
+        if file_name is None:
+            file_name = f'{self.bbox.string}.tif'
+        tiffio.write_volume(file_name, self.numpy().array,
+                            compression=compression)
+
+    @classmethod
+    def from_tif(cls, file_name: str, voxel_offset=None, dtype=None,
+                 voxel_size=None):
+        """Load from a TIFF file or a directory of per-section TIFFs
+        (reference chunk/base.py:209-236)."""
+        from . import tiffio
+        arr = tiffio.read_volume(file_name, dtype=dtype)
+        return cls(arr,
+                   voxel_offset=voxel_offset or (0, 0, 0),
+                   voxel_size=voxel_size)
 
     @classmethod
     def from_npy(cls, file_name: str, voxel_offset=(0, 0, 0),

@@ -416,6 +416,62 @@ def load_npy(tasks, name, file_name, voxel_offset, output_chunk_name):
         yield task
 
 
+@main.command('load-tif')
+@click.option('--name', type=str, default='load-tif', help='name of operator')
+@click.option('--file-name', '-f', required=True,
+              type=click.Path(exists=True, dir_okay=True),
+              help='TIFF file, or a directory of per-section *.tif* files')
+@click.option('--voxel-offset', '-v', type=click.INT, nargs=3, default=None,
+              callback=default_none, help='global offset of this chunk')
+@click.option('--voxel-size', '-s', type=click.INT, nargs=3, default=None,
+              callback=default_none, help='physical voxel size (nm)')
+@click.option('--dtype', '-d', default=None,
+              type=click.Choice(['uint8', 'uint16', 'uint32', 'uint64',
+                                 'float32', 'float64', 'float16']),
+              help='convert to data type')
+@click.option('--output-chunk-name', '-o', type=str,
+              default=DEFAULT_CHUNK_NAME, help='output chunk name')
+@operator
+def load_tif(tasks, name, file_name, voxel_offset, voxel_size, dtype,
+             output_chunk_name):
+    """Read TIFF files (reference flow.py:918-950; in-repo codec)."""
+    for task in tasks:
+        if task is not None:
+            start = time()
+            task[output_chunk_name] = Chunk.from_tif(
+                file_name, dtype=dtype, voxel_offset=voxel_offset,
+                voxel_size=voxel_size)
+            task['log']['timer'][name] = time() - start
+        yield task
+
+
+@main.command('save-tif')
+@click.option('--name', type=str, default='save-tif', help='name of operator')
+@click.option('--input-chunk-name', '-i', type=str,
+              default=DEFAULT_CHUNK_NAME, help='input chunk name')
+@click.option('--file-name', '-f', default=None, type=str,
+              help='output .tif/.tiff path; default <bbox>.tif')
+@click.option('--dtype', '-t', default=None,
+              type=click.Choice(['uint8', 'uint16', 'uint32', 'uint64',
+                                 'float32', 'float64']),
+              help='convert to this data type before saving')
+@click.option('--compression', '-c', type=click.Choice(['', 'zlib']),
+              default='zlib',
+              help='codec (in-repo writer: raw or zlib/deflate)')
+@operator
+def save_tif(tasks, name, input_chunk_name, file_name, dtype, compression):
+    """Save chunk as a TIFF file (reference flow.py:953-974)."""
+    for task in tasks:
+        if task is not None:
+            start = time()
+            chunk = task[input_chunk_name]
+            if dtype:
+                chunk = chunk.astype(dtype)
+            chunk.to_tif(file_name, compression=compression)
+            task['log']['timer'][name] = time() - start
+        yield task
+
+
 @main.command('save-log')
 @click.option('--output-path', '-o', type=str, required=True,
               help='directory for per-task log JSON files')
@@ -597,3 +653,7 @@ def skip_none(tasks, input_name, touch, prefix, suffix):
                     Path(f"{prefix}{task['bbox'].string}{suffix}").touch()
                 task = None
         yield task
+
+
+if __name__ == '__main__':  # python -m chunkflow_amd.flow <ops...>
+    main()
