@@ -646,6 +646,192 @@ __global__ __launch_bounds__(512, 1) void k_conv3_zring(
     }
 }
 
+// Software-pipelined variant of k_conv3_zring: the NEXT plane's global
+// loads are issued into registers first, the 18 taps that only touch the
+// two already-resident planes run while those loads are in flight, then
+// the registers drain to the ring slot and ONE barrier releases the last
+// 9 taps. One barrier per z (the plain kernel needs two): iteration z's
+// store targets the slot plane z-2 occupied, whose last readers (dzi=0 of
+// iteration z-1) sit before iteration z-1's barrier.
+template <int C, int K, int TY, int TX, int NTK>
+__global__ __launch_bounds__(512, 1) void k_conv3_zring_pl(
+    const float* __restrict__ in, const float* __restrict__ wgt,
+    const float* __restrict__ bias, const float* __restrict__ res,
+    float* __restrict__ out, int N, int D, int H, int W, int do_elu,
+    int j0) {
+    constexpr int PC = padc(C);
+    constexpr int SX = TX + 2;
+    constexpr int SY = TY + 2;
+    constexpr int KK = C / 4;
+    constexpr int NT = NTK;
+    constexpr int NW = 8;
+    constexpr int XT = TX / 16;
+    constexpr int M_TILES = (TY * XT) / NW;
+    constexpr int C4N = C / 4;
+    constexpr int LV = (SY * SX * C4N + 511) / 512;  // loads held per thread
+    static_assert((TY * XT) % NW == 0, "");
+
+    __shared__ float ring[3 * SY * SX * PC];
+    __shared__ float wall[27 * NT * C * 16];
+
+    const int n = blockIdx.z;
+    const int y0 = blockIdx.y * TY;
+    const int x0 = blockIdx.x * TX;
+    const int tid = threadIdx.x;
+    const int wave = tid >> 6;
+    const int lane = tid & 63;
+
+    for (int idx = tid; idx < 27 * NT * C * 16; idx += 512) {
+        const int j = idx & 15;
+        const int c = (idx >> 4) % C;
+        const int nt = (idx >> 4) / C % NT;
+        const int tap = (idx >> 4) / C / NT;
+        const int jg = j0 + nt * 16 + j;
+        wall[idx] =
+            jg < K ? wgt[((long long)tap * C + c) * K + jg] : 0.f;
+    }
+
+    const bool xy_interior = y0 >= 1 && y0 + TY + 1 <= H && x0 >= 1 &&
+                             x0 + TX + 1 <= W;
+
+    // clamp-then-zero load of this thread's share of global plane P
+    auto plane_load = [&](int P, f32x4 (&vals)[LV]) {
+        const bool zin = P >= 0 && P < D;
+        const bool interior = zin && xy_interior;
+#pragma unroll
+        for (int li = 0; li < LV; ++li) {
+            const int idx = tid + li * 512;
+            if (idx >= SY * SX * C4N) break;
+            const int c4 = idx % C4N;
+            const int v = idx / C4N;
+            const int gy = y0 + v / SX - 1;
+            const int gx = x0 + v % SX - 1;
+            if (interior) {
+                vals[li] = *reinterpret_cast<const f32x4*>(
+                    in + ((((long long)n * D + P) * H + gy) * W + gx) * C +
+                    c4 * 4);
+            } else {
+                const bool ok = zin && gy >= 0 && gy < H && gx >= 0 &&
+                                gx < W;
+                vals[li] = *reinterpret_cast<const f32x4*>(
+                    in + ((((long long)n * D + (zin ? P : 0)) * H +
+                           (ok ? gy : 0)) * W + (ok ? gx : 0)) * C +
+                    c4 * 4);
+                if (!ok) vals[li] = {0.f, 0.f, 0.f, 0.f};
+            }
+        }
+    };
+    auto plane_store = [&](int P, const f32x4 (&vals)[LV]) {
+        const int slot = ((P + 1) % 3 + 3) % 3;
+#pragma unroll
+        for (int li = 0; li < LV; ++li) {
+            const int idx = tid + li * 512;
+            if (idx >= SY * SX * C4N) break;
+            const int c4 = idx % C4N;
+            const int v = idx / C4N;
+            *reinterpret_cast<f32x4*>(
+                &ring[(slot * SY * SX + v) * PC + c4 * 4]) = vals[li];
+        }
+    };
+
+    {
+        f32x4 v0[LV], v1[LV];
+        plane_load(-1, v0);
+        plane_load(0, v1);
+        plane_store(-1, v0);
+        plane_store(0, v1);
+    }
+    __syncthreads();
+
+    int tmy[M_TILES], tmx[M_TILES];
+#pragma unroll
+    for (int m = 0; m < M_TILES; ++m) {
+        const int g = wave * M_TILES + m;
+        tmx[m] = (g % XT) * 16;
+        tmy[m] = g / XT;
+    }
+    const int a_row = lane & 15;
+    const int a_k = lane >> 4;
+    const int col16 = lane & 15;
+    const int colj = lane & 15;
+    const int rbase = (lane >> 4) * 4;
+
+    for (int z = 0; z < D; ++z) {
+        f32x4 acc[M_TILES][NT];
+#pragma unroll
+        for (int m = 0; m < M_TILES; ++m)
+#pragma unroll
+            for (int t = 0; t < NT; ++t)
+                acc[m][t] = {0.f, 0.f, 0.f, 0.f};
+
+        auto compute_dzi = [&](int dzi) {
+            const int slot = ((z + dzi) % 3 + 3) % 3;  // plane z + dzi - 1
+            const float* plane = &ring[slot * SY * SX * PC];
+#pragma unroll
+            for (int tl = 0; tl < 9; ++tl) {
+                const int tap = dzi * 9 + tl;
+                const int dy = tl / 3 - 1;
+                const int dx = tl % 3 - 1;
+                const float* arow[M_TILES];
+#pragma unroll
+                for (int m = 0; m < M_TILES; ++m) {
+                    arow[m] = &plane[((1 + tmy[m] + dy) * SX +
+                                      (1 + tmx[m] + dx)) * PC +
+                                     a_row * PC + a_k];
+                }
+                const float* wblk = &wall[tap * NT * C * 16];
+#pragma unroll
+                for (int nt = 0; nt < NT; ++nt) {
+#pragma unroll
+                    for (int kk = 0; kk < KK; ++kk) {
+                        const float b =
+                            wblk[(nt * C + kk * 4 + a_k) * 16 + col16];
+#pragma unroll
+                        for (int m = 0; m < M_TILES; ++m) {
+                            const float a = arow[m][kk * 4];
+                            acc[m][nt] =
+                                __builtin_amdgcn_mfma_f32_16x16x4f32(
+                                    a, b, acc[m][nt], 0, 0, 0);
+                        }
+                    }
+                }
+            }
+        };
+
+        f32x4 vals[LV];
+        plane_load(z + 1, vals);   // in flight across the next 18 taps
+        compute_dzi(0);
+        compute_dzi(1);
+        plane_store(z + 1, vals);
+        __syncthreads();
+        compute_dzi(2);
+
+#pragma unroll
+        for (int m = 0; m < M_TILES; ++m) {
+            const int gy = y0 + tmy[m];
+            if (gy >= H) continue;
+#pragma unroll
+            for (int t = 0; t < NT; ++t) {
+                const int j = j0 + t * 16 + colj;
+                if (j >= K) continue;
+                const float bj = bias ? bias[j] : 0.f;
+#pragma unroll
+                for (int r = 0; r < 4; ++r) {
+                    const int gx = x0 + tmx[m] + rbase + r;
+                    if (gx >= W) continue;
+                    long long o =
+                        ((((long long)n * D + z) * H + gy) * W + gx) * K +
+                        j;
+                    float v = acc[m][t][r] + bj;
+                    if (res) v += res[o];
+                    if (do_elu) v = v > 0.f ? v : expm1f(v);
+                    out[o] = v;
+                }
+            }
+        }
+    }
+}
+
 }  // namespace
 
 extern "C" int cfx_conv3_ndhwc_zring(cfx_ctx* ctx, const float* in,
@@ -654,21 +840,38 @@ extern "C" int cfx_conv3_ndhwc_zring(cfx_ctx* ctx, const float* in,
                                      int N, int D, int H, int W, int C,
                                      int K, int do_elu) {
     dim3 grid((W + 15) / 16, (H + 7) / 8, (unsigned)N);
+    // CFX_ZRING_PL=0 falls back to the barrier-per-phase kernel (the
+    // pipelined one measured faster on hardware and is the default)
+    static const int use_pl = [] {
+        const char* e = getenv("CFX_ZRING_PL");
+        return e ? atoi(e) : 1;
+    }();
     hipEvent_t e0;
     if (prof_begin(ctx, &e0)) return -1;
     if (C == 28 && K == 28) {
         // wall holds both K tiles (96.8 KB + 60.5 KB ring)
-        hipLaunchKernelGGL((k_conv3_zring<28, 28, 8, 16, 2>), grid,
-                           dim3(512), 0, ctx->stream, in, wgt, bias,
-                           residual, out, N, D, H, W, do_elu, 0);
+        if (use_pl)
+            hipLaunchKernelGGL((k_conv3_zring_pl<28, 28, 8, 16, 2>), grid,
+                               dim3(512), 0, ctx->stream, in, wgt, bias,
+                               residual, out, N, D, H, W, do_elu, 0);
+        else
+            hipLaunchKernelGGL((k_conv3_zring<28, 28, 8, 16, 2>), grid,
+                               dim3(512), 0, ctx->stream, in, wgt, bias,
+                               residual, out, N, D, H, W, do_elu, 0);
     } else if (C == 36 && K == 36) {
         // one 16-wide K tile per launch (wall 62.2 KB + ring 77.8 KB);
         // the input re-read costs ~3x HBM traffic of a 113 MB activation
         // per conv — negligible next to the compute
         for (int j0 = 0; j0 < 36; j0 += 16) {
-            hipLaunchKernelGGL((k_conv3_zring<36, 36, 8, 16, 1>), grid,
-                               dim3(512), 0, ctx->stream, in, wgt, bias,
-                               residual, out, N, D, H, W, do_elu, j0);
+            if (use_pl)
+                hipLaunchKernelGGL((k_conv3_zring_pl<36, 36, 8, 16, 1>),
+                                   grid, dim3(512), 0, ctx->stream, in, wgt,
+                                   bias, residual, out, N, D, H, W, do_elu,
+                                   j0);
+            else
+                hipLaunchKernelGGL((k_conv3_zring<36, 36, 8, 16, 1>), grid,
+                                   dim3(512), 0, ctx->stream, in, wgt, bias,
+                                   residual, out, N, D, H, W, do_elu, j0);
         }
     } else {
         g_err = "cfx_conv3_ndhwc_zring: width not instantiated";

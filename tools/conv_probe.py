@@ -54,13 +54,14 @@ def probe(C, D, H, W, N=12, iters=10, w32=False, zring=False):
            'mine_TF': flops / tm / 1e12, 'torch_TF': flops / tt / 1e12,
            'speedup': tt / tm}, flush=True)
 
-probe(28, 20, 256, 256)
-probe(28, 20, 256, 256, w32=True)
-probe(28, 20, 256, 256, zring=True)
-probe(36, 20, 128, 128, zring=True)
-probe(36, 20, 128, 128)
-probe(48, 20, 64, 64)
-probe(64, 20, 32, 32)
+if __name__ == '__main__' and '--bf16' not in sys.argv:
+    probe(28, 20, 256, 256)
+    probe(28, 20, 256, 256, w32=True)
+    probe(28, 20, 256, 256, zring=True)
+    probe(36, 20, 128, 128, zring=True)
+    probe(36, 20, 128, 128)
+    probe(48, 20, 64, 64)
+    probe(64, 20, 32, 32)
 
 
 def probe_bf16(C=28, D=20, H=256, W=256, N=24, iters=8):
@@ -96,5 +97,5 @@ def probe_bf16(C=28, D=20, H=256, W=256, N=24, iters=8):
            'torch_TF': flops / tt / 1e12, 'speedup': tt / tm}, flush=True)
 
 
-if '--bf16' in sys.argv:
+if __name__ == '__main__' and '--bf16' in sys.argv:
     probe_bf16(28, 32, 256, 256, N=24)
