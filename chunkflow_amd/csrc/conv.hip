@@ -1024,6 +1024,155 @@ __global__ __launch_bounds__(512, 1) void k_conv3_zring_bf16(
     }
 }
 
+// Software-pipelined bf16 ring (same one-barrier-per-z scheme as
+// k_conv3_zring_pl: next plane's global loads fly over the first 18 taps,
+// drain to LDS, one barrier, last 9 taps). The plain bf16 kernel is
+// staging-bound — only 2 MFMAs per tap — so the overlap matters more here
+// than in f32.
+template <int C, int K, int TY, int TX>
+__global__ __launch_bounds__(512, 1) void k_conv3_zring_bf16_pl(
+    const cfx_bf16* __restrict__ in, const cfx_bf16* __restrict__ wgt,
+    const float* __restrict__ bias, const cfx_bf16* __restrict__ res,
+    cfx_bf16* __restrict__ out, int N, int D, int H, int W, int do_elu) {
+    constexpr int CP = 32;
+    constexpr int PCB = CP + 8;
+    constexpr int SX = TX + 2;
+    constexpr int SY = TY + 2;
+    constexpr int KK = CP / 16;
+    constexpr int C4 = CP / 4;
+    constexpr int LV = (SY * SX * C4 + 511) / 512;
+    typedef cfx_bf16 bf16x4 __attribute__((ext_vector_type(4)));
+    static_assert(C <= CP && K <= 32, "");
+
+    __shared__ cfx_bf16 ring[3 * SY * SX * PCB];
+    __shared__ cfx_bf16 wall[27 * 32 * PCB];
+
+    const int n = blockIdx.z;
+    const int y0 = blockIdx.y * TY;
+    const int x0 = blockIdx.x * TX;
+    const int tid = threadIdx.x;
+    const int wave = tid >> 6;
+    const int lane = tid & 63;
+
+    for (int idx = tid; idx < 27 * 32 * CP; idx += 512) {
+        const int c = idx % CP;
+        const int j = (idx / CP) % 32;
+        const int tap = idx / (CP * 32);
+        wall[(tap * 32 + j) * PCB + c] = wgt[(tap * 32 + j) * 32 + c];
+    }
+
+    const bool xy_interior = y0 >= 1 && y0 + TY + 1 <= H && x0 >= 1 &&
+                             x0 + TX + 1 <= W;
+
+    auto plane_load = [&](int P, bf16x4 (&vals)[LV]) {
+        const bool zin = P >= 0 && P < D;
+        const bool interior = zin && xy_interior;
+#pragma unroll
+        for (int li = 0; li < LV; ++li) {
+            const int idx = tid + li * 512;
+            if (idx >= SY * SX * C4) break;
+            const int c4 = idx % C4;
+            const int v = idx / C4;
+            const int gy = y0 + v / SX - 1;
+            const int gx = x0 + v % SX - 1;
+            vals[li] = {};
+            if (c4 * 4 < C) {
+                if (interior) {
+                    vals[li] = *reinterpret_cast<const bf16x4*>(
+                        in + ((((long long)n * D + P) * H + gy) * W + gx)
+                                 * C + c4 * 4);
+                } else {
+                    const bool ok = zin && gy >= 0 && gy < H && gx >= 0 &&
+                                    gx < W;
+                    vals[li] = *reinterpret_cast<const bf16x4*>(
+                        in + ((((long long)n * D + (zin ? P : 0)) * H +
+                               (ok ? gy : 0)) * W + (ok ? gx : 0)) * C +
+                        c4 * 4);
+                    if (!ok) vals[li] = {};
+                }
+            }
+        }
+    };
+    auto plane_store = [&](int P, const bf16x4 (&vals)[LV]) {
+        const int slot = ((P + 1) % 3 + 3) % 3;
+#pragma unroll
+        for (int li = 0; li < LV; ++li) {
+            const int idx = tid + li * 512;
+            if (idx >= SY * SX * C4) break;
+            const int c4 = idx % C4;
+            const int v = idx / C4;
+            *reinterpret_cast<bf16x4*>(
+                &ring[((slot * SY + v / SX) * SX + v % SX) * PCB +
+                      c4 * 4]) = vals[li];
+        }
+    };
+
+    {
+        bf16x4 v0[LV], v1[LV];
+        plane_load(-1, v0);
+        plane_load(0, v1);
+        plane_store(-1, v0);
+        plane_store(0, v1);
+    }
+    __syncthreads();
+
+    const int ax = lane & 31;
+    const int khalf = (lane >> 5) * 8;
+
+    for (int z = 0; z < D; ++z) {
+        f32x16 acc = {};
+        auto compute_dzi = [&](int dzi) {
+            const int slot = ((z + dzi) % 3 + 3) % 3;
+            const cfx_bf16* plane = &ring[slot * SY * SX * PCB];
+#pragma unroll
+            for (int tl = 0; tl < 9; ++tl) {
+                const int tap = dzi * 9 + tl;
+                const int dy = tl / 3 - 1;
+                const int dx = tl % 3 - 1;
+                const cfx_bf16* arow =
+                    &plane[((1 + wave + dy) * SX + (1 + dx) + ax) * PCB +
+                           khalf];
+                const cfx_bf16* brow = &wall[(tap * 32 + ax) * PCB + khalf];
+#pragma unroll
+                for (int kk = 0; kk < KK; ++kk) {
+                    const bf16x8 a =
+                        *reinterpret_cast<const bf16x8*>(arow + kk * 16);
+                    const bf16x8 b =
+                        *reinterpret_cast<const bf16x8*>(brow + kk * 16);
+                    acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, acc,
+                                                                  0, 0, 0);
+                }
+            }
+        };
+
+        bf16x4 vals[LV];
+        plane_load(z + 1, vals);
+        compute_dzi(0);
+        compute_dzi(1);
+        plane_store(z + 1, vals);
+        __syncthreads();
+        compute_dzi(2);
+
+        const int gy = y0 + wave;
+        const int j = lane & 31;
+        if (gy < H && j < K) {
+            const float bj = bias ? bias[j] : 0.f;
+#pragma unroll
+            for (int r = 0; r < 16; ++r) {
+                const int row = (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
+                const int gx = x0 + row;
+                if (gx >= W) continue;
+                long long o =
+                    ((((long long)n * D + z) * H + gy) * W + gx) * K + j;
+                float v = acc[r] + bj;
+                if (res) v += (float)res[o];
+                if (do_elu) v = v > 0.f ? v : expm1f(v);
+                out[o] = (cfx_bf16)v;
+            }
+        }
+    }
+}
+
 }  // namespace
 
 extern "C" int cfx_conv3_ndhwc_bf16(cfx_ctx* ctx, const void* in,
@@ -1036,13 +1185,24 @@ extern "C" int cfx_conv3_ndhwc_bf16(cfx_ctx* ctx, const void* in,
         return -1;
     }
     dim3 grid((W + 31) / 32, (H + 7) / 8, (unsigned)N);
+    static const int use_pl = [] {
+        const char* e = getenv("CFX_ZRING_PL");
+        return e ? atoi(e) : 1;
+    }();
     hipEvent_t e0;
     if (prof_begin(ctx, &e0)) return -1;
-    hipLaunchKernelGGL((k_conv3_zring_bf16<28, 28, 8, 32>), grid, dim3(512),
-                       0, ctx->stream, (const cfx_bf16*)in,
-                       (const cfx_bf16*)wgt, bias,
-                       (const cfx_bf16*)residual, (cfx_bf16*)out, N, D, H,
-                       W, do_elu);
+    if (use_pl)
+        hipLaunchKernelGGL((k_conv3_zring_bf16_pl<28, 28, 8, 32>), grid,
+                           dim3(512), 0, ctx->stream, (const cfx_bf16*)in,
+                           (const cfx_bf16*)wgt, bias,
+                           (const cfx_bf16*)residual, (cfx_bf16*)out, N, D,
+                           H, W, do_elu);
+    else
+        hipLaunchKernelGGL((k_conv3_zring_bf16<28, 28, 8, 32>), grid,
+                           dim3(512), 0, ctx->stream, (const cfx_bf16*)in,
+                           (const cfx_bf16*)wgt, bias,
+                           (const cfx_bf16*)residual, (cfx_bf16*)out, N, D,
+                           H, W, do_elu);
     CFX_CHECK(hipGetLastError());
     double flops = 2.0 * 27.0 * 28 * 28 * (double)N * D * H * W;
     if (prof_end(ctx, e0, CFX_K_CONV, flops)) return -1;
