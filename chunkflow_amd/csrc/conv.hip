@@ -1185,9 +1185,11 @@ extern "C" int cfx_conv3_ndhwc_bf16(cfx_ctx* ctx, const void* in,
         return -1;
     }
     dim3 grid((W + 31) / 32, (H + 7) / 8, (unsigned)N);
+    // default 0 here: measured 159.0 TF plain vs 157.1 pipelined — the
+    // bf16 ring is LDS-issue-bound, not staging-latency-bound
     static const int use_pl = [] {
         const char* e = getenv("CFX_ZRING_PL");
-        return e ? atoi(e) : 1;
+        return e ? atoi(e) : 0;
     }();
     hipEvent_t e0;
     if (prof_begin(ctx, &e0)) return -1;
