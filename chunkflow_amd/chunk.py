@@ -11,6 +11,8 @@ host methods below are the reference semantics at the pipeline boundary.
 """
 from __future__ import annotations
 
+import os
+
 import numpy as np
 
 from .cartesian import BoundingBox, Cartesian
@@ -298,6 +300,87 @@ class Chunk:
         return cls(arr,
                    voxel_offset=voxel_offset or (0, 0, 0),
                    voxel_size=voxel_size)
+
+    def to_h5(self, file_name: str, with_offset: bool = True,
+              chunk_size=(8, 8, 8), compression='gzip', voxel_size=None):
+        """Save via the in-repo HDF5 codec, keeping the reference's file
+        shape (chunk/base.py:368-410): datasets /main, /voxel_offset,
+        /voxel_size; a non-.h5 file_name gets the bbox string appended.
+        chunk_size/compression are accepted for signature parity but the
+        codec stores contiguously (a lossless storage-layout difference;
+        h5py reads either)."""
+        from . import h5io
+        del chunk_size, compression
+        if not file_name.endswith('.h5'):
+            file_name += self.bbox.string + '.h5'
+        if os.path.exists(file_name):
+            os.remove(file_name)
+        ds = {'main': self.numpy().array}
+        if voxel_size is None and self.voxel_size is not None:
+            voxel_size = tuple(self.voxel_size)
+        if voxel_size is not None:
+            ds['voxel_size'] = np.asarray(voxel_size, dtype=np.int64)
+        if with_offset and self.voxel_offset is not None:
+            ds['voxel_offset'] = np.asarray(tuple(self.voxel_offset),
+                                            dtype=np.int64)
+        h5io.write_h5(file_name, ds)
+        return file_name
+
+    @classmethod
+    def from_h5(cls, file_name: str, voxel_offset=None, dataset_path=None,
+                voxel_size=None, cutout_start=None, cutout_stop=None,
+                cutout_size=None, dtype=None):
+        """Load via the in-repo HDF5 codec with the reference's semantics
+        (chunk/base.py:267-366): default dataset = first key without
+        'global'/'offset'/'unique'; /voxel_offset & /voxel_size datasets;
+        cutout in GLOBAL coordinates; a non-HDF5 file_name is a prefix and
+        a missing/empty per-bbox file returns None."""
+        from . import h5io
+        file_name = os.path.expanduser(file_name)
+        if cutout_start is not None and cutout_size is not None and \
+                cutout_stop is None:
+            cutout_stop = tuple(t + s for t, s in
+                                zip(cutout_start, cutout_size))
+        if not (os.path.isfile(file_name) and h5io.is_hdf5(file_name)):
+            assert cutout_start is not None and cutout_stop is not None
+            bbox = BoundingBox(cutout_start, cutout_stop)
+            file_name += f'{bbox.string}.h5'
+            if not os.path.exists(file_name) or \
+                    os.path.getsize(file_name) == 0:
+                return None
+        data = h5io.read_h5(file_name)
+        if dataset_path is None:
+            for key in data:
+                if 'global' not in key and 'offset' not in key and \
+                        'unique' not in key:
+                    dataset_path = key
+                    break
+        arr = data[dataset_path.lstrip('/')
+                   if dataset_path.lstrip('/') in data else dataset_path]
+        if voxel_offset is None:
+            voxel_offset = tuple(data['voxel_offset']) \
+                if 'voxel_offset' in data else (0, 0, 0)
+        if voxel_size is None and 'voxel_size' in data:
+            voxel_size = tuple(data['voxel_size'])
+        if cutout_start is None:
+            cutout_start = tuple(voxel_offset)
+        if cutout_size is None:
+            cutout_size = arr.shape[-3:]
+        if cutout_stop is None:
+            cutout_stop = tuple(t + s for t, s in
+                                zip(cutout_start, cutout_size))
+        for c, v in zip(cutout_start, voxel_offset):
+            assert c >= v, 'cutout must start at/after the voxel offset'
+        arr = arr[...,
+                  cutout_start[0] - voxel_offset[0]:
+                  cutout_stop[0] - voxel_offset[0],
+                  cutout_start[1] - voxel_offset[1]:
+                  cutout_stop[1] - voxel_offset[1],
+                  cutout_start[2] - voxel_offset[2]:
+                  cutout_stop[2] - voxel_offset[2]]
+        if dtype is not None:
+            arr = arr.astype(dtype)
+        return cls(arr, voxel_offset=cutout_start, voxel_size=voxel_size)
 
     @classmethod
     def from_npy(cls, file_name: str, voxel_offset=(0, 0, 0),

@@ -416,6 +416,90 @@ def load_npy(tasks, name, file_name, voxel_offset, output_chunk_name):
         yield task
 
 
+@main.command('load-h5')
+@click.option('--name', type=str, default='load-h5', help='name of operator')
+@click.option('--file-name', '-f', type=str, required=True,
+              help='HDF5 file, or a per-bbox file prefix')
+@click.option('--dataset-path', '-d', type=str, default=None,
+              help='dataset path inside the HDF5 file')
+@click.option('--dtype', '-e', default=None,
+              type=click.Choice(['float32', 'float64', 'uint16', 'uint32',
+                                 'uint64', 'uint8']),
+              help='transform data type')
+@click.option('--voxel-offset', '-v', type=click.INT, nargs=3, default=None,
+              callback=default_none, help='voxel offset of the dataset')
+@click.option('--voxel-size', '-x', type=click.INT, nargs=3, default=None,
+              callback=default_none, help='physical voxel size (nm)')
+@click.option('--cutout-start', '-t', type=click.INT, nargs=3,
+              callback=default_none, help='cutout start (global)')
+@click.option('--cutout-stop', '-p', type=click.INT, nargs=3,
+              callback=default_none, help='cutout stop (global)')
+@click.option('--cutout-size', '-s', type=click.INT, nargs=3,
+              callback=default_none, help='cutout size')
+@click.option('--set-bbox/--no-set-bbox', default=False,
+              help='set the task bbox from the loaded chunk')
+@click.option('--output-chunk-name', '-o', type=str,
+              default=DEFAULT_CHUNK_NAME, help='output chunk name')
+@operator
+def load_h5(tasks, name, file_name, dataset_path, dtype, voxel_offset,
+            voxel_size, cutout_start, cutout_stop, cutout_size, set_bbox,
+            output_chunk_name):
+    """Read HDF5 files (reference flow.py:976-1065; in-repo codec —
+    a task bbox supplies the cutout and per-bbox file name)."""
+    for task in tasks:
+        if task is not None:
+            start = time()
+            cs, cp = cutout_start, cutout_stop
+            csz = cutout_size
+            fname = file_name
+            if 'bbox' in task and cutout_start is None:
+                bbox = task['bbox']
+                cs, cp = tuple(bbox.start), tuple(bbox.stop)
+                csz = tuple(bbox.shape)
+                if not file_name.endswith('.h5'):
+                    fname = f'{file_name}{bbox.string}.h5'
+            chunk = Chunk.from_h5(
+                fname, dataset_path=dataset_path,
+                voxel_offset=voxel_offset, voxel_size=voxel_size,
+                cutout_start=cs, cutout_stop=cp, cutout_size=csz,
+                dtype=dtype)
+            if chunk is not None and dtype is not None:
+                chunk = chunk.astype(dtype)
+            task[output_chunk_name] = chunk
+            if set_bbox and chunk is not None:
+                task['bbox'] = chunk.bbox
+            task['log']['timer'][name] = time() - start
+        yield task
+
+
+@main.command('save-h5')
+@click.option('--name', type=str, default='save-h5', help='name of operator')
+@click.option('--input-name', '-i', type=str, default=DEFAULT_CHUNK_NAME,
+              help='input chunk name')
+@click.option('--file-name', '-f', type=str, required=True,
+              help='file name or prefix of the output HDF5 file')
+@click.option('--with-offset/--without-offset', default=True,
+              help='store the /voxel_offset dataset or not')
+@click.option('--voxel-size', '-v', type=click.INT, nargs=3, default=None,
+              callback=default_none, help='voxel size to store')
+@click.option('--dtype', '-d', type=str, default=None,
+              help='data type conversion before saving')
+@operator
+def save_h5(tasks, name, input_name, file_name, with_offset, voxel_size,
+            dtype):
+    """Save chunk to an HDF5 file (reference flow.py:1068-1120)."""
+    for task in tasks:
+        if task is not None:
+            start = time()
+            chunk = task[input_name]
+            if dtype is not None:
+                chunk = chunk.astype(dtype)
+            chunk.to_h5(file_name, with_offset=with_offset,
+                        voxel_size=voxel_size)
+            task['log']['timer'][name] = time() - start
+        yield task
+
+
 @main.command('load-tif')
 @click.option('--name', type=str, default='load-tif', help='name of operator')
 @click.option('--file-name', '-f', required=True,
