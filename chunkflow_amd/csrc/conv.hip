@@ -832,6 +832,198 @@ __global__ __launch_bounds__(512, 1) void k_conv3_zring_pl(
     }
 }
 
+// C=48 persistent-z ring: the full 27-tap weight wall (82.9 KB per
+// 16-wide K tile) no longer fits LDS beside the 3-plane ring (103.7 KB),
+// so the wall is DOUBLE-BUFFERED per 9-tap dzi group (2 x 27.6 KB) and
+// each group's stage overlaps the previous group's MFMAs. K covered by
+// three j0 launches. Per z: 3 barriers; wall restage is ~3% of the z's
+// MFMA time (weights come from L2 after the first z). Buffer parity is
+// (3*z + dzi) % 2 — odd group count flips the pairing every z, and each
+// write lands on a buffer whose last reader sat before the previous
+// barrier (see the B1/B2/B3 comments).
+template <int C, int K, int TY, int TX>
+__global__ __launch_bounds__(512, 1) void k_conv3_zring_dw(
+    const float* __restrict__ in, const float* __restrict__ wgt,
+    const float* __restrict__ bias, const float* __restrict__ res,
+    float* __restrict__ out, int N, int D, int H, int W, int do_elu,
+    int j0) {
+    // ring stride: C+2 when C%32==16 (48 -> 50; mod 32 = 18 keeps the
+    // A reads bank-conflict-free) — the +4 pad of padc() would push the
+    // ring + double wall 3.7 KB past the 160 KB LDS. Stride 50 dwords is
+    // only 8-byte aligned, so the ring staging stores are b64.
+    constexpr int PC = (C % 32 == 16) ? C + 2 : padc(C);
+    constexpr int SX = TX + 2;
+    constexpr int SY = TY + 2;
+    constexpr int KK = C / 4;
+    constexpr int NW = 8;
+    constexpr int XT = TX / 16;
+    constexpr int M_TILES = (TY * XT) / NW;
+    constexpr int C4N = C / 4;
+    constexpr int LV = (SY * SX * C4N + 511) / 512;
+    static_assert((TY * XT) % NW == 0, "");
+
+    __shared__ float ring[3 * SY * SX * PC];
+    __shared__ float wall[2][9 * C * 16];
+
+    const int n = blockIdx.z;
+    const int y0 = blockIdx.y * TY;
+    const int x0 = blockIdx.x * TX;
+    const int tid = threadIdx.x;
+    const int wave = tid >> 6;
+    const int lane = tid & 63;
+
+    // stage the 9-tap group g (taps 9g..9g+8) into wall[buf]
+    auto wall_store = [&](int g, int buf) {
+        for (int idx = tid; idx < 9 * C * 16; idx += 512) {
+            const int j = idx & 15;
+            const int c = (idx >> 4) % C;
+            const int tl = (idx >> 4) / C;
+            const int jg = j0 + j;
+            wall[buf][idx] =
+                jg < K ? wgt[((long long)(9 * g + tl) * C + c) * K + jg]
+                       : 0.f;
+        }
+    };
+
+    const bool xy_interior = y0 >= 1 && y0 + TY + 1 <= H && x0 >= 1 &&
+                             x0 + TX + 1 <= W;
+    auto plane_load = [&](int P, f32x4 (&vals)[LV]) {
+        const bool zin = P >= 0 && P < D;
+        const bool interior = zin && xy_interior;
+#pragma unroll
+        for (int li = 0; li < LV; ++li) {
+            const int idx = tid + li * 512;
+            if (idx >= SY * SX * C4N) break;
+            const int c4 = idx % C4N;
+            const int v = idx / C4N;
+            const int gy = y0 + v / SX - 1;
+            const int gx = x0 + v % SX - 1;
+            if (interior) {
+                vals[li] = *reinterpret_cast<const f32x4*>(
+                    in + ((((long long)n * D + P) * H + gy) * W + gx) * C +
+                    c4 * 4);
+            } else {
+                const bool ok = zin && gy >= 0 && gy < H && gx >= 0 &&
+                                gx < W;
+                vals[li] = *reinterpret_cast<const f32x4*>(
+                    in + ((((long long)n * D + (zin ? P : 0)) * H +
+                           (ok ? gy : 0)) * W + (ok ? gx : 0)) * C +
+                    c4 * 4);
+                if (!ok) vals[li] = {0.f, 0.f, 0.f, 0.f};
+            }
+        }
+    };
+    typedef float f32x2 __attribute__((ext_vector_type(2)));
+    auto plane_store = [&](int P, const f32x4 (&vals)[LV]) {
+        const int slot = ((P + 1) % 3 + 3) % 3;
+#pragma unroll
+        for (int li = 0; li < LV; ++li) {
+            const int idx = tid + li * 512;
+            if (idx >= SY * SX * C4N) break;
+            const int c4 = idx % C4N;
+            const int v = idx / C4N;
+            float* dst = &ring[(slot * SY * SX + v) * PC + c4 * 4];
+            *reinterpret_cast<f32x2*>(dst) = f32x2{vals[li][0], vals[li][1]};
+            *reinterpret_cast<f32x2*>(dst + 2) =
+                f32x2{vals[li][2], vals[li][3]};
+        }
+    };
+
+    {
+        f32x4 v0[LV], v1[LV];
+        plane_load(-1, v0);
+        plane_load(0, v1);
+        plane_store(-1, v0);
+        plane_store(0, v1);
+        wall_store(0, 0);  // g0 of z=0 -> buf (3*0+0)%2 = 0
+    }
+    __syncthreads();
+
+    int tmy[M_TILES], tmx[M_TILES];
+#pragma unroll
+    for (int m = 0; m < M_TILES; ++m) {
+        const int g = wave * M_TILES + m;
+        tmx[m] = (g % XT) * 16;
+        tmy[m] = g / XT;
+    }
+    const int a_row = lane & 15;
+    const int a_k = lane >> 4;
+    const int col16 = lane & 15;
+    const int colj = lane & 15;
+    const int rbase = (lane >> 4) * 4;
+
+    for (int z = 0; z < D; ++z) {
+        f32x4 acc[M_TILES];
+#pragma unroll
+        for (int m = 0; m < M_TILES; ++m)
+            acc[m] = {0.f, 0.f, 0.f, 0.f};
+
+        auto compute_dzi = [&](int dzi, int buf) {
+            const int slot = ((z + dzi) % 3 + 3) % 3;
+            const float* plane = &ring[slot * SY * SX * PC];
+            const float* wb = wall[buf];
+#pragma unroll
+            for (int tl = 0; tl < 9; ++tl) {
+                const int dy = tl / 3 - 1;
+                const int dx = tl % 3 - 1;
+                const float* arow[M_TILES];
+#pragma unroll
+                for (int m = 0; m < M_TILES; ++m) {
+                    arow[m] = &plane[((1 + tmy[m] + dy) * SX +
+                                      (1 + tmx[m] + dx)) * PC +
+                                     a_row * PC + a_k];
+                }
+                const float* wblk = &wb[tl * C * 16];
+#pragma unroll
+                for (int kk = 0; kk < KK; ++kk) {
+                    const float b = wblk[(kk * 4 + a_k) * 16 + col16];
+#pragma unroll
+                    for (int m = 0; m < M_TILES; ++m) {
+                        const float a = arow[m][kk * 4];
+                        acc[m] = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                            a, b, acc[m], 0, 0, 0);
+                    }
+                }
+            }
+        };
+
+        const int p = (3 * z) % 2;  // buffer of this z's g0
+        f32x4 vals[LV];
+        plane_load(z + 1, vals);
+        wall_store(1, 1 - p);      // g1 -> other buf (last read: g2 of z-1)
+        compute_dzi(0, p);
+        __syncthreads();           // B1: g1 visible; g0 buf free
+        wall_store(2, p);          // g2 overwrites g0's buf
+        compute_dzi(1, 1 - p);
+        plane_store(z + 1, vals);
+        __syncthreads();           // B2: g2 + plane z+1 visible; g1 buf free
+        if (z + 1 < D)
+            wall_store(0, 1 - p);  // next z's g0 -> (3(z+1))%2 == 1-p
+        compute_dzi(2, p);
+
+#pragma unroll
+        for (int m = 0; m < M_TILES; ++m) {
+            const int gy = y0 + tmy[m];
+            if (gy >= H) continue;
+            const int j = j0 + colj;
+            if (j >= K) continue;
+            const float bj = bias ? bias[j] : 0.f;
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                const int gx = x0 + tmx[m] + rbase + r;
+                if (gx >= W) continue;
+                long long o =
+                    ((((long long)n * D + z) * H + gy) * W + gx) * K + j;
+                float v = acc[m][r] + bj;
+                if (res) v += res[o];
+                if (do_elu) v = v > 0.f ? v : expm1f(v);
+                out[o] = v;
+            }
+        }
+        __syncthreads();           // B3: next z's g0 visible; g2 buf free
+    }
+}
+
 }  // namespace
 
 extern "C" int cfx_conv3_ndhwc_zring(cfx_ctx* ctx, const float* in,
@@ -872,6 +1064,14 @@ extern "C" int cfx_conv3_ndhwc_zring(cfx_ctx* ctx, const float* in,
                 hipLaunchKernelGGL((k_conv3_zring<36, 36, 8, 16, 1>), grid,
                                    dim3(512), 0, ctx->stream, in, wgt, bias,
                                    residual, out, N, D, H, W, do_elu, j0);
+        }
+    } else if (C == 48 && K == 48) {
+        // double-buffered per-dzi weight wall (full wall no longer fits
+        // beside the ring); one 16-wide K tile per launch
+        for (int j0 = 0; j0 < 48; j0 += 16) {
+            hipLaunchKernelGGL((k_conv3_zring_dw<48, 48, 8, 16>), grid,
+                               dim3(512), 0, ctx->stream, in, wgt, bias,
+                               residual, out, N, D, H, W, do_elu, j0);
         }
     } else {
         g_err = "cfx_conv3_ndhwc_zring: width not instantiated";

@@ -14,3 +14,6 @@ print('CFX_ZRING_PL =', os.environ.get('CFX_ZRING_PL', '(default 1)'),
       flush=True)
 probe(28, 20, 256, 256, zring=True)
 probe(36, 20, 128, 128, zring=True)
+if '--c48' in sys.argv:
+    probe(48, 20, 64, 64, zring=True)
+    probe(48, 20, 64, 64)
