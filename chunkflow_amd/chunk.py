@@ -158,7 +158,10 @@ class Chunk:
     def astype(self, dtype) -> 'Chunk':
         if _is_tensor(self.array):
             tdt = {'float32': torch.float32, 'float16': torch.float16,
-                   'uint8': torch.uint8}[np.dtype(dtype).name]
+                   'float64': torch.float64, 'uint8': torch.uint8,
+                   'int32': torch.int32, 'int64': torch.int64,
+                   'uint32': torch.uint32, 'uint16': torch.uint16,
+                   'bool': torch.bool}[np.dtype(dtype).name]
             return Chunk(self.array.to(tdt), voxel_offset=self.voxel_offset,
                          voxel_size=self.voxel_size)
         return Chunk(self.array.astype(dtype), voxel_offset=self.voxel_offset,
