@@ -1179,27 +1179,53 @@ __global__ __launch_bounds__(512, 1) void k_conv3_zring_bf16(
         __syncthreads();
 
         f32x16 acc = {};
-        for (int dzi = 0; dzi < 3; ++dzi) {
-            const int slot = ((z + dzi) % 3 + 3) % 3;
-            const cfx_bf16* plane = &ring[slot * SY * SX * PCB];
+        {
+            // Explicit 4-pair (2-tap) fragment prefetch pipeline. The
+            // compiler's natural schedule issues a pair's ds_reads only
+            // ~1 MFMA before their s_waitcnt, so every tap parks for the
+            // LDS latency (PMC: 67% SQ_WAIT_ANY). With distance 4 the
+            // 4 MFMAs in flight (~128 cyc) cover the read latency and the
+            // waitcnts become lgkmcnt(6) partial waits.
+            const cfx_bf16* planes[3];
 #pragma unroll
-            for (int tl = 0; tl < 9; ++tl) {
-                const int tap = dzi * 9 + tl;
-                const int dy = tl / 3 - 1;
-                const int dx = tl % 3 - 1;
-                const cfx_bf16* arow =
-                    &plane[((1 + wave + dy) * SX + (1 + dx) + ax) * PCB +
-                           khalf];
-                const cfx_bf16* brow = &wall[(tap * 32 + ax) * PCB + khalf];
+            for (int dzi = 0; dzi < 3; ++dzi)
+                planes[dzi] =
+                    &ring[(((z + dzi) % 3 + 3) % 3) * SY * SX * PCB];
+            auto addrA = [&](int p) {
+                const int tap = p >> 1, kk = p & 1;
+                const int dzi = tap / 9, tl = tap % 9;
+                const int dy = tl / 3 - 1, dx = tl % 3 - 1;
+                return reinterpret_cast<const bf16x8*>(
+                    &planes[dzi][((1 + wave + dy) * SX + (1 + dx) + ax) *
+                                     PCB + khalf + kk * 16]);
+            };
+            auto addrB = [&](int p) {
+                const int tap = p >> 1, kk = p & 1;
+                return reinterpret_cast<const bf16x8*>(
+                    &wall[(tap * 32 + ax) * PCB + khalf + kk * 16]);
+            };
+            constexpr int PD = 4;        // pairs in flight
+            constexpr int NP = 27 * KK;  // 54 A*B pairs per z
+            bf16x8 abuf[PD], bbuf[PD];
 #pragma unroll
-                for (int kk = 0; kk < KK; ++kk) {
-                    const bf16x8 a =
-                        *reinterpret_cast<const bf16x8*>(arow + kk * 16);
-                    const bf16x8 b =
-                        *reinterpret_cast<const bf16x8*>(brow + kk * 16);
-                    acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, acc,
-                                                                  0, 0, 0);
+            for (int p = 0; p < PD; ++p) {
+                abuf[p] = *addrA(p);
+                bbuf[p] = *addrB(p);
+            }
+#pragma unroll
+            for (int p = 0; p < NP; ++p) {
+                const int s = p % PD;
+                acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                    abuf[s], bbuf[s], acc, 0, 0, 0);
+                // sched_barriers pin the issue order (mfma p, then the
+                // p+PD reads) — without them the scheduler collapses the
+                // pipeline back to distance-1 and re-exposes the latency
+                __builtin_amdgcn_sched_barrier(0);
+                if (p + PD < NP) {
+                    abuf[s] = *addrA(p + PD);
+                    bbuf[s] = *addrB(p + PD);
                 }
+                __builtin_amdgcn_sched_barrier(0);
             }
         }
 
