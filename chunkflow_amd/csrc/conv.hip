@@ -1098,7 +1098,11 @@ typedef __bf16 cfx_bf16;
 typedef cfx_bf16 bf16x8 __attribute__((ext_vector_type(8)));
 typedef float f32x16 __attribute__((ext_vector_type(16)));
 
-template <int C, int K, int TY, int TX>
+// MODE: 0 = full; 1 = skip the MFMA mainloop; 2 = skip the epilogue
+// stores/res; 3 = skip per-z plane staging (wrong results; timing only) —
+// phase-ablation instrumentation for the stall hunt (tools/conv_probe
+// --bf16-ablate).
+template <int C, int K, int TY, int TX, int MODE = 0>
 __global__ __launch_bounds__(512, 1) void k_conv3_zring_bf16(
     const cfx_bf16* __restrict__ in,    // (N, D, H, W, C)
     const cfx_bf16* __restrict__ wgt,   // (27, 32, 32): [tap][j][c], padded
@@ -1175,11 +1179,11 @@ __global__ __launch_bounds__(512, 1) void k_conv3_zring_bf16(
     const int khalf = (lane >> 5) * 8;   // this lane's k sub-range base
 
     for (int z = 0; z < D; ++z) {
-        stage_plane(z + 1);
+        if (MODE != 3) stage_plane(z + 1);
         __syncthreads();
 
         f32x16 acc = {};
-        {
+        if (MODE != 1) {
             // Explicit 4-pair (2-tap) fragment prefetch pipeline. The
             // compiler's natural schedule issues a pair's ds_reads only
             // ~1 MFMA before their s_waitcnt, so every tap parks for the
@@ -1231,7 +1235,7 @@ __global__ __launch_bounds__(512, 1) void k_conv3_zring_bf16(
 
         const int gy = y0 + wave;
         const int j = lane & 31;
-        if (gy < H && j < K) {
+        if (MODE != 2 && gy < H && j < K) {
             const float bj = bias ? bias[j] : 0.f;
 #pragma unroll
             for (int r = 0; r < 16; ++r) {
@@ -1419,8 +1423,30 @@ extern "C" int cfx_conv3_ndhwc_bf16(cfx_ctx* ctx, const void* in,
     }();
     hipEvent_t e0;
     if (prof_begin(ctx, &e0)) return -1;
+    static const int mode = [] {
+        const char* e = getenv("CFX_BF16_MODE");  // phase ablation (timing)
+        return e ? atoi(e) : 0;
+    }();
     if (use_pl)
         hipLaunchKernelGGL((k_conv3_zring_bf16_pl<28, 28, 8, 32>), grid,
+                           dim3(512), 0, ctx->stream, (const cfx_bf16*)in,
+                           (const cfx_bf16*)wgt, bias,
+                           (const cfx_bf16*)residual, (cfx_bf16*)out, N, D,
+                           H, W, do_elu);
+    else if (mode == 1)
+        hipLaunchKernelGGL((k_conv3_zring_bf16<28, 28, 8, 32, 1>), grid,
+                           dim3(512), 0, ctx->stream, (const cfx_bf16*)in,
+                           (const cfx_bf16*)wgt, bias,
+                           (const cfx_bf16*)residual, (cfx_bf16*)out, N, D,
+                           H, W, do_elu);
+    else if (mode == 2)
+        hipLaunchKernelGGL((k_conv3_zring_bf16<28, 28, 8, 32, 2>), grid,
+                           dim3(512), 0, ctx->stream, (const cfx_bf16*)in,
+                           (const cfx_bf16*)wgt, bias,
+                           (const cfx_bf16*)residual, (cfx_bf16*)out, N, D,
+                           H, W, do_elu);
+    else if (mode == 3)
+        hipLaunchKernelGGL((k_conv3_zring_bf16<28, 28, 8, 32, 3>), grid,
                            dim3(512), 0, ctx->stream, (const cfx_bf16*)in,
                            (const cfx_bf16*)wgt, bias,
                            (const cfx_bf16*)residual, (cfx_bf16*)out, N, D,
