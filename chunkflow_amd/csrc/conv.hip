@@ -1135,40 +1135,55 @@ __global__ __launch_bounds__(512, 1) void k_conv3_zring_bf16(
         wall[(tap * 32 + j) * PCB + c] = wgt[(tap * 32 + j) * 32 + c];
     }
 
+    // Staging is load-phase/store-phase with EXPLICIT register arrays and
+    // unconditional clamped loads: a single dynamic loop makes the
+    // compiler round-robin two registers and emit a full vmcnt(0) drain
+    // per 8-byte load (~900 cy of HBM latency each, serialized — the
+    // disassembly showed gload/waitcnt(0) pairs and the phase ablation
+    // priced it at ~2.6 ms of the 5.9 ms launch).
+    constexpr int C4 = CP / 4;
+    constexpr int STOT = SY * SX * C4;
+    constexpr int SLV = (STOT + 511) / 512;
+    typedef cfx_bf16 bf16x4 __attribute__((ext_vector_type(4)));
     auto stage_plane = [&](int P) {
         const int slot = ((P + 1) % 3 + 3) % 3;
         const bool zin = P >= 0 && P < D;
         const bool interior = zin && y0 >= 1 && y0 + TY + 1 <= H &&
                               x0 >= 1 && x0 + TX + 1 <= W;
-        // copy C bf16 per voxel as 4-channel chunks (C % 4 == 0), zero the
-        // pad channels once via the c4 == C/4 slot
-        constexpr int C4 = CP / 4;
-        typedef cfx_bf16 bf16x4 __attribute__((ext_vector_type(4)));
-        for (int idx = tid; idx < SY * SX * C4; idx += 512) {
+        // Loads are UNCONDITIONAL from clamped addresses (any masked or
+        // pad lane reads a safe in-bounds element); the zero-select runs
+        // in the store phase, after the whole batch is in flight — a
+        // select attached to each load forces a full vmcnt(0) drain per
+        // load (exec-masked load + v_cndmask), which is the serialization
+        // this replaces.
+        bf16x4 vals[SLV];
+        bool keep[SLV];
+#pragma unroll
+        for (int li = 0; li < SLV; ++li) {
+            const int idx = min(tid + li * 512, STOT - 1);
             const int c4 = idx % C4;
             const int v = idx / C4;
-            const int sx = v % SX;
-            const int sy = v / SX;
-            const int gy = y0 + sy - 1;
-            const int gx = x0 + sx - 1;
-            bf16x4 val = {};
-            if (c4 * 4 < C) {
-                if (interior) {
-                    val = *reinterpret_cast<const bf16x4*>(
-                        in + ((((long long)n * D + P) * H + gy) * W + gx)
-                                 * C + c4 * 4);
-                } else {
-                    const bool ok = zin && gy >= 0 && gy < H && gx >= 0 &&
-                                    gx < W;
-                    val = *reinterpret_cast<const bf16x4*>(
-                        in + ((((long long)n * D + (zin ? P : 0)) * H +
-                               (ok ? gy : 0)) * W + (ok ? gx : 0)) * C +
-                        c4 * 4);
-                    if (!ok) val = {};
-                }
-            }
+            const int gy = y0 + v / SX - 1;
+            const int gx = x0 + v % SX - 1;
+            const bool cok = c4 * 4 < C;
+            const bool ok = cok && zin &&
+                            (interior || (gy >= 0 && gy < H && gx >= 0 &&
+                                          gx < W));
+            keep[li] = ok;
+            vals[li] = *reinterpret_cast<const bf16x4*>(
+                in + ((((long long)n * D + (zin ? P : 0)) * H +
+                       (ok ? gy : 0)) * W + (ok ? gx : 0)) * C +
+                (cok ? c4 * 4 : 0));
+        }
+#pragma unroll
+        for (int li = 0; li < SLV; ++li) {
+            const int idx = tid + li * 512;
+            if (idx >= STOT) break;
+            const int c4 = idx % C4;
+            const int v = idx / C4;
             *reinterpret_cast<bf16x4*>(
-                &ring[((slot * SY + sy) * SX + sx) * PCB + c4 * 4]) = val;
+                &ring[((slot * SY + v / SX) * SX + v % SX) * PCB +
+                      c4 * 4]) = keep[li] ? vals[li] : bf16x4{};
         }
     };
 
@@ -1237,6 +1252,21 @@ __global__ __launch_bounds__(512, 1) void k_conv3_zring_bf16(
         const int j = lane & 31;
         if (MODE != 2 && gy < H && j < K) {
             const float bj = bias ? bias[j] : 0.f;
+            // residual reads batched up front from clamped addresses
+            // (a load inside the store loop serializes: one vmcnt(0)
+            // drain per 2-byte read)
+            cfx_bf16 rv[16];
+            if (res) {
+#pragma unroll
+                for (int r = 0; r < 16; ++r) {
+                    const int row =
+                        (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
+                    const int gx = min(x0 + row, W - 1);
+                    rv[r] = res[
+                        ((((long long)n * D + z) * H + gy) * W + gx) * K +
+                        j];
+                }
+            }
 #pragma unroll
             for (int r = 0; r < 16; ++r) {
                 const int row = (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
@@ -1245,7 +1275,7 @@ __global__ __launch_bounds__(512, 1) void k_conv3_zring_bf16(
                 long long o =
                     ((((long long)n * D + z) * H + gy) * W + gx) * K + j;
                 float v = acc[r] + bj;
-                if (res) v += (float)res[o];
+                if (res) v += (float)rv[r];
                 if (do_elu) v = v > 0.f ? v : expm1f(v);
                 out[o] = (cfx_bf16)v;
             }
