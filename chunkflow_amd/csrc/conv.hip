@@ -619,6 +619,25 @@ __global__ __launch_bounds__(512, 1) void k_conv3_zring(
         }
 
         // epilogue for this z
+        float rv[M_TILES][NT][4];
+        if (res) {  // residual reads batched from clamped addresses (a
+                    // load in the store loop costs a vmcnt(0) drain each)
+#pragma unroll
+            for (int m = 0; m < M_TILES; ++m) {
+                const int gym = min(y0 + tmy[m], H - 1);
+#pragma unroll
+                for (int t = 0; t < NT; ++t) {
+                    const int jm = min(j0 + t * 16 + colj, K - 1);
+#pragma unroll
+                    for (int r = 0; r < 4; ++r) {
+                        const int gxm = min(x0 + tmx[m] + rbase + r, W - 1);
+                        rv[m][t][r] = res[
+                            ((((long long)n * D + z) * H + gym) * W + gxm) *
+                                K + jm];
+                    }
+                }
+            }
+        }
 #pragma unroll
         for (int m = 0; m < M_TILES; ++m) {
             const int gy = y0 + tmy[m];
@@ -636,7 +655,7 @@ __global__ __launch_bounds__(512, 1) void k_conv3_zring(
                         ((((long long)n * D + z) * H + gy) * W + gx) * K +
                         j;
                     float v = acc[m][t][r] + bj;
-                    if (res) v += res[o];
+                    if (res) v += rv[m][t][r];
                     if (do_elu) v = v > 0.f ? v : expm1f(v);
                     out[o] = v;
                 }
@@ -806,6 +825,25 @@ __global__ __launch_bounds__(512, 1) void k_conv3_zring_pl(
         __syncthreads();
         compute_dzi(2);
 
+        float rv[M_TILES][NT][4];
+        if (res) {  // residual reads batched from clamped addresses (a
+                    // load in the store loop costs a vmcnt(0) drain each)
+#pragma unroll
+            for (int m = 0; m < M_TILES; ++m) {
+                const int gym = min(y0 + tmy[m], H - 1);
+#pragma unroll
+                for (int t = 0; t < NT; ++t) {
+                    const int jm = min(j0 + t * 16 + colj, K - 1);
+#pragma unroll
+                    for (int r = 0; r < 4; ++r) {
+                        const int gxm = min(x0 + tmx[m] + rbase + r, W - 1);
+                        rv[m][t][r] = res[
+                            ((((long long)n * D + z) * H + gym) * W + gxm) *
+                                K + jm];
+                    }
+                }
+            }
+        }
 #pragma unroll
         for (int m = 0; m < M_TILES; ++m) {
             const int gy = y0 + tmy[m];
@@ -823,7 +861,7 @@ __global__ __launch_bounds__(512, 1) void k_conv3_zring_pl(
                         ((((long long)n * D + z) * H + gy) * W + gx) * K +
                         j;
                     float v = acc[m][t][r] + bj;
-                    if (res) v += res[o];
+                    if (res) v += rv[m][t][r];
                     if (do_elu) v = v > 0.f ? v : expm1f(v);
                     out[o] = v;
                 }
