@@ -1362,36 +1362,32 @@ __global__ __launch_bounds__(512, 1) void k_conv3_zring_bf16_pl(
     const bool xy_interior = y0 >= 1 && y0 + TY + 1 <= H && x0 >= 1 &&
                              x0 + TX + 1 <= W;
 
-    auto plane_load = [&](int P, bf16x4 (&vals)[LV]) {
+    // batched unconditional clamped loads + deferred zero-select (see
+    // the plain kernel's stage_plane comment: a per-load select costs a
+    // full vmcnt(0) drain per 8-byte load)
+    auto plane_load = [&](int P, bf16x4 (&vals)[LV], bool (&keep)[LV]) {
         const bool zin = P >= 0 && P < D;
         const bool interior = zin && xy_interior;
 #pragma unroll
         for (int li = 0; li < LV; ++li) {
-            const int idx = tid + li * 512;
-            if (idx >= SY * SX * C4) break;
+            const int idx = min(tid + li * 512, SY * SX * C4 - 1);
             const int c4 = idx % C4;
             const int v = idx / C4;
             const int gy = y0 + v / SX - 1;
             const int gx = x0 + v % SX - 1;
-            vals[li] = {};
-            if (c4 * 4 < C) {
-                if (interior) {
-                    vals[li] = *reinterpret_cast<const bf16x4*>(
-                        in + ((((long long)n * D + P) * H + gy) * W + gx)
-                                 * C + c4 * 4);
-                } else {
-                    const bool ok = zin && gy >= 0 && gy < H && gx >= 0 &&
-                                    gx < W;
-                    vals[li] = *reinterpret_cast<const bf16x4*>(
-                        in + ((((long long)n * D + (zin ? P : 0)) * H +
-                               (ok ? gy : 0)) * W + (ok ? gx : 0)) * C +
-                        c4 * 4);
-                    if (!ok) vals[li] = {};
-                }
-            }
+            const bool cok = c4 * 4 < C;
+            const bool ok = cok && zin &&
+                            (interior || (gy >= 0 && gy < H && gx >= 0 &&
+                                          gx < W));
+            keep[li] = ok;
+            vals[li] = *reinterpret_cast<const bf16x4*>(
+                in + ((((long long)n * D + (zin ? P : 0)) * H +
+                       (ok ? gy : 0)) * W + (ok ? gx : 0)) * C +
+                (cok ? c4 * 4 : 0));
         }
     };
-    auto plane_store = [&](int P, const bf16x4 (&vals)[LV]) {
+    auto plane_store = [&](int P, const bf16x4 (&vals)[LV],
+                           const bool (&keep)[LV]) {
         const int slot = ((P + 1) % 3 + 3) % 3;
 #pragma unroll
         for (int li = 0; li < LV; ++li) {
@@ -1401,16 +1397,17 @@ __global__ __launch_bounds__(512, 1) void k_conv3_zring_bf16_pl(
             const int v = idx / C4;
             *reinterpret_cast<bf16x4*>(
                 &ring[((slot * SY + v / SX) * SX + v % SX) * PCB +
-                      c4 * 4]) = vals[li];
+                      c4 * 4]) = keep[li] ? vals[li] : bf16x4{};
         }
     };
 
     {
         bf16x4 v0[LV], v1[LV];
-        plane_load(-1, v0);
-        plane_load(0, v1);
-        plane_store(-1, v0);
-        plane_store(0, v1);
+        bool k0[LV], k1[LV];
+        plane_load(-1, v0, k0);
+        plane_load(0, v1, k1);
+        plane_store(-1, v0, k0);
+        plane_store(0, v1, k1);
     }
     __syncthreads();
 
@@ -1444,10 +1441,11 @@ __global__ __launch_bounds__(512, 1) void k_conv3_zring_bf16_pl(
         };
 
         bf16x4 vals[LV];
-        plane_load(z + 1, vals);
+        bool keep[LV];
+        plane_load(z + 1, vals, keep);
         compute_dzi(0);
         compute_dzi(1);
-        plane_store(z + 1, vals);
+        plane_store(z + 1, vals, keep);
         __syncthreads();
         compute_dzi(2);
 
@@ -1455,6 +1453,18 @@ __global__ __launch_bounds__(512, 1) void k_conv3_zring_bf16_pl(
         const int j = lane & 31;
         if (gy < H && j < K) {
             const float bj = bias ? bias[j] : 0.f;
+            cfx_bf16 rv[16];
+            if (res) {  // batched residual reads (clamped addresses)
+#pragma unroll
+                for (int r = 0; r < 16; ++r) {
+                    const int row =
+                        (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
+                    const int gx = min(x0 + row, W - 1);
+                    rv[r] = res[
+                        ((((long long)n * D + z) * H + gy) * W + gx) * K +
+                        j];
+                }
+            }
 #pragma unroll
             for (int r = 0; r < 16; ++r) {
                 const int row = (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
@@ -1463,7 +1473,7 @@ __global__ __launch_bounds__(512, 1) void k_conv3_zring_bf16_pl(
                 long long o =
                     ((((long long)n * D + z) * H + gy) * W + gx) * K + j;
                 float v = acc[r] + bj;
-                if (res) v += (float)res[o];
+                if (res) v += (float)rv[r];
                 if (do_elu) v = v > 0.f ? v : expm1f(v);
                 out[o] = (cfx_bf16)v;
             }
