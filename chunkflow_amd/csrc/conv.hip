@@ -776,12 +776,22 @@ __global__ __launch_bounds__(512, 1) void k_conv3_zring_pl(
     const int rbase = (lane >> 4) * 4;
 
     for (int z = 0; z < D; ++z) {
+        // Two accumulator chains per m-tile: consecutive MFMAs on the
+        // SAME accumulator pay the 40-cyc dependent latency against a
+        // 32-cyc issue slot (PMC: 51.6% SQ_WAIT_INST_ANY with the naive
+        // nt-outer order). kk-outer alternates the NT=2 accumulators
+        // (bit-identical sums — each chain keeps its order); NT==1
+        // splits by kk parity into acc/acc2 (a benign reassociation,
+        // folded in the epilogue).
         f32x4 acc[M_TILES][NT];
+        f32x4 acc2[M_TILES];
 #pragma unroll
-        for (int m = 0; m < M_TILES; ++m)
+        for (int m = 0; m < M_TILES; ++m) {
+            acc2[m] = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
             for (int t = 0; t < NT; ++t)
                 acc[m][t] = {0.f, 0.f, 0.f, 0.f};
+        }
 
         auto compute_dzi = [&](int dzi) {
             const int slot = ((z + dzi) % 3 + 3) % 3;  // plane z + dzi - 1
@@ -800,17 +810,21 @@ __global__ __launch_bounds__(512, 1) void k_conv3_zring_pl(
                 }
                 const float* wblk = &wall[tap * NT * C * 16];
 #pragma unroll
-                for (int nt = 0; nt < NT; ++nt) {
+                for (int kk = 0; kk < KK; ++kk) {
+                    float a[M_TILES];
 #pragma unroll
-                    for (int kk = 0; kk < KK; ++kk) {
+                    for (int m = 0; m < M_TILES; ++m)
+                        a[m] = arow[m][kk * 4];
+#pragma unroll
+                    for (int nt = 0; nt < NT; ++nt) {
                         const float b =
                             wblk[(nt * C + kk * 4 + a_k) * 16 + col16];
 #pragma unroll
                         for (int m = 0; m < M_TILES; ++m) {
-                            const float a = arow[m][kk * 4];
-                            acc[m][nt] =
-                                __builtin_amdgcn_mfma_f32_16x16x4f32(
-                                    a, b, acc[m][nt], 0, 0, 0);
+                            f32x4& dst = (NT == 1 && (kk & 1))
+                                             ? acc2[m] : acc[m][nt];
+                            dst = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                                a[m], b, dst, 0, 0, 0);
                         }
                     }
                 }
@@ -861,6 +875,7 @@ __global__ __launch_bounds__(512, 1) void k_conv3_zring_pl(
                         ((((long long)n * D + z) * H + gy) * W + gx) * K +
                         j;
                     float v = acc[m][t][r] + bj;
+                    if (NT == 1) v += acc2[m][r];
                     if (res) v += rv[m][t][r];
                     if (do_elu) v = v > 0.f ? v : expm1f(v);
                     out[o] = v;
