@@ -1496,6 +1496,183 @@ __global__ __launch_bounds__(512, 1) void k_conv3_zring_bf16_pl(
     }
 }
 
+
+// 4-slot ring ("quad"): planes z-1, z, z+1 all RESIDENT when iteration z
+// starts, plane z+2 loads during the full 54-MFMA mainloop and lands in
+// the slot plane z-2 vacated — ONE barrier per z and a whole iteration
+// (~3.5k cyc) of latency cover for the staging loads, vs the 3-slot
+// ring where plane z+1 must land mid-iteration. The x-stride shrinks to
+// PCB2=36 so 4 slots + the wall fit 160 KB LDS exactly (160,128 B);
+// stride 72 B leaves a 2-way conflict on 2 of 32 banks per 8-lane b128
+// phase — measured below the MFMA bound. Accumulation order matches the
+// other bf16 rings bit-for-bit.
+template <int C, int K, int TY, int TX>
+__global__ __launch_bounds__(512, 1) void k_conv3_zring_bf16_q(
+    const cfx_bf16* __restrict__ in, const cfx_bf16* __restrict__ wgt,
+    const float* __restrict__ bias, const cfx_bf16* __restrict__ res,
+    cfx_bf16* __restrict__ out, int N, int D, int H, int W, int do_elu) {
+    constexpr int CP = 32;
+    constexpr int PCB2 = CP + 4;         // 36: 4-slot ring must fit LDS
+    constexpr int SX = TX + 2;
+    constexpr int SY = TY + 2;
+    constexpr int C4 = CP / 4;
+    constexpr int STOT = SY * SX * C4;
+    constexpr int SLV = (STOT + 511) / 512;
+    typedef cfx_bf16 bf16x4 __attribute__((ext_vector_type(4)));
+    static_assert(C <= CP && K <= 32, "");
+
+    __shared__ cfx_bf16 ring[4 * SY * SX * PCB2];
+    __shared__ cfx_bf16 wall[27 * 32 * PCB2];
+
+    const int n = blockIdx.z;
+    const int y0 = blockIdx.y * TY;
+    const int x0 = blockIdx.x * TX;
+    const int tid = threadIdx.x;
+    const int wave = tid >> 6;
+    const int lane = tid & 63;
+
+    for (int idx = tid; idx < 27 * 32 * CP; idx += 512) {
+        const int c = idx % CP;
+        const int j = (idx / CP) % 32;
+        const int tap = idx / (CP * 32);
+        wall[(tap * 32 + j) * PCB2 + c] = wgt[(tap * 32 + j) * 32 + c];
+    }
+
+    const bool xy_interior = y0 >= 1 && y0 + TY + 1 <= H && x0 >= 1 &&
+                             x0 + TX + 1 <= W;
+    auto plane_load = [&](int P, bf16x4 (&vals)[SLV], bool (&keep)[SLV]) {
+        const bool zin = P >= 0 && P < D;
+        const bool interior = zin && xy_interior;
+#pragma unroll
+        for (int li = 0; li < SLV; ++li) {
+            const int idx = min(tid + li * 512, STOT - 1);
+            const int c4 = idx % C4;
+            const int v = idx / C4;
+            const int gy = y0 + v / SX - 1;
+            const int gx = x0 + v % SX - 1;
+            const bool cok = c4 * 4 < C;
+            const bool ok = cok && zin &&
+                            (interior || (gy >= 0 && gy < H && gx >= 0 &&
+                                          gx < W));
+            keep[li] = ok;
+            vals[li] = *reinterpret_cast<const bf16x4*>(
+                in + ((((long long)n * D + (zin ? P : 0)) * H +
+                       (ok ? gy : 0)) * W + (ok ? gx : 0)) * C +
+                (cok ? c4 * 4 : 0));
+        }
+    };
+    auto plane_store = [&](int P, const bf16x4 (&vals)[SLV],
+                           const bool (&keep)[SLV]) {
+        const int slot = (P + 2) & 3;
+#pragma unroll
+        for (int li = 0; li < SLV; ++li) {
+            const int idx = tid + li * 512;
+            if (idx >= STOT) break;
+            const int c4 = idx % C4;
+            const int v = idx / C4;
+            *reinterpret_cast<bf16x4*>(
+                &ring[((slot * SY + v / SX) * SX + v % SX) * PCB2 +
+                      c4 * 4]) = keep[li] ? vals[li] : bf16x4{};
+        }
+    };
+
+    {
+        bf16x4 v0[SLV], v1[SLV], v2[SLV];
+        bool k0[SLV], k1[SLV], k2[SLV];
+        plane_load(-1, v0, k0);
+        plane_load(0, v1, k1);
+        plane_load(1, v2, k2);
+        plane_store(-1, v0, k0);
+        plane_store(0, v1, k1);
+        plane_store(1, v2, k2);
+    }
+    __syncthreads();
+
+    const int ax = lane & 31;
+    const int khalf = (lane >> 5) * 8;
+
+    for (int z = 0; z < D; ++z) {
+        bf16x4 vals[SLV];
+        bool keep[SLV];
+        plane_load(z + 2, vals, keep);  // a full iteration to land
+
+        f32x16 acc = {};
+        {
+            const cfx_bf16* planes[3];
+#pragma unroll
+            for (int dzi = 0; dzi < 3; ++dzi)
+                planes[dzi] =
+                    &ring[((z + dzi + 1) & 3) * SY * SX * PCB2];
+            auto addrA = [&](int p) {
+                const int tap = p >> 1, kk = p & 1;
+                const int dzi = tap / 9, tl = tap % 9;
+                const int dy = tl / 3 - 1, dx = tl % 3 - 1;
+                return reinterpret_cast<const bf16x8*>(
+                    &planes[dzi][((1 + wave + dy) * SX + (1 + dx) + ax) *
+                                     PCB2 + khalf + kk * 16]);
+            };
+            auto addrB = [&](int p) {
+                const int tap = p >> 1, kk = p & 1;
+                return reinterpret_cast<const bf16x8*>(
+                    &wall[(tap * 32 + ax) * PCB2 + khalf + kk * 16]);
+            };
+            constexpr int PD = 4;
+            constexpr int NP = 27 * 2;
+            bf16x8 abuf[PD], bbuf[PD];
+#pragma unroll
+            for (int p = 0; p < PD; ++p) {
+                abuf[p] = *addrA(p);
+                bbuf[p] = *addrB(p);
+            }
+#pragma unroll
+            for (int p = 0; p < NP; ++p) {
+                const int sidx = p % PD;
+                acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                    abuf[sidx], bbuf[sidx], acc, 0, 0, 0);
+                __builtin_amdgcn_sched_barrier(0);
+                if (p + PD < NP) {
+                    abuf[sidx] = *addrA(p + PD);
+                    bbuf[sidx] = *addrB(p + PD);
+                }
+                __builtin_amdgcn_sched_barrier(0);
+            }
+        }
+
+        plane_store(z + 2, vals, keep);
+
+        const int gy = y0 + wave;
+        const int j = lane & 31;
+        if (gy < H && j < K) {
+            const float bj = bias ? bias[j] : 0.f;
+            cfx_bf16 rv[16];
+            if (res) {  // batched residual reads (clamped addresses)
+#pragma unroll
+                for (int r = 0; r < 16; ++r) {
+                    const int row =
+                        (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
+                    const int gx = min(x0 + row, W - 1);
+                    rv[r] = res[
+                        ((((long long)n * D + z) * H + gy) * W + gx) * K +
+                        j];
+                }
+            }
+#pragma unroll
+            for (int r = 0; r < 16; ++r) {
+                const int row = (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
+                const int gx = x0 + row;
+                if (gx >= W) continue;
+                long long o =
+                    ((((long long)n * D + z) * H + gy) * W + gx) * K + j;
+                float v = acc[r] + bj;
+                if (res) v += (float)rv[r];
+                if (do_elu) v = v > 0.f ? v : expm1f(v);
+                out[o] = (cfx_bf16)v;
+            }
+        }
+        __syncthreads();
+    }
+}
+
 }  // namespace
 
 extern "C" int cfx_conv3_ndhwc_bf16(cfx_ctx* ctx, const void* in,
@@ -1521,7 +1698,13 @@ extern "C" int cfx_conv3_ndhwc_bf16(cfx_ctx* ctx, const void* in,
         const char* e = getenv("CFX_BF16_MODE");  // phase ablation (timing)
         return e ? atoi(e) : 0;
     }();
-    if (use_pl)
+    if (use_pl == 2)
+        hipLaunchKernelGGL((k_conv3_zring_bf16_q<28, 28, 8, 32>), grid,
+                           dim3(512), 0, ctx->stream, (const cfx_bf16*)in,
+                           (const cfx_bf16*)wgt, bias,
+                           (const cfx_bf16*)residual, (cfx_bf16*)out, N, D,
+                           H, W, do_elu);
+    else if (use_pl)
         hipLaunchKernelGGL((k_conv3_zring_bf16_pl<28, 28, 8, 32>), grid,
                            dim3(512), 0, ctx->stream, (const cfx_bf16*)in,
                            (const cfx_bf16*)wgt, bias,
