@@ -10,7 +10,6 @@ from __future__ import annotations
 
 import itertools
 from collections import namedtuple
-from math import ceil
 from typing import Union
 
 

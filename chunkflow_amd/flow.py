@@ -10,8 +10,10 @@ operators (reference chunkflow/flow/flow.py):
   inference             :1852-1933
   crop-margin           :2053-2084
   plugin                :1751-1800
-  connected-components  :1803-1829 (host path; GPU kernel is a 'next' row)
+  connected-components  :1803-1829 (gfx950 union-find on device chunks)
   normalize-contrast    :1672-1710 (device histogram/LUT kernels)
+  load-h5 / save-h5     :976-1120  (in-repo HDF5 codec — h5io.py)
+  load-tif / save-tif   :918-974   (in-repo TIFF codec — tiffio.py)
   log-summary           :1633-1647 (voxels/sec report shape)
 plus the at-least-once resume family (skip-task-by-file, mark-complete,
 skip-all-zero, skip-none — SURVEY.md §5), copy-var/delete-var, and npy
@@ -24,7 +26,7 @@ from time import time
 import click
 import numpy as np
 
-from .cartesian import BoundingBox, BoundingBoxes, Cartesian
+from .cartesian import BoundingBox, BoundingBoxes
 from .chunk import Chunk
 from .plugin import Plugin
 from .runtime import (DEFAULT_CHUNK_NAME, default_none, generator,

@@ -7,7 +7,6 @@ k = 8*(l>>5)+e; D (f32x16): col j = l&31, row = (reg&3) + 8*(reg>>2) +
 """
 import os
 import subprocess
-import sys
 
 import numpy as np
 
