@@ -102,3 +102,33 @@ def test_intra_chunk_shard_world2(tmp_path):
     ref = oracle_inference(arr, (10, 32, 32), (2, 8, 8),
                            num_output_channels=2, batch_size=3)
     np.testing.assert_allclose(got, ref, rtol=1e-5, atol=1e-6)
+
+
+@pytest.mark.timeout(240)
+def test_shard_and_stitch_world4(tmp_path):
+    """World 4 — the same fan-in rank 0 sees on the 8-GPU node (every
+    non-zero rank sends concurrently)."""
+    result = str(tmp_path / 'vol4.npy')
+    mp.spawn(_worker, args=(4, 29537, result), nprocs=4, join=True)
+    vol = np.load(result)
+    assert vol.shape == (3, 4, 8, 32)
+    for i in range(4):
+        np.testing.assert_array_equal(vol[:, :, :, i * 8:(i + 1) * 8],
+                                      np.full((3, 4, 8, 8), i + 1,
+                                              dtype=np.float32))
+
+
+@pytest.mark.timeout(240)
+def test_intra_chunk_shard_world3(tmp_path):
+    """Intra-chunk sharding with a world size that does NOT divide the
+    patch count evenly (ragged shard)."""
+    from oracle import oracle_inference
+    rng = np.random.RandomState(23)
+    arr = rng.randint(0, 256, size=(20, 68, 72), dtype=np.uint8)
+    result = str(tmp_path / 'intra3.npy')
+    np.save(result + '.in.npy', arr)
+    mp.spawn(_intra_worker, args=(3, 29539, result), nprocs=3, join=True)
+    got = np.load(result)
+    ref = oracle_inference(arr, (10, 32, 32), (2, 8, 8),
+                           num_output_channels=2, batch_size=3)
+    np.testing.assert_allclose(got, ref, rtol=1e-5, atol=1e-6)
