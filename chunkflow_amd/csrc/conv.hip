@@ -672,7 +672,9 @@ __global__ __launch_bounds__(512, 1) void k_conv3_zring(
 // 9 taps. One barrier per z (the plain kernel needs two): iteration z's
 // store targets the slot plane z-2 occupied, whose last readers (dzi=0 of
 // iteration z-1) sit before iteration z-1's barrier.
-template <int C, int K, int TY, int TX, int NTK>
+// MODE (ablation, CFX_F32_MODE): 0 full, 1 skip mainloop, 3 skip per-z
+// staging (timing only, wrong results)
+template <int C, int K, int TY, int TX, int NTK, int MODE = 0>
 __global__ __launch_bounds__(512, 1) void k_conv3_zring_pl(
     const float* __restrict__ in, const float* __restrict__ wgt,
     const float* __restrict__ bias, const float* __restrict__ res,
@@ -713,34 +715,30 @@ __global__ __launch_bounds__(512, 1) void k_conv3_zring_pl(
     const bool xy_interior = y0 >= 1 && y0 + TY + 1 <= H && x0 >= 1 &&
                              x0 + TX + 1 <= W;
 
-    // clamp-then-zero load of this thread's share of global plane P
-    auto plane_load = [&](int P, f32x4 (&vals)[LV]) {
+    // Unconditional clamped loads + deferred zero-select (see the bf16
+    // twin: a select attached to each load makes hipcc emit a full
+    // vmcnt(0) drain per load, serializing the HBM latency)
+    auto plane_load = [&](int P, f32x4 (&vals)[LV], bool (&keep)[LV]) {
         const bool zin = P >= 0 && P < D;
         const bool interior = zin && xy_interior;
 #pragma unroll
         for (int li = 0; li < LV; ++li) {
-            const int idx = tid + li * 512;
-            if (idx >= SY * SX * C4N) break;
+            const int idx = min(tid + li * 512, SY * SX * C4N - 1);
             const int c4 = idx % C4N;
             const int v = idx / C4N;
             const int gy = y0 + v / SX - 1;
             const int gx = x0 + v % SX - 1;
-            if (interior) {
-                vals[li] = *reinterpret_cast<const f32x4*>(
-                    in + ((((long long)n * D + P) * H + gy) * W + gx) * C +
-                    c4 * 4);
-            } else {
-                const bool ok = zin && gy >= 0 && gy < H && gx >= 0 &&
-                                gx < W;
-                vals[li] = *reinterpret_cast<const f32x4*>(
-                    in + ((((long long)n * D + (zin ? P : 0)) * H +
-                           (ok ? gy : 0)) * W + (ok ? gx : 0)) * C +
-                    c4 * 4);
-                if (!ok) vals[li] = {0.f, 0.f, 0.f, 0.f};
-            }
+            const bool ok = zin && (interior ||
+                                    (gy >= 0 && gy < H && gx >= 0 &&
+                                     gx < W));
+            keep[li] = ok;
+            vals[li] = *reinterpret_cast<const f32x4*>(
+                in + ((((long long)n * D + (zin ? P : 0)) * H +
+                       (ok ? gy : 0)) * W + (ok ? gx : 0)) * C + c4 * 4);
         }
     };
-    auto plane_store = [&](int P, const f32x4 (&vals)[LV]) {
+    auto plane_store = [&](int P, const f32x4 (&vals)[LV],
+                           const bool (&keep)[LV]) {
         const int slot = ((P + 1) % 3 + 3) % 3;
 #pragma unroll
         for (int li = 0; li < LV; ++li) {
@@ -749,16 +747,18 @@ __global__ __launch_bounds__(512, 1) void k_conv3_zring_pl(
             const int c4 = idx % C4N;
             const int v = idx / C4N;
             *reinterpret_cast<f32x4*>(
-                &ring[(slot * SY * SX + v) * PC + c4 * 4]) = vals[li];
+                &ring[(slot * SY * SX + v) * PC + c4 * 4]) =
+                keep[li] ? vals[li] : f32x4{0.f, 0.f, 0.f, 0.f};
         }
     };
 
     {
         f32x4 v0[LV], v1[LV];
-        plane_load(-1, v0);
-        plane_load(0, v1);
-        plane_store(-1, v0);
-        plane_store(0, v1);
+        bool k0[LV], k1[LV];
+        plane_load(-1, v0, k0);
+        plane_load(0, v1, k1);
+        plane_store(-1, v0, k0);
+        plane_store(0, v1, k1);
     }
     __syncthreads();
 
@@ -832,12 +832,16 @@ __global__ __launch_bounds__(512, 1) void k_conv3_zring_pl(
         };
 
         f32x4 vals[LV];
-        plane_load(z + 1, vals);   // in flight across the next 18 taps
-        compute_dzi(0);
-        compute_dzi(1);
-        plane_store(z + 1, vals);
+        bool keep[LV];
+        if (MODE != 3)  // flies over the next 18 taps
+            plane_load(z + 1, vals, keep);
+        if (MODE != 1) {
+            compute_dzi(0);
+            compute_dzi(1);
+        }
+        if (MODE != 3) plane_store(z + 1, vals, keep);
         __syncthreads();
-        compute_dzi(2);
+        if (MODE != 1) compute_dzi(2);
 
         float rv[M_TILES][NT][4];
         if (res) {  // residual reads batched from clamped addresses (a
@@ -1091,11 +1095,23 @@ extern "C" int cfx_conv3_ndhwc_zring(cfx_ctx* ctx, const float* in,
         const char* e = getenv("CFX_ZRING_PL");
         return e ? atoi(e) : 1;
     }();
+    static const int f32mode = [] {
+        const char* e = getenv("CFX_F32_MODE");  // phase ablation (timing)
+        return e ? atoi(e) : 0;
+    }();
     hipEvent_t e0;
     if (prof_begin(ctx, &e0)) return -1;
     if (C == 28 && K == 28) {
         // wall holds both K tiles (96.8 KB + 60.5 KB ring)
-        if (use_pl)
+        if (f32mode == 1)
+            hipLaunchKernelGGL((k_conv3_zring_pl<28, 28, 8, 16, 2, 1>),
+                               grid, dim3(512), 0, ctx->stream, in, wgt,
+                               bias, residual, out, N, D, H, W, do_elu, 0);
+        else if (f32mode == 3)
+            hipLaunchKernelGGL((k_conv3_zring_pl<28, 28, 8, 16, 2, 3>),
+                               grid, dim3(512), 0, ctx->stream, in, wgt,
+                               bias, residual, out, N, D, H, W, do_elu, 0);
+        else if (use_pl)
             hipLaunchKernelGGL((k_conv3_zring_pl<28, 28, 8, 16, 2>), grid,
                                dim3(512), 0, ctx->stream, in, wgt, bias,
                                residual, out, N, D, H, W, do_elu, 0);
