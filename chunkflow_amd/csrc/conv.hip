@@ -1447,38 +1447,70 @@ __global__ __launch_bounds__(512, 1) void k_conv3_zring_bf16_pl(
 
     for (int z = 0; z < D; ++z) {
         f32x16 acc = {};
-        auto compute_dzi = [&](int dzi) {
-            const int slot = ((z + dzi) % 3 + 3) % 3;
-            const cfx_bf16* plane = &ring[slot * SY * SX * PCB];
+        // fragment addressing for the flat pair pipeline (pair p: tap
+        // p>>1, kk p&1); same PD=4 prefetch-distance scheme as the plain
+        // kernel, split at pair 36 around the plane barrier
+        const cfx_bf16* planes[3];
 #pragma unroll
-            for (int tl = 0; tl < 9; ++tl) {
-                const int tap = dzi * 9 + tl;
-                const int dy = tl / 3 - 1;
-                const int dx = tl % 3 - 1;
-                const cfx_bf16* arow =
-                    &plane[((1 + wave + dy) * SX + (1 + dx) + ax) * PCB +
-                           khalf];
-                const cfx_bf16* brow = &wall[(tap * 32 + ax) * PCB + khalf];
-#pragma unroll
-                for (int kk = 0; kk < KK; ++kk) {
-                    const bf16x8 a =
-                        *reinterpret_cast<const bf16x8*>(arow + kk * 16);
-                    const bf16x8 b =
-                        *reinterpret_cast<const bf16x8*>(brow + kk * 16);
-                    acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, acc,
-                                                                  0, 0, 0);
-                }
-            }
+        for (int dzi = 0; dzi < 3; ++dzi)
+            planes[dzi] = &ring[(((z + dzi) % 3 + 3) % 3) * SY * SX * PCB];
+        auto addrA = [&](int p) {
+            const int tap = p >> 1, kk = p & 1;
+            const int dzi = tap / 9, tl = tap % 9;
+            const int dy = tl / 3 - 1, dx = tl % 3 - 1;
+            return reinterpret_cast<const bf16x8*>(
+                &planes[dzi][((1 + wave + dy) * SX + (1 + dx) + ax) * PCB +
+                             khalf + kk * 16]);
         };
+        auto addrB = [&](int p) {
+            const int tap = p >> 1, kk = p & 1;
+            return reinterpret_cast<const bf16x8*>(
+                &wall[(tap * 32 + ax) * PCB + khalf + kk * 16]);
+        };
+        constexpr int PD = 4;
 
         bf16x4 vals[LV];
         bool keep[LV];
         plane_load(z + 1, vals, keep);
-        compute_dzi(0);
-        compute_dzi(1);
-        plane_store(z + 1, vals, keep);
-        __syncthreads();
-        compute_dzi(2);
+        {
+            bf16x8 abuf[PD], bbuf[PD];
+#pragma unroll
+            for (int p = 0; p < PD; ++p) {
+                abuf[p] = *addrA(p);
+                bbuf[p] = *addrB(p);
+            }
+#pragma unroll
+            for (int p = 0; p < 36; ++p) {  // dzi 0,1: planes z-1, z
+                const int si = p % PD;
+                acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                    abuf[si], bbuf[si], acc, 0, 0, 0);
+                __builtin_amdgcn_sched_barrier(0);
+                if (p + PD < 36) {
+                    abuf[si] = *addrA(p + PD);
+                    bbuf[si] = *addrB(p + PD);
+                }
+                __builtin_amdgcn_sched_barrier(0);
+            }
+            plane_store(z + 1, vals, keep);
+            __syncthreads();
+#pragma unroll
+            for (int p = 36; p < 36 + PD; ++p) {
+                abuf[p % PD] = *addrA(p);
+                bbuf[p % PD] = *addrB(p);
+            }
+#pragma unroll
+            for (int p = 36; p < 54; ++p) {  // dzi 2: plane z + 1
+                const int si = p % PD;
+                acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                    abuf[si], bbuf[si], acc, 0, 0, 0);
+                __builtin_amdgcn_sched_barrier(0);
+                if (p + PD < 54) {
+                    abuf[si] = *addrA(p + PD);
+                    bbuf[si] = *addrB(p + PD);
+                }
+                __builtin_amdgcn_sched_barrier(0);
+            }
+        }
 
         const int gy = y0 + wave;
         const int j = lane & 31;
