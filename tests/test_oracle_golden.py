@@ -143,3 +143,32 @@ def test_oracle_vs_live_reference():
                            offset=(3, 4, 5))
     np.testing.assert_allclose(out, np.asarray(ref_out.array),
                                rtol=1e-6, atol=1e-7)
+
+
+@pytest.mark.skipif(not __import__('os').path.isdir('/root/reference'),
+                    reason='reference checkout not present (GPU box)')
+@pytest.mark.parametrize('seed', [101, 202, 303])
+def test_oracle_vs_live_reference_fuzz(seed):
+    """Randomized geometries against the LIVE reference (build container
+    only): chunk/patch/overlap drawn per seed, identity engine, masked
+    mode, random offset."""
+    from oracle.ref_harness import import_reference
+    Inferencer, Chunk, _ = import_reference()
+    rng = np.random.RandomState(seed)
+    patch = tuple(int(rng.randint(6, 13) * 2) for _ in range(3))
+    ov = tuple(max(2, (p // 4) // 2 * 2) for p in patch)
+    chunk = tuple(p + int(rng.randint(0, 3)) * (p - o)
+                  + int(rng.randint(0, 4))
+                  for p, o in zip(patch, ov))
+    off = tuple(int(v) for v in rng.randint(0, 9, size=3))
+    nch = int(rng.randint(1, 4))
+    bs = int(rng.randint(1, 5))
+    arr = rng.randint(0, 256, size=chunk).astype(np.uint8)
+    with Inferencer(None, None, patch, output_patch_overlap=ov,
+                    framework='identity', num_output_channels=nch,
+                    batch_size=bs, mask_output_chunk=True) as inf:
+        ref_out = inf(Chunk(arr.copy(), voxel_offset=off))
+    got = oracle_inference(arr, patch, ov, num_output_channels=nch,
+                           batch_size=bs, offset=off)
+    np.testing.assert_allclose(got, np.asarray(ref_out.array),
+                               rtol=1e-6, atol=1e-7)
