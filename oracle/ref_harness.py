@@ -42,6 +42,7 @@ def install_stubs():
     stub('h5py')
     stub('tifffile')
     stub('cc3d')
+    stub('cv2')  # chunk/image/adjust_grey.py imports it at module scope
     cv = stub('cloudvolume', CloudVolume=None)
     cvlib = stub('cloudvolume.lib', Bbox=_Bbox, Vec=_Vec,
                  yellow=lambda s: s)

@@ -335,3 +335,26 @@ def test_fastconv_eligibility_rules():
     blocks = [m for m in mod.InstantiatedModel.modules()
               if _resblock_like(m)]
     assert len(blocks) == 4  # the two 28- and two 36-wide ResBlocks
+
+
+@pytest.mark.skipif(not os.path.isdir('/root/reference'),
+                    reason='reference checkout not present (GPU box)')
+@pytest.mark.parametrize('seed', [5, 6, 7, 11])
+def test_normalize_contrast_vs_live_reference(seed):
+    """The contrast oracle is BIT-EQUAL to the live reference
+    Image.normalize_contrast on random uint8 volumes (including the
+    for-else whole-chunk second pass, chunk/image/base.py:30-132)."""
+    import sys
+    from oracle.ref_harness import install_stubs, REFERENCE_PATH
+    install_stubs()
+    if REFERENCE_PATH not in sys.path:
+        sys.path.insert(0, REFERENCE_PATH)
+    from chunkflow.chunk.image import Image as RefImage
+    from oracle.contrast import oracle_normalize_contrast
+    rng = np.random.RandomState(seed)
+    shape = tuple(int(v) for v in rng.randint(3, 20, size=3))
+    arr = rng.randint(0, 256, size=shape).astype(np.uint8)
+    img = RefImage(arr.copy())
+    img.normalize_contrast()
+    mine = oracle_normalize_contrast(arr.copy())
+    np.testing.assert_array_equal(mine, np.asarray(img.array))
