@@ -68,17 +68,25 @@ class Chunk:
                 arr = np.repeat(arr[None, ...], size[0], axis=0)
             if dtype == np.uint8:
                 arr = (arr * 255).astype(dtype)
+            elif dtype in (np.uint16, np.uint32, np.uint64):
+                # reference base.py:180-181: threshold then cc3d-label so
+                # the synthetic segmentation has per-component ids
+                from .connected import equal_value_label
+                arr = equal_value_label((arr > 0.5).astype(np.uint8), 6)
             elif np.issubdtype(dtype, np.floating):
                 arr = arr.astype(dtype)
             else:
                 raise NotImplementedError(
-                    f'sin pattern for {dtype} needs segmentation labeling '
-                    '(out of hot-path scope)')
+                    f'do not support this data type: {dtype}')
         elif pattern == 'random':
             if np.issubdtype(dtype, np.floating):
                 arr = np.random.rand(*size).astype(dtype)
             elif np.issubdtype(dtype, np.integer):
-                arr = np.random.randint(high, size=size, dtype=dtype)
+                # reference base.py:192-193 relabels with cc3d so values
+                # are per-component ids, not raw randints
+                from .connected import equal_value_label
+                arr = equal_value_label(
+                    np.random.randint(high, size=size, dtype=dtype), 6)
             else:
                 raise NotImplementedError(dtype)
         else:
@@ -340,8 +348,10 @@ class Chunk:
         a missing/empty per-bbox file returns None."""
         from . import h5io
         file_name = os.path.expanduser(file_name)
-        if cutout_start is not None and cutout_size is not None and \
-                cutout_stop is None:
+        if cutout_start is not None and cutout_size is not None:
+            # reference computes stop EARLY here (base.py:281-282), before
+            # the negative-entry fill below, so negative sizes combined
+            # with an explicit start still yield an empty cutout there too
             cutout_stop = tuple(t + s for t, s in
                                 zip(cutout_start, cutout_size))
         if not (os.path.isfile(file_name) and h5io.is_hdf5(file_name)):
@@ -369,6 +379,14 @@ class Chunk:
             cutout_start = tuple(voxel_offset)
         if cutout_size is None:
             cutout_size = arr.shape[-3:]
+        elif min(cutout_size) < 0:
+            # reference base.py:328-333: negative entries are filled from
+            # the dataset shape (only effective when stop wasn't set above)
+            cutout_size = list(cutout_size)
+            for idx in range(-1, -4, -1):
+                if cutout_size[idx] < 0:
+                    cutout_size[idx] = arr.shape[idx]
+            cutout_size = tuple(cutout_size)
         if cutout_stop is None:
             cutout_stop = tuple(t + s for t, s in
                                 zip(cutout_start, cutout_size))

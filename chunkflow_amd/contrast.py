@@ -88,24 +88,37 @@ def normalize_contrast(chunk: Chunk, lower_clip_fraction=0.01,
         D = t.shape[0]
         n_per_sec = t.shape[1] * t.shape[2]
 
-        def lut_pass(nsec, nps):
-            hist_dev = torch.empty((nsec, 256), dtype=torch.int32,
-                                   device=t.device)
-            ops.cfx.hist_u8(t.data_ptr(), nps, nsec, hist_dev.data_ptr())
-            hist = hist_dev.cpu().numpy().astype(np.int64)
+        # int32 device bins bound each count by voxels-per-section; the
+        # whole-chunk pass's histogram is derived on host in int64 by
+        # pushing the per-section histograms through their LUTs (exact:
+        # the apply is pointwise v -> lut[v]), so no 2^31 whole-chunk limit
+        assert n_per_sec < 2**31, \
+            'per-section voxel count overflows the int32 device histogram'
+        identity = np.arange(256, dtype=np.uint8)
+
+        def luts_from_hist(hist, nsec):
             luts = np.empty((nsec, 256), dtype=np.uint8)
-            identity = np.arange(256, dtype=np.uint8)
             for s in range(nsec):
                 lut = hist_to_lut(_bincount_like(hist[s]),
                                   lower_clip_fraction, upper_clip_fraction,
                                   minval=minval, maxval=maxval)
                 luts[s] = identity if lut is None else lut
-            lut_dev = torch.from_numpy(luts).to(t.device)
-            ops.cfx.lut_apply_u8(t.data_ptr(), nps, nsec,
-                                 lut_dev.data_ptr())
+            return luts
 
-        lut_pass(D, n_per_sec)              # per-section pass
-        lut_pass(1, D * n_per_sec)          # the for-else whole-chunk pass
+        hist_dev = torch.empty((D, 256), dtype=torch.int32, device=t.device)
+        ops.cfx.hist_u8(t.data_ptr(), n_per_sec, D, hist_dev.data_ptr())
+        hist = hist_dev.cpu().numpy().astype(np.int64)
+        luts = luts_from_hist(hist, D)                  # per-section pass
+        lut_dev = torch.from_numpy(luts).to(t.device)
+        ops.cfx.lut_apply_u8(t.data_ptr(), n_per_sec, D, lut_dev.data_ptr())
+
+        whole = np.zeros((1, 256), dtype=np.int64)      # post-apply hist
+        for s in range(D):
+            np.add.at(whole[0], luts[s], hist[s])
+        luts_w = luts_from_hist(whole, 1)               # for-else whole pass
+        lut_w_dev = torch.from_numpy(luts_w).to(t.device)
+        ops.cfx.lut_apply_u8(t.data_ptr(), D * n_per_sec, 1,
+                             lut_w_dev.data_ptr())
         return chunk
 
     arr = chunk.numpy().array
