@@ -219,8 +219,13 @@ def normalize_contrast(tasks, name, input_chunk_name, output_chunk_name,
               help='number of input channels')
 @click.option('--num-output-channels', '-c', type=click.INT, default=3,
               help='number of output channels')
-@click.option('--dtype', '-d', type=click.Choice(['float32', 'float16']),
-              default='float32', help='output data type.')
+@click.option('--dtype', '-d',
+              type=click.Choice(['float32', 'float16', 'bfloat16']),
+              default='float32',
+              help='output data type. bfloat16 is an MI355X extension of '
+                   'the reference choice (flow.py:1871-1875): compute runs '
+                   'in bf16 on the MFMA cores while blending stays f32, so '
+                   'the output chunk is float32 (numpy has no bfloat16).')
 @click.option('--framework', '-f',
               type=click.Choice(['universal', 'identity', 'pytorch']),
               default='universal', help='inference framework')

@@ -151,3 +151,27 @@ def test_universal_example_file(golden):
     sin = C.create(size=(20, 68, 72), dtype='uint8', pattern='sin').array
     np.testing.assert_allclose(got[0], sin.astype(np.float32) / 255.0,
                                rtol=1e-5, atol=1e-5)
+
+
+def test_inference_dtype_bfloat16(tmp_path, golden_dir, golden):
+    """--dtype bfloat16 (documented MI355X extension of the reference's
+    float32|float16 choice): bf16 compute, f32 blended output."""
+    out = tmp_path / 'out.npy'
+    run_cli(['create-chunk', '--size', '20', '68', '72',
+             '--dtype', 'uint8', '--pattern', 'sin',
+             'inference', '-s', '10', '32', '32',
+             '--output-patch-overlap', '2', '8', '8',
+             '--framework', 'pytorch', '--batch-size', '1',
+             '--dtype', 'bfloat16',
+             '-m', os.path.join(golden_dir, 'ref_model.py'),
+             '-w', os.path.join(golden_dir, 'ref_model_weights.pt'),
+             '--num-output-channels', '3', '--mask-output-chunk',
+             'save-npy', '-f', str(out)])
+    got = np.load(out)
+    assert got.shape == (3, 20, 68, 72) and got.dtype == np.float32
+    # bf16 compute through the 2-layer conv agrees with the committed f32
+    # reference output to bf16 accumulation error (loose per-element bound,
+    # tight in the mean)
+    ref = golden[1]['e2e_pytorch_out']
+    np.testing.assert_allclose(got, ref, rtol=0.15, atol=0.06)
+    assert float(np.abs(got - ref).mean()) < 2e-2
