@@ -55,10 +55,18 @@ from .runtime import (DEFAULT_CHUNK_NAME, default_none, generator,
               help='starting index of task list.')
 @click.option('--task-index-stop', '-p', type=click.INT, default=None,
               help='stop index of task list.')
+@click.option('--task-rank', type=click.INT, default=None,
+              help='worker rank for multi-GPU sharding (tasks with index % '
+                   'task-world == task-rank are kept). Defaults to the RANK '
+                   'env var under torchrun — the slurm-array-index analog '
+                   'of reference flow.py:151-156.')
+@click.option('--task-world', type=click.INT, default=None,
+              help='number of workers for multi-GPU sharding. Defaults to '
+                   'the WORLD_SIZE env var under torchrun.')
 @generator
 def generate_tasks(roi_start, roi_stop, roi_size, chunk_size, bounding_box,
                    grid_size, file_path, respect_chunk_size,
-                   task_index_start, task_index_stop):
+                   task_index_start, task_index_stop, task_rank, task_world):
     """Generate a batch of tasks."""
     if bounding_box is not None:
         bboxes = [BoundingBox.from_string(bounding_box)]
@@ -77,9 +85,20 @@ def generate_tasks(roi_start, roi_stop, roi_size, chunk_size, bounding_box,
         arr = np.array([list(b.start) + list(b.stop) for b in bboxes],
                        dtype=np.int64)
         np.save(file_path, arr)
+    # rank-aware sharding (torchrun worker mode, BASELINE config 3):
+    # every rank computes the SAME full bbox list (kept in state for the
+    # stitch operator), then keeps indices i % world == rank
+    if task_rank is None:
+        task_rank = int(os.environ.get('RANK', '0'))
+    if task_world is None:
+        task_world = int(os.environ.get('WORLD_SIZE', '1'))
+    assert 0 <= task_rank < task_world
+    state['all_bboxes'] = list(bboxes)
     bbox_num = len(bboxes)
     print(f'total number of tasks: {bbox_num}')
     for bbox_index, bbox in enumerate(bboxes):
+        if bbox_index % task_world != task_rank:
+            continue
         task = get_initial_task()
         task['bbox'] = bbox
         task['bbox_index'] = bbox_index
@@ -400,6 +419,74 @@ def save_npy(tasks, name, file_name, input_chunk_name):
                 fname = fname.replace('{bbox}', task['bbox'].string)
             task[input_chunk_name].to_npy(fname)
             task['log']['timer'][name] = time() - start
+        yield task
+
+
+@main.command('stitch')
+@click.option('--name', type=str, default='stitch', help='name of operator')
+@click.option('--input-chunk-name', '-i', type=str,
+              default=DEFAULT_CHUNK_NAME, help='input chunk name')
+@click.option('--output-volume-name', '-o', type=str, default='volume',
+              help='name of the stitched volume in the final task')
+@click.option('--file-name', '-f', type=str, default=None,
+              help='save the stitched volume as .npy on rank 0')
+@click.option('--backend', type=str, default=None,
+              help='torch.distributed backend override (tests use gloo; '
+                   'default nccl=RCCL on GPU, gloo otherwise)')
+@operator
+def stitch(tasks, name, input_chunk_name, output_volume_name, file_name,
+           backend):
+    """Gather per-rank output chunks into one volume on rank 0.
+
+    The terminal operator of the multi-GPU worker mode (BASELINE config 3):
+    each torchrun rank runs the same pipeline over its task shard
+    (generate-tasks --task-rank/--task-world); this op collects the rank's
+    output chunks, then performs the single RCCL-over-xGMI collective of
+    the whole pipeline — a p2p gather of disjoint sub-volumes to rank 0
+    (dispatch.stitch_to_rank0, the same code bench.py times). Rank 0 yields
+    one final task carrying the stitched volume; other ranks yield nothing.
+    """
+    import torch
+    from .dispatch import init_distributed, stitch_to_rank0
+    collected = {}
+    last_task = None
+    for task in tasks:
+        if task is not None:
+            start = time()
+            chunk = task[input_chunk_name]
+            t = chunk.array
+            if not isinstance(t, torch.Tensor):
+                t = torch.from_numpy(np.ascontiguousarray(t))
+            if t.dtype != torch.float32:
+                t = t.to(torch.float32)
+            if t.ndim == 3:
+                t = t[None]
+            idx = task.get('bbox_index', len(collected))
+            collected[idx] = (task.get('bbox'), t)
+            last_task = task
+            last_task['log']['timer'][name] = time() - start
+    rank, world = init_distributed(backend=backend)
+    all_bboxes = state.get('all_bboxes')
+    if all_bboxes is None:
+        assert world == 1, \
+            'stitch with world_size>1 needs generate-tasks in the pipeline'
+        all_bboxes = [b for b, _ in
+                      (collected[i] for i in sorted(collected))]
+    if not collected:
+        return
+    channels = next(iter(collected.values()))[1].shape[0]
+    device = next(iter(collected.values()))[1].device
+    local = {i: t for i, (_, t) in collected.items()}
+    volume = stitch_to_rank0(all_bboxes, local, channels, rank, world,
+                             device)
+    if rank == 0:
+        roi_start = tuple(min(b.start[d] for b in all_bboxes)
+                          for d in range(3))
+        vol_chunk = Chunk(volume, voxel_offset=roi_start)
+        if file_name:
+            vol_chunk.to_npy(file_name)
+        task = last_task or get_initial_task()
+        task[output_volume_name] = vol_chunk
         yield task
 
 

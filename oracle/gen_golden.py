@@ -160,6 +160,67 @@ def main():
         'weights': 'tests/golden/ref_model_weights.pt',
         'patch_size': (10, 32, 32), 'overlap': (2, 8, 8), 'batch_size': 1}
 
+    # --- TTA (augment=True) goldens: identity + conv engines ---------------
+    # (reference transform.py:114-156 driven by inferencer.py:420-431)
+    with Inferencer(None, None, (10, 32, 32),
+                    output_patch_overlap=(2, 8, 8), framework='identity',
+                    num_output_channels=3, batch_size=3,
+                    mask_output_chunk=True, augment=True) as inferencer:
+        out = inferencer(Chunk(chunk_arr.copy(), voxel_offset=(0, 0, 0)))
+    arrays['e2e_identity_augment_out'] = np.asarray(out.array)
+
+    with Inferencer(model_path, weight_path, (10, 32, 32),
+                    output_patch_overlap=(2, 8, 8), framework='pytorch',
+                    num_output_channels=3, batch_size=1,
+                    mask_output_chunk=True, augment=True) as inferencer:
+        out = inferencer(Chunk(chunk_arr.copy(), voxel_offset=(0, 0, 0)))
+    arrays['e2e_pytorch_augment_out'] = np.asarray(out.array)
+    meta['cases']['e2e_augment'] = {
+        'patch_size': (10, 32, 32), 'overlap': (2, 8, 8),
+        'note': 'augment=True goldens for identity (batch 3) and the '
+                '2-layer conv net (batch 1), same input/geometry as e2e'}
+
+    # --- full-RSUNet conv parity pin at a config-2-like geometry -----------
+    # (the real benchmark net through the reference
+    # Inferencer on torch-CPU; patch/pytorch.py:98-119 semantics)
+    rs_model = os.path.join(HERE, '..', 'examples', 'nets', 'rsunet.py')
+    rs_weights = os.path.join(GOLDEN_DIR, 'rsunet_weights.pt')
+    spec2 = importlib.util.spec_from_file_location('golden_rsunet', rs_model)
+    mod2 = importlib.util.module_from_spec(spec2)
+    spec2.loader.exec_module(mod2)
+    torch.save(mod2.InstantiatedModel.state_dict(), rs_weights)
+    wsum = {k: float(v.to(torch.float64).sum())
+            for k, v in mod2.InstantiatedModel.state_dict().items()}
+
+    rs_in = Chunk.create(size=(64, 256, 256), dtype='uint8',
+                         pattern='sin').array
+    with Inferencer(rs_model, rs_weights, (20, 128, 128),
+                    output_patch_overlap=(4, 32, 32), framework='pytorch',
+                    num_output_channels=3, batch_size=1,
+                    mask_output_chunk=True) as inferencer:
+        out = inferencer(Chunk(rs_in.copy(), voxel_offset=(0, 0, 0)))
+    rs_out = np.asarray(out.array)
+    assert rs_out.shape == (3, 64, 256, 256)
+    # full array is 50 MB — commit a strided subsample + a seeded random
+    # voxel sample + f64 global stats instead
+    arrays['rsunet_64x256x256_sub'] = rs_out[:, ::4, ::8, ::8].copy()
+    rng2 = np.random.RandomState(123)
+    idx = rng2.choice(rs_out.size, size=8192, replace=False)
+    idx.sort()
+    arrays['rsunet_64x256x256_sample_idx'] = idx.astype(np.int64)
+    arrays['rsunet_64x256x256_sample_val'] = rs_out.ravel()[idx].copy()
+    meta['cases']['rsunet_64x256x256'] = {
+        'model': 'examples/nets/rsunet.py',
+        'weights': 'tests/golden/rsunet_weights.pt',
+        'weight_f64_sums': wsum,
+        'input': "Chunk.create((64,256,256), uint8, pattern='sin')",
+        'input_crc32': crc(rs_in),
+        'patch_size': (20, 128, 128), 'overlap': (4, 32, 32),
+        'batch_size': 1,
+        'sum_f64': float(rs_out.astype(np.float64).sum()),
+        'min': float(rs_out.min()), 'max': float(rs_out.max()),
+    }
+
     np.savez_compressed(os.path.join(GOLDEN_DIR, 'golden.npz'), **arrays)
     with open(os.path.join(GOLDEN_DIR, 'golden.json'), 'w') as f:
         json.dump(meta, f, indent=1, default=str)
