@@ -132,3 +132,112 @@ def test_intra_chunk_shard_world3(tmp_path):
     ref = oracle_inference(arr, (10, 32, 32), (2, 8, 8),
                            num_output_channels=2, batch_size=3)
     np.testing.assert_allclose(got, ref, rtol=1e-5, atol=1e-6)
+
+
+# --------------------------------------------------------------------------
+# CLI-level worker mode (VERDICT r01 item 5): generate-tasks --task-rank /
+# --task-world sharding + the stitch operator, world 2 over gloo — the
+# config-3 pipeline end-to-end from the operator surface.
+# --------------------------------------------------------------------------
+PLUGIN_SRC = '''\
+import numpy as np
+def execute(chunk):
+    # bbox-dependent fill so a misplaced stitch block is visible
+    v = (sum(chunk.voxel_offset) % 200) + 1
+    return np.full(chunk.shape, v, dtype=np.uint8)
+'''
+
+
+def _cli_pipeline_args(plugin_path, out_npy):
+    return ['generate-tasks', '-c', '4', '8', '8',
+            '--roi-size', '4', '8', '32',
+            'create-chunk',
+            'plugin', '-f', plugin_path, '-i', 'chunk', '-o', 'chunk',
+            'inference', '-s', '4', '8', '8',
+            '--output-patch-overlap', '1', '2', '2',
+            '--framework', 'identity', '--num-output-channels', '3',
+            '--mask-output-chunk',
+            'stitch', '--backend', 'gloo', '-f', out_npy]
+
+
+@pytest.mark.timeout(240)
+def test_cli_worker_mode_world2(tmp_path):
+    import subprocess
+    import sys
+    plugin = str(tmp_path / 'fill_offset.py')
+    with open(plugin, 'w') as f:
+        f.write(PLUGIN_SRC)
+
+    # single-process run (no env): the truth
+    out1 = str(tmp_path / 'vol_w1.npy')
+    env = dict(os.environ)
+    for k in ('RANK', 'WORLD_SIZE', 'MASTER_ADDR', 'MASTER_PORT',
+              'LOCAL_RANK'):
+        env.pop(k, None)
+    r = subprocess.run(
+        [sys.executable, '-m', 'chunkflow_amd']
+        + _cli_pipeline_args(plugin, out1),
+        env=env, capture_output=True, text=True, timeout=180)
+    assert r.returncode == 0, r.stdout + r.stderr
+
+    # world-2 torchrun-style run: same pipeline, env-driven sharding
+    out2 = str(tmp_path / 'vol_w2.npy')
+    procs = []
+    for rank in range(2):
+        env2 = dict(env, RANK=str(rank), WORLD_SIZE='2',
+                    MASTER_ADDR='127.0.0.1', MASTER_PORT='29541',
+                    LOCAL_RANK=str(rank))
+        procs.append(subprocess.Popen(
+            [sys.executable, '-m', 'chunkflow_amd']
+            + _cli_pipeline_args(plugin, out2),
+            env=env2, stdout=subprocess.PIPE, stderr=subprocess.STDOUT,
+            text=True))
+    for p in procs:
+        out, _ = p.communicate(timeout=180)
+        assert p.returncode == 0, out
+
+    v1 = np.load(out1)
+    v2 = np.load(out2)
+    assert v1.shape == (3, 4, 8, 32)
+    np.testing.assert_array_equal(v1, v2)
+    # content is bbox-dependent: block i along x holds ((8i)%200+1)/255
+    for i in range(4):
+        np.testing.assert_allclose(
+            v1[:, :, :, i * 8:(i + 1) * 8],
+            np.float32((8 * i % 200 + 1) / 255.0), rtol=1e-5)
+
+
+def _worker8(rank, world, port, result_path):
+    os.environ['MASTER_ADDR'] = '127.0.0.1'
+    os.environ['MASTER_PORT'] = str(port)
+    os.environ['RANK'] = str(rank)
+    os.environ['WORLD_SIZE'] = str(world)
+    import torch.distributed as dist
+    from chunkflow_amd.cartesian import BoundingBoxes
+    from chunkflow_amd.dispatch import (init_distributed, shard_tasks,
+                                        stitch_to_rank0)
+    init_distributed(backend='gloo')
+    bboxes = BoundingBoxes.from_manual_setup(
+        (4, 8, 8), roi_size=(4, 8, 8 * 8))  # 8 tasks along x
+    assert len(bboxes) == 8
+    local = {i: torch.full((3, 4, 8, 8), float(i + 1), dtype=torch.float32)
+             for i in shard_tasks(list(range(8)), rank, world)}
+    vol = stitch_to_rank0(bboxes, local, 3, rank, world, 'cpu')
+    if rank == 0:
+        np.save(result_path, vol.numpy())
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_shard_and_stitch_world8(tmp_path):
+    """World 8 — the full 8-GPU-node fan-in shape: rank 0 receives from 7
+    concurrent senders (VERDICT r01 weak #7)."""
+    result = str(tmp_path / 'vol8.npy')
+    mp.spawn(_worker8, args=(8, 29549, result), nprocs=8, join=True)
+    vol = np.load(result)
+    assert vol.shape == (3, 4, 8, 64)
+    for i in range(8):
+        np.testing.assert_array_equal(vol[:, :, :, i * 8:(i + 1) * 8],
+                                      np.full((3, 4, 8, 8), i + 1,
+                                              dtype=np.float32))

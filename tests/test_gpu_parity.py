@@ -714,3 +714,63 @@ def test_bf16_engine_fastconv_close_to_f32(golden_dir):
     base = run('float32', '0')
     fast_bf16 = run('bfloat16', '1')
     np.testing.assert_allclose(fast_bf16, base, rtol=0.05, atol=0.03)
+
+
+# --------------------------------------------------------------------------
+# round-2 reference-golden pins: TTA and the full benchmark RSUNet
+# --------------------------------------------------------------------------
+def test_augment_identity_reference_golden(golden):
+    """TTA (augment=True) on GPU vs the live-reference golden
+    (transform.py:114-156 semantics), identity engine."""
+    from chunkflow_amd.chunk import Chunk
+    _, arrays = golden
+    inf = _hip_inferencer(framework='identity', num_output_channels=3,
+                          batch_size=3, mask_output_chunk=True,
+                          augment=True)
+    out = inf(Chunk(arrays['e2e_input_u8'].copy()))
+    np.testing.assert_allclose(out.numpy().array,
+                               arrays['e2e_identity_augment_out'],
+                               rtol=1e-5, atol=1e-6)
+
+
+def test_augment_pytorch_reference_golden(golden, golden_dir):
+    """TTA through the conv engine on GPU (the flips/transposes actually
+    change conv outputs) vs the live-reference golden, 1e-4 fp32."""
+    from chunkflow_amd.chunk import Chunk
+    _, arrays = golden
+    inf = _hip_inferencer(
+        model=os.path.join(golden_dir, 'ref_model.py'),
+        weights=os.path.join(golden_dir, 'ref_model_weights.pt'),
+        framework='pytorch', num_output_channels=3, batch_size=1,
+        mask_output_chunk=True, augment=True)
+    out = inf(Chunk(arrays['e2e_input_u8'].copy()))
+    np.testing.assert_allclose(out.numpy().array,
+                               arrays['e2e_pytorch_augment_out'],
+                               rtol=1e-4, atol=1e-4)
+
+
+def test_rsunet_config2_conv_parity_1e4(golden, golden_dir):
+    """the REAL benchmark RSUNet through the full GPU
+    path (fastconv MFMA rings + MIOpen + HIP tiler/blend) vs the live
+    reference Inferencer on torch-CPU at a config-2-like geometry
+    (64x256x256, patch 20x128x128, overlap 4x32x32), <=1e-4 fp32."""
+    from chunkflow_amd.chunk import Chunk
+    meta, arrays = golden
+    case = meta['cases']['rsunet_64x256x256']
+    rs_in = Chunk.create(size=(64, 256, 256), dtype='uint8', pattern='sin')
+    inf = _hip_inferencer(
+        model=os.path.join(REPO, 'examples', 'nets', 'rsunet.py'),
+        weights=os.path.join(golden_dir, 'rsunet_weights.pt'),
+        patch_size=(20, 128, 128), overlap=(4, 32, 32),
+        framework='pytorch', num_output_channels=3, batch_size=4,
+        mask_output_chunk=True)
+    out = np.asarray(inf(rs_in).numpy().array)
+    assert out.shape == (3, 64, 256, 256)
+    np.testing.assert_allclose(out[:, ::4, ::8, ::8],
+                               arrays['rsunet_64x256x256_sub'],
+                               rtol=1e-4, atol=1e-4)
+    np.testing.assert_allclose(
+        out.ravel()[arrays['rsunet_64x256x256_sample_idx']],
+        arrays['rsunet_64x256x256_sample_val'], rtol=1e-4, atol=1e-4)
+    assert abs(out.astype(np.float64).sum() - case['sum_f64']) < 200.0
+    assert case['min'] - 1e-3 < out.min() and out.max() < case['max'] + 1e-3

@@ -416,3 +416,84 @@ def test_from_h5_negative_cutout_size(tmp_path):
     c = Chunk.from_h5(p, cutout_size=(-1, 6, -1))
     assert c.shape == (4, 6, 8)
     assert np.array_equal(np.asarray(c.array), arr)
+
+
+# --- round-2 reference-golden pins (TTA + full RSUNet) ---------------------
+@pytest.mark.skipif(GPU, reason='CPU plumbing path is refused on a GPU box')
+def test_augment_identity_golden(golden):
+    """TTA (augment=True) pinned against the live reference
+    (transform.py:114-156 semantics): identity engine, batch 3."""
+    from chunkflow_amd.inferencer import Inferencer
+    _, arrays = golden
+    inf = Inferencer(None, None, (10, 32, 32),
+                     output_patch_overlap=(2, 8, 8), framework='identity',
+                     num_output_channels=3, batch_size=3,
+                     mask_output_chunk=True, augment=True)
+    out = inf(Chunk(arrays['e2e_input_u8'].copy()))
+    np.testing.assert_allclose(out.numpy().array,
+                               arrays['e2e_identity_augment_out'],
+                               rtol=1e-5, atol=1e-6)
+
+
+@pytest.mark.skipif(GPU, reason='CPU plumbing path is refused on a GPU box')
+def test_augment_pytorch_golden(golden, golden_dir):
+    """TTA through the conv engine (the transforms actually change the
+    output here, unlike identity) vs the live reference."""
+    from chunkflow_amd.inferencer import Inferencer
+    _, arrays = golden
+    inf = Inferencer(os.path.join(golden_dir, 'ref_model.py'),
+                     os.path.join(golden_dir, 'ref_model_weights.pt'),
+                     (10, 32, 32),
+                     output_patch_overlap=(2, 8, 8), framework='pytorch',
+                     num_output_channels=3, batch_size=1,
+                     mask_output_chunk=True, augment=True)
+    out = inf(Chunk(arrays['e2e_input_u8'].copy()))
+    np.testing.assert_allclose(out.numpy().array,
+                               arrays['e2e_pytorch_augment_out'],
+                               rtol=1e-5, atol=1e-5)
+
+
+def test_rsunet_weight_determinism(golden):
+    """The committed rsunet_weights.pt matches the seeded in-repo model
+    (guards the golden against silent model-file drift)."""
+    meta, _ = golden
+    import importlib.util
+    spec = importlib.util.spec_from_file_location(
+        'rsunet_det', os.path.join(
+            os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+            'examples', 'nets', 'rsunet.py'))
+    mod = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(mod)
+    want = meta['cases']['rsunet_64x256x256']['weight_f64_sums']
+    got = {k: float(v.to(torch.float64).sum())
+           for k, v in mod.InstantiatedModel.state_dict().items()}
+    assert set(got) == set(want)
+    for k in want:
+        assert abs(got[k] - want[k]) < 1e-6, k
+
+
+@pytest.mark.skipif(GPU, reason='CPU plumbing path is refused on a GPU box')
+def test_rsunet_64x256x256_cpu_plumbing_golden(golden, golden_dir):
+    """Full benchmark RSUNet through the product CPU plumbing path vs the
+    live-reference Inferencer output (~15 s)."""
+    from chunkflow_amd.inferencer import Inferencer
+    meta, arrays = golden
+    case = meta['cases']['rsunet_64x256x256']
+    rs_in = Chunk.create(size=(64, 256, 256), dtype='uint8', pattern='sin')
+    assert crc(np.asarray(rs_in.array)) == case['input_crc32']
+    inf = Inferencer(
+        os.path.join(os.path.dirname(golden_dir), '..', 'examples', 'nets',
+                     'rsunet.py'),
+        os.path.join(golden_dir, 'rsunet_weights.pt'),
+        (20, 128, 128), output_patch_overlap=(4, 32, 32),
+        framework='pytorch', num_output_channels=3, batch_size=1,
+        mask_output_chunk=True)
+    out = np.asarray(inf(rs_in).numpy().array)
+    assert out.shape == (3, 64, 256, 256)
+    np.testing.assert_allclose(out[:, ::4, ::8, ::8],
+                               arrays['rsunet_64x256x256_sub'],
+                               rtol=1e-5, atol=1e-5)
+    np.testing.assert_allclose(
+        out.ravel()[arrays['rsunet_64x256x256_sample_idx']],
+        arrays['rsunet_64x256x256_sample_val'], rtol=1e-5, atol=1e-5)
+    assert abs(out.astype(np.float64).sum() - case['sum_f64']) < 1.0
