@@ -8,6 +8,7 @@ SRC = os.path.join(PKG_DIR, 'csrc', 'cfx.hip')
 SRC_CC = os.path.join(PKG_DIR, 'csrc', 'cc.hip')
 SRC_IMG = os.path.join(PKG_DIR, 'csrc', 'image.hip')
 SRC_CONV = os.path.join(PKG_DIR, 'csrc', 'conv.hip')
+SRC_UPDOWN = os.path.join(PKG_DIR, 'csrc', 'updown.hip')
 SRC_INT = os.path.join(PKG_DIR, 'csrc', 'cfx_internal.h')
 HEADER = os.path.join(PKG_DIR, '..', 'include', 'chunkflow_amd.h')
 
@@ -17,7 +18,8 @@ def so_is_fresh() -> bool:
         return False
     so_mtime = os.path.getmtime(SO_PATH)
     return all(os.path.getmtime(p) <= so_mtime
-               for p in (SRC, SRC_CC, SRC_IMG, SRC_CONV, SRC_INT, HEADER))
+               for p in (SRC, SRC_CC, SRC_IMG, SRC_CONV, SRC_UPDOWN, SRC_INT,
+                         HEADER))
 
 
 def build(force: bool = False) -> str:
@@ -31,7 +33,7 @@ def build(force: bool = False) -> str:
     cmd = [
         'hipcc', '--offload-arch=gfx950', '-O3', '-std=c++17',
         '-ffp-contract=off', '-fPIC', '-shared', SRC, SRC_CC, SRC_IMG,
-        SRC_CONV, '-o', SO_PATH,
+        SRC_CONV, SRC_UPDOWN, '-o', SO_PATH,
     ]
     subprocess.run(cmd, check=True)
     return SO_PATH

@@ -1545,6 +1545,202 @@ __global__ __launch_bounds__(512, 1) void k_conv3_zring_bf16_pl(
 }
 
 
+// Sliced bf16 ring: the _pl kernel generalized to a c-slice [c0, c0+CL)
+// of a wider channel dimension (runtime stride CS) and a j-tile
+// [j0, j0+32) of a wider K (runtime stride KS). Widths 36 and 48 run as
+// four launches — (c 0..31, CPV=32) + (c 32.., CPV=16), each for j-tiles
+// j0=0 and j0=32 — reusing the PCB=CP+8 LDS layout whose strides (40/24
+// elements) are the measured conflict-free ones. The second c-half launch
+// chains through `res` = its own output (partial sums round through bf16
+// between halves; bias/ELU apply only on the final half). Weight pack:
+// [27][64][48] zero-padded, [tap][j][c].
+template <int CPV, int TY, int TX>
+__global__ __launch_bounds__(512, 1) void k_conv3_zring_bf16_s(
+    const cfx_bf16* __restrict__ in, const cfx_bf16* __restrict__ wgt,
+    const float* __restrict__ bias, const cfx_bf16* __restrict__ res,
+    cfx_bf16* __restrict__ out, int N, int D, int H, int W, int CS,
+    int CL, int c0, int KS, int Ktot, int j0, int do_elu) {
+    constexpr int CP = CPV;
+    constexpr int PCB = CP + 8;
+    constexpr int SX = TX + 2;
+    constexpr int SY = TY + 2;
+    constexpr int KK = CP / 16;
+    constexpr int C4 = CP / 4;
+    constexpr int LV = (SY * SX * C4 + 511) / 512;
+    typedef cfx_bf16 bf16x4 __attribute__((ext_vector_type(4)));
+    constexpr int WPK = 64;  // weight pack j rows
+    constexpr int WPC = 48;  // weight pack c width
+
+    __shared__ cfx_bf16 ring[3 * SY * SX * PCB];
+    __shared__ cfx_bf16 wall[27 * 32 * PCB];
+
+    const int n = blockIdx.z;
+    const int y0 = blockIdx.y * TY;
+    const int x0 = blockIdx.x * TX;
+    const int tid = threadIdx.x;
+    const int wave = tid >> 6;
+    const int lane = tid & 63;
+
+    for (int idx = tid; idx < 27 * 32 * CP; idx += 512) {
+        const int c = idx % CP;
+        const int j = (idx / CP) % 32;
+        const int tap = idx / (CP * 32);
+        wall[(tap * 32 + j) * PCB + c] =
+            wgt[(tap * WPK + j0 + j) * WPC + c0 + c];
+    }
+
+    const bool xy_interior = y0 >= 1 && y0 + TY + 1 <= H && x0 >= 1 &&
+                             x0 + TX + 1 <= W;
+
+    auto plane_load = [&](int P, bf16x4 (&vals)[LV], bool (&keep)[LV]) {
+        const bool zin = P >= 0 && P < D;
+        const bool interior = zin && xy_interior;
+#pragma unroll
+        for (int li = 0; li < LV; ++li) {
+            const int idx = min(tid + li * 512, SY * SX * C4 - 1);
+            const int c4 = idx % C4;
+            const int v = idx / C4;
+            const int gy = y0 + v / SX - 1;
+            const int gx = x0 + v % SX - 1;
+            const bool cok = c4 * 4 < CL;
+            const bool ok = cok && zin &&
+                            (interior || (gy >= 0 && gy < H && gx >= 0 &&
+                                          gx < W));
+            keep[li] = ok;
+            vals[li] = *reinterpret_cast<const bf16x4*>(
+                in + (((long long)n * D + (zin ? P : 0)) * H +
+                      (ok ? gy : 0)) * (long long)W * CS +
+                (ok ? gx : 0) * (long long)CS + c0 + (cok ? c4 * 4 : 0));
+        }
+    };
+    auto plane_store = [&](int P, const bf16x4 (&vals)[LV],
+                           const bool (&keep)[LV]) {
+        const int slot = ((P + 1) % 3 + 3) % 3;
+#pragma unroll
+        for (int li = 0; li < LV; ++li) {
+            const int idx = tid + li * 512;
+            if (idx >= SY * SX * C4) break;
+            const int c4 = idx % C4;
+            const int v = idx / C4;
+            *reinterpret_cast<bf16x4*>(
+                &ring[((slot * SY + v / SX) * SX + v % SX) * PCB +
+                      c4 * 4]) = keep[li] ? vals[li] : bf16x4{};
+        }
+    };
+
+    {
+        bf16x4 v0[LV], v1[LV];
+        bool k0[LV], k1[LV];
+        plane_load(-1, v0, k0);
+        plane_load(0, v1, k1);
+        plane_store(-1, v0, k0);
+        plane_store(0, v1, k1);
+    }
+    __syncthreads();
+
+    const int ax = lane & 31;
+    const int khalf = (lane >> 5) * 8;
+
+    for (int z = 0; z < D; ++z) {
+        f32x16 acc = {};
+        const cfx_bf16* planes[3];
+#pragma unroll
+        for (int dzi = 0; dzi < 3; ++dzi)
+            planes[dzi] = &ring[(((z + dzi) % 3 + 3) % 3) * SY * SX * PCB];
+        auto addrA = [&](int p) {
+            const int tap = p / KK, kk = p % KK;
+            const int dzi = tap / 9, tl = tap % 9;
+            const int dy = tl / 3 - 1, dx = tl % 3 - 1;
+            return reinterpret_cast<const bf16x8*>(
+                &planes[dzi][((1 + wave + dy) * SX + (1 + dx) + ax) * PCB +
+                             khalf + kk * 16]);
+        };
+        auto addrB = [&](int p) {
+            const int tap = p / KK, kk = p % KK;
+            return reinterpret_cast<const bf16x8*>(
+                &wall[(tap * 32 + ax) * PCB + khalf + kk * 16]);
+        };
+        constexpr int PD = 4;
+        constexpr int NP1 = 18 * KK;  // dzi 0,1: planes z-1, z
+        constexpr int NP2 = 27 * KK;
+
+        bf16x4 vals[LV];
+        bool keep[LV];
+        plane_load(z + 1, vals, keep);
+        {
+            bf16x8 abuf[PD], bbuf[PD];
+#pragma unroll
+            for (int p = 0; p < PD; ++p) {
+                abuf[p] = *addrA(p);
+                bbuf[p] = *addrB(p);
+            }
+#pragma unroll
+            for (int p = 0; p < NP1; ++p) {
+                const int si = p % PD;
+                acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                    abuf[si], bbuf[si], acc, 0, 0, 0);
+                __builtin_amdgcn_sched_barrier(0);
+                if (p + PD < NP1) {
+                    abuf[si] = *addrA(p + PD);
+                    bbuf[si] = *addrB(p + PD);
+                }
+                __builtin_amdgcn_sched_barrier(0);
+            }
+            plane_store(z + 1, vals, keep);
+            __syncthreads();
+#pragma unroll
+            for (int p = NP1; p < NP1 + PD; ++p) {
+                abuf[p % PD] = *addrA(p);
+                bbuf[p % PD] = *addrB(p);
+            }
+#pragma unroll
+            for (int p = NP1; p < NP2; ++p) {
+                const int si = p % PD;
+                acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                    abuf[si], bbuf[si], acc, 0, 0, 0);
+                __builtin_amdgcn_sched_barrier(0);
+                if (p + PD < NP2) {
+                    abuf[si] = *addrA(p + PD);
+                    bbuf[si] = *addrB(p + PD);
+                }
+                __builtin_amdgcn_sched_barrier(0);
+            }
+        }
+
+        const int gy = y0 + wave;
+        const int j = lane & 31;
+        if (gy < H && j0 + j < Ktot) {
+            const float bj = bias ? bias[j0 + j] : 0.f;
+            cfx_bf16 rv[16];
+            if (res) {
+#pragma unroll
+                for (int r = 0; r < 16; ++r) {
+                    const int row =
+                        (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
+                    const int gx = min(x0 + row, W - 1);
+                    rv[r] = res[
+                        (((long long)n * D + z) * H + gy) * (long long)W *
+                            KS + (long long)gx * KS + j0 + j];
+                }
+            }
+#pragma unroll
+            for (int r = 0; r < 16; ++r) {
+                const int row = (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
+                const int gx = x0 + row;
+                if (gx >= W) continue;
+                long long o = (((long long)n * D + z) * H + gy) *
+                                  (long long)W * KS +
+                              (long long)gx * KS + j0 + j;
+                float v = acc[r] + bj;
+                if (res) v += (float)rv[r];
+                if (do_elu) v = v > 0.f ? v : expm1f(v);
+                out[o] = (cfx_bf16)v;
+            }
+        }
+    }
+}
+
+
 // 4-slot ring ("quad"): planes z-1, z, z+1 all RESIDENT when iteration z
 // starts, plane z+2 loads during the full 54-MFMA mainloop and lands in
 // the slot plane z-2 vacated — ONE barrier per z and a whole iteration
@@ -1728,8 +1924,34 @@ extern "C" int cfx_conv3_ndhwc_bf16(cfx_ctx* ctx, const void* in,
                                     const void* residual, void* out, int N,
                                     int D, int H, int W, int C, int K,
                                     int do_elu) {
+    if ((C == 36 && K == 36) || (C == 48 && K == 48)) {
+        // sliced schedule: (c-half, j-tile) x 4 launches; wgt pack is
+        // [27][64][48] zero-padded. Bias/ELU only on the final c-half,
+        // which chains through res = its own output.
+        dim3 grid((W + 31) / 32, (H + 7) / 8, (unsigned)N);
+        hipEvent_t e0;
+        if (prof_begin(ctx, &e0)) return -1;
+        const int CL2 = C - 32;
+        for (int j0 = 0; j0 < K; j0 += 32) {
+            hipLaunchKernelGGL((k_conv3_zring_bf16_s<32, 8, 32>), grid,
+                               dim3(512), 0, ctx->stream,
+                               (const cfx_bf16*)in, (const cfx_bf16*)wgt,
+                               nullptr, (const cfx_bf16*)residual,
+                               (cfx_bf16*)out, N, D, H, W, C, 32, 0, K, K,
+                               j0, 0);
+            hipLaunchKernelGGL((k_conv3_zring_bf16_s<16, 8, 32>), grid,
+                               dim3(512), 0, ctx->stream,
+                               (const cfx_bf16*)in, (const cfx_bf16*)wgt,
+                               bias, (const cfx_bf16*)out, (cfx_bf16*)out,
+                               N, D, H, W, C, CL2, 32, K, K, j0, do_elu);
+        }
+        CFX_CHECK(hipGetLastError());
+        double flops = 2.0 * 27.0 * C * K * (double)N * D * H * W;
+        if (prof_end(ctx, e0, CFX_K_CONV, flops)) return -1;
+        return 0;
+    }
     if (C != 28 || K != 28) {
-        g_err = "cfx_conv3_ndhwc_bf16: only C == K == 28 instantiated";
+        g_err = "cfx_conv3_ndhwc_bf16: only C == K == 28/36/48 instantiated";
         return -1;
     }
     dim3 grid((W + 31) / 32, (H + 7) / 8, (unsigned)N);

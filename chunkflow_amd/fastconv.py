@@ -139,7 +139,7 @@ def maybe_accelerate(model: nn.Module, device_index: int = 0) -> int:
     return count
 
 
-BF16_WIDTHS = (28,)
+BF16_WIDTHS = (28, 36, 48)
 
 
 class CfxConv3dBF16(nn.Module):
@@ -152,7 +152,12 @@ class CfxConv3dBF16(nn.Module):
         self.K = conv.out_channels
         self.device_index = device_index
         w = conv.weight.detach().float()  # (K, C, 3, 3, 3)
-        pack = torch.zeros(27, 32, 32)
+        # C <= 32 uses the (27, 32, 32) pack of the single-tile ring;
+        # 36/48 use the (27, 64, 48) pack of the sliced 4-launch schedule
+        if self.C <= 32:
+            pack = torch.zeros(27, 32, 32)
+        else:
+            pack = torch.zeros(27, 64, 48)
         pack[:, :self.K, :self.C] = w.permute(2, 3, 4, 0, 1) \
             .reshape(27, self.K, self.C)
         self.register_buffer('wpack', pack.to(torch.bfloat16).contiguous())
@@ -237,5 +242,124 @@ def maybe_accelerate_bf16(model: nn.Module, device_index: int = 0) -> int:
             elif elig(child):
                 setattr(parent, name,
                         CfxConv3dBF16(child, device_index).to(dev))
+                count += 1
+    return count
+
+
+# --------------------------------------------------------------------------
+# up/down-sampling convs: the RSUNet (1,2,2)-kernel, (1,2,2)-stride shapes
+# are HBM-streaming work; csrc/updown.hip replaces MIOpen's implicit GEMM
+# (bf16 bwd_data runs ~20x below the stream roofline there).
+# --------------------------------------------------------------------------
+class CfxUpConv3d(nn.Module):
+    """Drop-in for nn.ConvTranspose3d(kernel=(1,2,2), stride=(1,2,2))."""
+
+    def __init__(self, conv: nn.ConvTranspose3d, device_index: int = 0,
+                 bf16: bool = False):
+        super().__init__()
+        self.C = conv.in_channels
+        self.K = conv.out_channels
+        self.device_index = device_index
+        self.bf16 = bf16
+        w = conv.weight.detach().float()      # (C, K, 1, 2, 2)
+        pack = w[:, :, 0].permute(2, 3, 0, 1).reshape(4, self.C, self.K)
+        dt = torch.bfloat16 if bf16 else torch.float32
+        self.register_buffer('wpack', pack.to(dt).contiguous())
+        if conv.bias is not None:
+            self.register_buffer('bias', conv.bias.detach().float())
+        else:
+            self.bias = None
+
+    def forward(self, x):
+        x = x.contiguous(memory_format=torch.channels_last_3d)
+        n, c, d, h, w = x.shape
+        dt = torch.bfloat16 if self.bf16 else torch.float32
+        assert x.dtype == dt
+        out = torch.empty((n, self.K, d, 2 * h, 2 * w), dtype=dt,
+                          device=x.device,
+                          memory_format=torch.channels_last_3d)
+        get_cfx(self.device_index).upconv_2x2(
+            x.data_ptr(), self.wpack.data_ptr(),
+            self.bias.data_ptr() if self.bias is not None else None,
+            out.data_ptr(), n, d, h, w, self.C, self.K, bf16=self.bf16)
+        return out
+
+
+class CfxDownConv3d(nn.Module):
+    """Drop-in for nn.Conv3d(kernel=(1,2,2), stride=(1,2,2))."""
+
+    def __init__(self, conv: nn.Conv3d, device_index: int = 0,
+                 bf16: bool = False):
+        super().__init__()
+        self.C = conv.in_channels
+        self.K = conv.out_channels
+        self.device_index = device_index
+        self.bf16 = bf16
+        w = conv.weight.detach().float()      # (K, C, 1, 2, 2)
+        pack = w[:, :, 0].permute(2, 3, 1, 0).reshape(4, self.C, self.K)
+        dt = torch.bfloat16 if bf16 else torch.float32
+        self.register_buffer('wpack', pack.to(dt).contiguous())
+        if conv.bias is not None:
+            self.register_buffer('bias', conv.bias.detach().float())
+        else:
+            self.bias = None
+
+    def forward(self, x):
+        x = x.contiguous(memory_format=torch.channels_last_3d)
+        n, c, d, h, w = x.shape
+        dt = torch.bfloat16 if self.bf16 else torch.float32
+        assert x.dtype == dt
+        out = torch.empty((n, self.K, d, h // 2, w // 2), dtype=dt,
+                          device=x.device,
+                          memory_format=torch.channels_last_3d)
+        get_cfx(self.device_index).downconv_2x2(
+            x.data_ptr(), self.wpack.data_ptr(),
+            self.bias.data_ptr() if self.bias is not None else None,
+            out.data_ptr(), n, d, h, w, self.C, self.K, bf16=self.bf16)
+        return out
+
+
+def _up_eligible(m) -> bool:
+    return (isinstance(m, nn.ConvTranspose3d)
+            and m.kernel_size == (1, 2, 2) and m.stride == (1, 2, 2)
+            and m.padding == (0, 0, 0) and m.output_padding == (0, 0, 0)
+            and m.dilation == (1, 1, 1) and m.groups == 1
+            and m.in_channels <= 64 and m.out_channels <= 64)
+
+
+def _down_eligible(m) -> bool:
+    return (isinstance(m, nn.Conv3d)
+            and m.kernel_size == (1, 2, 2) and m.stride == (1, 2, 2)
+            and m.padding == (0, 0, 0) and m.dilation == (1, 1, 1)
+            and m.groups == 1
+            and m.in_channels <= 64 and m.out_channels <= 64)
+
+
+@torch.no_grad()
+def _updown_matches(orig, repl, dev, bf16):
+    x = torch.randn(1, repl.C, 3, 12, 16, device=dev)
+    if bf16:
+        x = x.to(torch.bfloat16)
+    x = x.contiguous(memory_format=torch.channels_last_3d)
+    want = orig.to(dev)(x).float()
+    got = repl(x).float()
+    tol = 0.05 if bf16 else 1e-4
+    return bool(torch.allclose(got, want, rtol=tol, atol=tol))
+
+
+def accelerate_updown(model: nn.Module, device_index: int = 0,
+                      bf16: bool = False) -> int:
+    """Swap eligible up/down-sampling convs in-place; returns count."""
+    count = 0
+    dev = f'cuda:{device_index}'
+    for parent in list(model.modules()):
+        for name, child in list(parent.named_children()):
+            repl = None
+            if _up_eligible(child):
+                repl = CfxUpConv3d(child, device_index, bf16=bf16).to(dev)
+            elif _down_eligible(child):
+                repl = CfxDownConv3d(child, device_index, bf16=bf16).to(dev)
+            if repl is not None and _updown_matches(child, repl, dev, bf16):
+                setattr(parent, name, repl)
                 count += 1
     return count

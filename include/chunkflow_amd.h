@@ -143,13 +143,28 @@ int cfx_conv3_ndhwc_zring(cfx_ctx* ctx, const float* in, const float* wgt,
                           const float* bias, const float* residual,
                           float* out, int N, int D, int H, int W, int C,
                           int K, int do_elu);
-/* bf16 persistent-z ring (v_mfma_f32_32x32x16_bf16; C == K == 28):
- * in/out/residual are bf16 NDHWC, wgt is a bf16 (27, 32, 32) [tap][j][c]
- * zero-padded pack, bias stays f32; epilogue math in f32 */
+/* bf16 persistent-z ring (v_mfma_f32_32x32x16_bf16; C == K in
+ * {28, 36, 48}): in/out/residual are bf16 NDHWC, wgt is a bf16 zero-padded
+ * [tap][j][c] pack — (27, 32, 32) for C == 28, (27, 64, 48) for 36/48
+ * (those run as four c-half x j-tile launches; partial sums round through
+ * bf16 between the halves) — bias stays f32; epilogue math in f32 */
 int cfx_conv3_ndhwc_bf16(cfx_ctx* ctx, const void* in, const void* wgt,
                          const float* bias, const void* residual,
                          void* out, int N, int D, int H, int W, int C,
                          int K, int do_elu);
+
+/* ---- up/down-sampling convs (RSUNet (1,2,2)-kernel, (1,2,2)-stride) ----- */
+/* ConvTranspose3d: out (N,D,2H,2W,K) from in (N,D,H,W,C), NDHWC; wgt
+ * packed [parity q=(py<<1)|px][C][K] in the compute dtype; bias f32 or
+ * NULL; is_bf16 selects bf16 vs f32 tensors. HBM-streaming VALU kernel. */
+int cfx_upconv_2x2(cfx_ctx* ctx, const void* in, const void* wgt,
+                   const float* bias, void* out, int N, int D, int H,
+                   int W, int C, int K, int is_bf16);
+/* strided Conv3d (downsample): out (N,D,H/2,W/2,K) from in (N,D,H,W,C);
+ * same wgt pack/convention as cfx_upconv_2x2 */
+int cfx_downconv_2x2(cfx_ctx* ctx, const void* in, const void* wgt,
+                     const float* bias, void* out, int N, int D, int H,
+                     int W, int C, int K, int is_bf16);
 /* the 32x32x2-MFMA variant (C == K == 28 instantiated) */
 int cfx_conv3_ndhwc_w32(cfx_ctx* ctx, const float* in, const float* wgt,
                         const float* bias, const float* residual,

@@ -774,3 +774,130 @@ def test_rsunet_config2_conv_parity_1e4(golden, golden_dir):
         arrays['rsunet_64x256x256_sample_val'], rtol=1e-4, atol=1e-4)
     assert abs(out.astype(np.float64).sum() - case['sum_f64']) < 200.0
     assert case['min'] - 1e-3 < out.min() and out.max() < case['max'] + 1e-3
+
+
+# --------------------------------------------------------------------------
+# round-2 kernels: up/down-sampling convs + sliced bf16 ring (C=36/48)
+# --------------------------------------------------------------------------
+@pytest.mark.parametrize('C,K,bias', [(36, 28, True), (48, 36, True),
+                                      (64, 48, True), (28, 28, False)])
+def test_upconv_2x2_f32_vs_torch(C, K, bias):
+    """csrc/updown.hip transposed conv vs torch ConvTranspose3d, f32."""
+    from chunkflow_amd.fastconv import CfxUpConv3d
+    torch.manual_seed(3)
+    conv = torch.nn.ConvTranspose3d(C, K, (1, 2, 2), stride=(1, 2, 2),
+                                    bias=bias).cuda()
+    x = torch.randn(2, C, 3, 13, 21, device='cuda') \
+        .contiguous(memory_format=torch.channels_last_3d)
+    want = conv(x)
+    got = CfxUpConv3d(conv, 0, bf16=False).cuda()(x)
+    assert got.shape == want.shape
+    torch.testing.assert_close(got, want, rtol=1e-4, atol=1e-4)
+
+
+@pytest.mark.parametrize('C,K', [(36, 28), (64, 48)])
+def test_upconv_2x2_bf16_vs_torch(C, K):
+    from chunkflow_amd.fastconv import CfxUpConv3d
+    torch.manual_seed(4)
+    conv = torch.nn.ConvTranspose3d(C, K, (1, 2, 2),
+                                    stride=(1, 2, 2)).cuda()
+    x = torch.randn(2, C, 3, 9, 33, device='cuda')
+    want = conv.to(torch.bfloat16)(
+        x.to(torch.bfloat16)
+        .contiguous(memory_format=torch.channels_last_3d)).float()
+    got = CfxUpConv3d(conv, 0, bf16=True).cuda()(
+        x.to(torch.bfloat16)
+        .contiguous(memory_format=torch.channels_last_3d)).float()
+    torch.testing.assert_close(got, want, rtol=0.05, atol=0.05)
+
+
+@pytest.mark.parametrize('C,K,bias', [(28, 36, True), (36, 48, True),
+                                      (48, 64, False)])
+def test_downconv_2x2_f32_vs_torch(C, K, bias):
+    from chunkflow_amd.fastconv import CfxDownConv3d
+    torch.manual_seed(5)
+    conv = torch.nn.Conv3d(C, K, (1, 2, 2), stride=(1, 2, 2),
+                           bias=bias).cuda()
+    x = torch.randn(2, C, 3, 14, 22, device='cuda') \
+        .contiguous(memory_format=torch.channels_last_3d)
+    want = conv(x)
+    got = CfxDownConv3d(conv, 0, bf16=False).cuda()(x)
+    assert got.shape == want.shape
+    torch.testing.assert_close(got, want, rtol=1e-4, atol=1e-4)
+
+
+def test_downconv_2x2_bf16_vs_torch():
+    from chunkflow_amd.fastconv import CfxDownConv3d
+    torch.manual_seed(6)
+    conv = torch.nn.Conv3d(36, 48, (1, 2, 2), stride=(1, 2, 2)).cuda()
+    x = torch.randn(1, 36, 4, 16, 40, device='cuda').to(torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last_3d)
+    want = conv.to(torch.bfloat16)(x).float()
+    got = CfxDownConv3d(conv, 0, bf16=True).cuda()(x).float()
+    torch.testing.assert_close(got, want, rtol=0.05, atol=0.05)
+
+
+@pytest.mark.parametrize('C', [36, 48])
+def test_bf16_sliced_ring_vs_f32_conv(C):
+    """The 4-launch sliced bf16 ring (c-halves x j-tiles) vs the same conv
+    in f32: the bf16 engine contract (partial sums round through bf16)."""
+    from chunkflow_amd.fastconv import CfxConv3dBF16
+    torch.manual_seed(7)
+    conv = torch.nn.Conv3d(C, C, 3, padding=1).cuda()
+    x = torch.randn(2, C, 5, 37, 41, device='cuda') \
+        .contiguous(memory_format=torch.channels_last_3d)
+    want = conv(x).float()
+    got = CfxConv3dBF16(conv, 0).cuda()._run(
+        x.to(torch.bfloat16)
+        .contiguous(memory_format=torch.channels_last_3d)).float()
+    # bf16 fragments + bf16 inter-half rounding: bf16-level tolerance,
+    # tight in the mean
+    torch.testing.assert_close(got, want, rtol=0.08, atol=0.08)
+    assert float((got - want).abs().mean()) < 0.02
+
+
+@pytest.mark.parametrize('C', [36, 48])
+def test_bf16_sliced_ring_residual_elu(C):
+    """Fused bias+residual+ELU epilogue of the sliced schedule vs torch."""
+    from chunkflow_amd.fastconv import CfxConv3dBF16
+    torch.manual_seed(8)
+    conv = torch.nn.Conv3d(C, C, 3, padding=1).cuda()
+    x = torch.randn(1, C, 4, 18, 35, device='cuda')
+    r = torch.randn(1, C, 4, 18, 35, device='cuda')
+    want = torch.nn.functional.elu(conv(x) + r).float()
+    xb = x.to(torch.bfloat16).cuda() \
+        .contiguous(memory_format=torch.channels_last_3d)
+    rb = r.to(torch.bfloat16).cuda() \
+        .contiguous(memory_format=torch.channels_last_3d)
+    got = CfxConv3dBF16(conv, 0).cuda()._run(xb, residual=rb,
+                                             elu=True).float()
+    torch.testing.assert_close(got, want, rtol=0.08, atol=0.08)
+    assert float((got - want).abs().mean()) < 0.02
+
+
+def test_rsunet_bf16_full_surgery_runs():
+    """Full RSUNet bf16 surgery (rings 28/36/48 + up/down kernels) stays
+    close to the torch-bf16 model output on a real patch shape."""
+    import importlib.util
+    from chunkflow_amd.fastconv import (maybe_accelerate_bf16,
+                                        accelerate_updown)
+    spec = importlib.util.spec_from_file_location(
+        'rsunet_t', os.path.join(REPO, 'examples', 'nets', 'rsunet.py'))
+    mod = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(mod)
+    model = mod.InstantiatedModel.cuda().eval().to(torch.bfloat16) \
+        .to(memory_format=torch.channels_last_3d)
+    x = torch.randn(1, 1, 4, 64, 64, device='cuda').to(torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last_3d)
+    with torch.no_grad():
+        want = model(x).float()
+    import copy
+    m2 = copy.deepcopy(model)
+    n_rb = maybe_accelerate_bf16(m2, 0)
+    n_ud = accelerate_updown(m2, 0, bf16=True)
+    assert n_rb == 6, n_rb   # enc+dec ResBlocks at widths 28/36/48
+    assert n_ud == 6, n_ud   # 3 down + 3 up
+    with torch.no_grad():
+        got = m2(x).float()
+    # sigmoid output in [0,1]: absolute tolerance
+    torch.testing.assert_close(got, want, rtol=0.1, atol=0.03)
