@@ -1,16 +1,20 @@
 // Up/down-sampling convolutions of the RSUNet encoder/decoder on gfx950.
 //
 // The reference runs these through torch -> MIOpen (ConvTranspose3d /
-// strided Conv3d). On MI355X the (1,2,2)-kernel, (1,2,2)-stride shapes are
-// pure HBM-streaming work (one input vector read + K-vector write per
-// spatial position; zero tap overlap), but MIOpen's bf16 bwd_data
-// implicit-GEMM runs them ~20x slower than the roofline. These kernels are
-// plain VALU gather/scatter streams: weights LDS-resident, input rows
-// served from L1 (threads of one workgroup share rows), f32 accumulation.
+// strided Conv3d). The (1,2,2)-kernel, (1,2,2)-stride shapes are pure
+// HBM-streaming work (a C-vector read + K-vector write per spatial
+// position; no tap overlap), and MIOpen's bf16 bwd_data implicit GEMM is
+// far off that roofline on the big up-conv (measured 8.0 ms vs ~0.2 ms
+// algorithmic at 36->28 x 128^2, profiles/updown_probe_r02.json). These
+// kernels are VALU streams: input row and weights staged to LDS in
+// batched coalesced loads (a per-load global read next to its use costs a
+// full vmcnt drain — the round-1 staging lesson), per-thread accumulators
+// over 8 x-positions so each weight vector is read once per c.
 //
-// Layouts: NDHWC (channels_last_3d); transposed-conv weights packed
-// [parity q=(py<<1)|px][C][K]; down-conv weights packed the same.
-// bias f32 (may be NULL). No activation (RSUNet applies none here).
+// Layouts: NDHWC (channels_last_3d); weights packed
+// [parity q=(py<<1)|px][C][K] (stored in LDS as [c][k][q] float4 so one
+// ds_read_b128 serves all four parities). bias f32 (may be NULL).
+// No activation (RSUNet applies none here). f32 accumulation.
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
 
@@ -21,73 +25,106 @@ namespace {
 typedef __hip_bfloat16 cfx_bf16;
 
 template <typename T>
-__device__ __forceinline__ float ld(const T* p);
+__device__ __forceinline__ float ldf(const T* p);
 template <>
-__device__ __forceinline__ float ld<float>(const float* p) { return *p; }
+__device__ __forceinline__ float ldf<float>(const float* p) { return *p; }
 template <>
-__device__ __forceinline__ float ld<cfx_bf16>(const cfx_bf16* p) {
+__device__ __forceinline__ float ldf<cfx_bf16>(const cfx_bf16* p) {
     return __bfloat162float(*p);
 }
 template <typename T>
-__device__ __forceinline__ void st(T* p, float v);
+__device__ __forceinline__ void stf(T* p, float v);
 template <>
-__device__ __forceinline__ void st<float>(float* p, float v) { *p = v; }
+__device__ __forceinline__ void stf<float>(float* p, float v) { *p = v; }
 template <>
-__device__ __forceinline__ void st<cfx_bf16>(cfx_bf16* p, float v) {
+__device__ __forceinline__ void stf<cfx_bf16>(cfx_bf16* p, float v) {
     *p = __float2bfloat16(v);
+}
+
+// cooperative stage of `rows` input rows (each `xn` positions x C
+// channels, row r at in + row_off[r]) into s_in[r * xn * CPAD + x * CPAD
+// + c] as f32, batched so the loads pipeline (no per-load drain).
+template <typename T>
+__device__ __forceinline__ void stage_rows(
+    float* s_in, const T* in, const long long* row_off, int rows, int xn,
+    int xvalid, int C, int CPAD, int tid, int nthreads) {
+    const int total = rows * xn * C;
+    for (int idx = tid; idx < total; idx += nthreads) {
+        const int c = idx % C;
+        const int x = (idx / C) % xn;
+        const int r = idx / (C * xn);
+        const float v = x < xvalid
+                            ? ldf(&in[row_off[r] + (long long)x * C + c])
+                            : 0.f;
+        s_in[(r * xn + x) * CPAD + c] = v;
+    }
 }
 
 // ---- transposed conv (1,2,2), stride (1,2,2) ------------------------------
 // out[n, z, 2y+py, 2x+px, k] = bias[k] + sum_c in[n, z, y, x, c] * w[q][c][k]
 // One workgroup: one (n, z, y) input row, XI input x positions, all K.
-// threads 256 = 8 pos-slots x 32 k-slots; weights staged to LDS as
-// [c][k][q] float4 so one ds_read_b128 serves all four parities.
+// 256 threads = 8 pos-slots x 32 k-slots; each thread accumulates XI/8
+// x-positions x 4 parities so w4 is read once per (c, k).
 template <typename T, int XI>
-__global__ __launch_bounds__(256, 1) void k_upconv2(
+__global__ __launch_bounds__(256, 2) void k_upconv2(
     const T* __restrict__ in, const T* __restrict__ wgt,
     const float* __restrict__ bias, T* __restrict__ out, int N, int D,
     int H, int W, int C, int K) {
-    extern __shared__ float s_w[];  // [C][K][4], K <= 64
+    constexpr int PX = XI / 8;  // x-positions per thread
+    extern __shared__ float smem[];
+    float* s_w = smem;                    // [C][K][4]
+    float* s_in = smem + C * K * 4;       // [XI][CPAD]
+    const int CPAD = C | 1;
     const int tid = threadIdx.x;
     const int kslot = tid & 31;
     const int pslot = tid >> 5;
 
-    // stage weights: wgt is [4][C][K] -> LDS [c][k][q]
     for (int idx = tid; idx < C * K * 4; idx += 256) {
         const int q = idx & 3;
         const int k = (idx >> 2) % K;
         const int c = (idx >> 2) / K;
-        s_w[(c * K + k) * 4 + q] = ld(&wgt[(q * C + c) * K + k]);
+        s_w[(c * K + k) * 4 + q] = ldf(&wgt[(q * C + c) * K + k]);
     }
-    __syncthreads();
 
     const int nz = blockIdx.z;           // n * D + z
     const int y = blockIdx.y;
     const int x0 = blockIdx.x * XI;
-    const long long in_row =
-        (((long long)nz * H) + y) * W * C;
-    const long long out_base = ((long long)nz * 2 * H) * 2 * W * K;
+    const int xvalid = min(XI, W - x0);
+    long long row_off[1] = {((((long long)nz * H) + y) * W + x0) * C};
+    stage_rows(s_in, in, row_off, 1, XI, xvalid, C, CPAD, tid, 256);
+    __syncthreads();
 
-    for (int xi = x0 + pslot; xi < min(x0 + XI, W); xi += 8) {
-        const T* ip = in + in_row + (long long)xi * C;
-        for (int k = kslot; k < K; k += 32) {
-            const float bj = bias ? bias[k] : 0.f;
-            float a0 = bj, a1 = bj, a2 = bj, a3 = bj;
-            for (int c = 0; c < C; ++c) {
-                const float v = ld(&ip[c]);  // same addr across k: broadcast
-                const float4 w4 =
-                    *reinterpret_cast<const float4*>(&s_w[(c * K + k) * 4]);
-                a0 += v * w4.x;
-                a1 += v * w4.y;
-                a2 += v * w4.z;
-                a3 += v * w4.w;
+    const long long out_base =
+        ((((long long)nz * 2 * H) + 2 * y) * 2 * W) * K;
+
+    for (int k = kslot; k < K; k += 32) {
+        const float bj = bias ? bias[k] : 0.f;
+        float acc[PX][4];
+#pragma unroll
+        for (int p = 0; p < PX; ++p)
+#pragma unroll
+            for (int q = 0; q < 4; ++q) acc[p][q] = bj;
+        for (int c = 0; c < C; ++c) {
+            const float4 w4 =
+                *reinterpret_cast<const float4*>(&s_w[(c * K + k) * 4]);
+#pragma unroll
+            for (int p = 0; p < PX; ++p) {
+                const float v = s_in[(pslot + p * 8) * CPAD + c];
+                acc[p][0] += v * w4.x;
+                acc[p][1] += v * w4.y;
+                acc[p][2] += v * w4.z;
+                acc[p][3] += v * w4.w;
             }
-            T* op = out + out_base + ((long long)2 * y * 2 * W +
-                                      2 * xi) * K + k;
-            st(op, a0);                       // (py=0, px=0)
-            st(op + K, a1);                   // (0, 1)
-            st(op + (long long)2 * W * K, a2);      // (1, 0)
-            st(op + (long long)2 * W * K + K, a3);  // (1, 1)
+        }
+#pragma unroll
+        for (int p = 0; p < PX; ++p) {
+            const int xi = x0 + pslot + p * 8;
+            if (xi >= W) continue;
+            T* op = out + out_base + (long long)2 * xi * K + k;
+            stf(op, acc[p][0]);
+            stf(op + K, acc[p][1]);
+            stf(op + (long long)2 * W * K, acc[p][2]);
+            stf(op + (long long)2 * W * K + K, acc[p][3]);
         }
     }
 }
@@ -96,11 +133,15 @@ __global__ __launch_bounds__(256, 1) void k_upconv2(
 // out[n, z, y, x, k] = bias[k] + sum_q sum_c in[n, z, 2y+qy, 2x+qx, c] *
 //                      w[q][c][k];  H, W are INPUT dims (out H/2 x W/2).
 template <typename T, int XO>
-__global__ __launch_bounds__(256, 1) void k_downconv2(
+__global__ __launch_bounds__(256, 2) void k_downconv2(
     const T* __restrict__ in, const T* __restrict__ wgt,
     const float* __restrict__ bias, T* __restrict__ out, int N, int D,
     int H, int W, int C, int K) {
-    extern __shared__ float s_w[];  // [c][k][q]
+    constexpr int PX = XO / 8;
+    extern __shared__ float smem[];
+    float* s_w = smem;                       // [C][K][4]
+    float* s_in = smem + C * K * 4;          // [2 rows][2*XO][CPAD]
+    const int CPAD = C | 1;
     const int tid = threadIdx.x;
     const int kslot = tid & 31;
     const int pslot = tid >> 5;
@@ -110,33 +151,43 @@ __global__ __launch_bounds__(256, 1) void k_downconv2(
         const int q = idx & 3;
         const int k = (idx >> 2) % K;
         const int c = (idx >> 2) / K;
-        s_w[(c * K + k) * 4 + q] = ld(&wgt[(q * C + c) * K + k]);
+        s_w[(c * K + k) * 4 + q] = ldf(&wgt[(q * C + c) * K + k]);
     }
-    __syncthreads();
 
     const int nz = blockIdx.z;
     const int y = blockIdx.y;            // output row
-    const int x0 = blockIdx.x * XO;
-    const long long in_row0 =
-        (((long long)nz * H) + 2 * y) * W * C;
+    const int x0 = blockIdx.x * XO;      // output x base
+    const int xvalid = min(2 * XO, W - 2 * x0);
+    long long row_off[2] = {
+        ((((long long)nz * H) + 2 * y) * W + 2 * x0) * C,
+        ((((long long)nz * H) + 2 * y + 1) * W + 2 * x0) * C};
+    stage_rows(s_in, in, row_off, 2, 2 * XO, xvalid, C, CPAD, tid, 256);
+    __syncthreads();
+
     const long long out_row = (((long long)nz * HO) + y) * WO * K;
 
-    for (int xo = x0 + pslot; xo < min(x0 + XO, WO); xo += 8) {
-        const T* i00 = in + in_row0 + (long long)2 * xo * C;
-        const T* i01 = i00 + C;
-        const T* i10 = i00 + (long long)W * C;
-        const T* i11 = i10 + C;
-        for (int k = kslot; k < K; k += 32) {
-            float acc = bias ? bias[k] : 0.f;
-            for (int c = 0; c < C; ++c) {
-                const float4 w4 =
-                    *reinterpret_cast<const float4*>(&s_w[(c * K + k) * 4]);
-                acc += ld(&i00[c]) * w4.x;
-                acc += ld(&i01[c]) * w4.y;
-                acc += ld(&i10[c]) * w4.z;
-                acc += ld(&i11[c]) * w4.w;
+    for (int k = kslot; k < K; k += 32) {
+        const float bj = bias ? bias[k] : 0.f;
+        float acc[PX];
+#pragma unroll
+        for (int p = 0; p < PX; ++p) acc[p] = bj;
+        for (int c = 0; c < C; ++c) {
+            const float4 w4 =
+                *reinterpret_cast<const float4*>(&s_w[(c * K + k) * 4]);
+#pragma unroll
+            for (int p = 0; p < PX; ++p) {
+                const int xl = 2 * (pslot + p * 8);  // local input x
+                acc[p] += s_in[xl * CPAD + c] * w4.x;
+                acc[p] += s_in[(xl + 1) * CPAD + c] * w4.y;
+                acc[p] += s_in[(2 * XO + xl) * CPAD + c] * w4.z;
+                acc[p] += s_in[(2 * XO + xl + 1) * CPAD + c] * w4.w;
             }
-            st(&out[out_row + (long long)xo * K + k], acc);
+        }
+#pragma unroll
+        for (int p = 0; p < PX; ++p) {
+            const int xo = x0 + pslot + p * 8;
+            if (xo >= WO) continue;
+            stf(&out[out_row + (long long)xo * K + k], acc[p]);
         }
     }
 }
@@ -152,7 +203,8 @@ extern "C" int cfx_upconv_2x2(cfx_ctx* ctx, const void* in, const void* wgt,
     }
     constexpr int XI = 64;
     dim3 grid((W + XI - 1) / XI, H, (unsigned)(N * D));
-    const size_t shmem = (size_t)C * K * 4 * sizeof(float);
+    const size_t shmem =
+        ((size_t)C * K * 4 + (size_t)XI * (C | 1)) * sizeof(float);
     hipEvent_t e0;
     if (prof_begin(ctx, &e0)) return -1;
     if (is_bf16)
@@ -184,7 +236,8 @@ extern "C" int cfx_downconv_2x2(cfx_ctx* ctx, const void* in,
     }
     constexpr int XO = 64;
     dim3 grid((W / 2 + XO - 1) / XO, H / 2, (unsigned)(N * D));
-    const size_t shmem = (size_t)C * K * 4 * sizeof(float);
+    const size_t shmem =
+        ((size_t)C * K * 4 + (size_t)2 * 2 * XO * (C | 1)) * sizeof(float);
     hipEvent_t e0;
     if (prof_begin(ctx, &e0)) return -1;
     if (is_bf16)

@@ -139,7 +139,11 @@ def maybe_accelerate(model: nn.Module, device_index: int = 0) -> int:
     return count
 
 
-BF16_WIDTHS = (28, 36, 48)
+# bf16 ring widths that BEAT MIOpen bf16 in steady state (measured,
+# profiles/updown_probe_r02.json): C=28 534 TF (3.5x), C=36 sliced 205 vs
+# 195 TF. C=48 sliced measured-rejected: 269-295 TF vs MIOpen's 348-355 on
+# the 64^2 extent (kernel kept callable + tested for the record).
+BF16_WIDTHS = (28, 36)
 
 
 class CfxConv3dBF16(nn.Module):

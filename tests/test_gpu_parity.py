@@ -895,7 +895,9 @@ def test_rsunet_bf16_full_surgery_runs():
     m2 = copy.deepcopy(model)
     n_rb = maybe_accelerate_bf16(m2, 0)
     n_ud = accelerate_updown(m2, 0, bf16=True)
-    assert n_rb == 6, n_rb   # enc+dec ResBlocks at widths 28/36/48
+    from chunkflow_amd.fastconv import BF16_WIDTHS
+    want_rb = 2 * len([w for w in (28, 36, 48) if w in BF16_WIDTHS])
+    assert n_rb == want_rb, (n_rb, want_rb)  # enc+dec blocks per width
     assert n_ud == 6, n_ud   # 3 down + 3 up
     with torch.no_grad():
         got = m2(x).float()
