@@ -192,7 +192,98 @@ __global__ __launch_bounds__(256, 2) void k_downconv2(
     }
 }
 
+// ---- input conv (1,5,5), pad (0,2,2), single input channel ---------------
+// RSUNet's conv_in (1 -> 28). MIOpen's implicit GEMM collapses to a
+// degenerate K-dim=25 GEMM here and runs ~39 ms (bf16) / ~7 ms (f32) per
+// batch-24 launch vs ~0.5 ms of algorithmic traffic. Plain 2-D stencil:
+// 5 input rows staged to LDS, per-thread weights in registers, PX
+// contiguous x positions per thread so each input value loads once.
+// out[n,z,y,x,k] = bias[k] + sum_{dy,dx} in[n,z,y+dy-2,x+dx-2] * w[k][tap]
+template <typename T, int XI, int PX>
+__global__ __launch_bounds__(256, 2) void k_conv155_c1(
+    const T* __restrict__ in, const T* __restrict__ wgt,
+    const float* __restrict__ bias, T* __restrict__ out, int N, int D,
+    int H, int W, int K) {
+    __shared__ float s_in[5][XI + 4];
+    const int tid = threadIdx.x;
+    const int kslot = tid & 31;
+    const int pslot = tid >> 5;          // 8 position-slots
+    const int nz = blockIdx.z;
+    const int y = blockIdx.y;
+    const int x0 = blockIdx.x * XI;
+
+    // stage 5 input rows (zero-padded at y/x edges), batched
+    const long long plane = (long long)nz * H;
+    for (int idx = tid; idx < 5 * (XI + 4); idx += 256) {
+        const int dy = idx / (XI + 4);
+        const int xl = idx % (XI + 4);
+        const int gy = y + dy - 2;
+        const int gx = x0 + xl - 2;
+        const bool ok = gy >= 0 && gy < H && gx >= 0 && gx < W;
+        s_in[dy][xl] = ok ? ldf(&in[(plane + gy) * W + gx]) : 0.f;
+    }
+    float wreg[25];
+#pragma unroll
+    for (int t = 0; t < 25; ++t)
+        wreg[t] = kslot < K ? ldf(&wgt[kslot * 25 + t]) : 0.f;
+    __syncthreads();
+
+    if (kslot >= K) return;
+    float acc[PX];
+    const float bj = bias ? bias[kslot] : 0.f;
+#pragma unroll
+    for (int p = 0; p < PX; ++p) acc[p] = bj;
+    // per-thread x window into registers: rows 5 x (PX + 4)
+    float v[5][PX + 4];
+    const int xl0 = pslot * PX;          // contiguous block per thread
+#pragma unroll
+    for (int dy = 0; dy < 5; ++dy)
+#pragma unroll
+        for (int i = 0; i < PX + 4; ++i) v[dy][i] = s_in[dy][xl0 + i];
+#pragma unroll
+    for (int dy = 0; dy < 5; ++dy)
+#pragma unroll
+        for (int dx = 0; dx < 5; ++dx) {
+            const float w = wreg[dy * 5 + dx];
+#pragma unroll
+            for (int p = 0; p < PX; ++p) acc[p] += v[dy][p + dx] * w;
+        }
+    const long long orow = (plane + y) * (long long)W * K;
+#pragma unroll
+    for (int p = 0; p < PX; ++p) {
+        const int gx = x0 + xl0 + p;
+        if (gx < W) stf(&out[orow + (long long)gx * K + kslot], acc[p]);
+    }
+}
+
 }  // namespace
+
+extern "C" int cfx_conv155_c1(cfx_ctx* ctx, const void* in, const void* wgt,
+                              const float* bias, void* out, int N, int D,
+                              int H, int W, int K, int is_bf16) {
+    if (K > 32) {
+        g_err = "cfx_conv155_c1: K <= 32 supported";
+        return -1;
+    }
+    constexpr int XI = 64, PX = 8;
+    dim3 grid((W + XI - 1) / XI, H, (unsigned)(N * D));
+    hipEvent_t e0;
+    if (prof_begin(ctx, &e0)) return -1;
+    if (is_bf16)
+        hipLaunchKernelGGL((k_conv155_c1<cfx_bf16, XI, PX>), grid,
+                           dim3(256), 0, ctx->stream, (const cfx_bf16*)in,
+                           (const cfx_bf16*)wgt, bias, (cfx_bf16*)out, N,
+                           D, H, W, K);
+    else
+        hipLaunchKernelGGL((k_conv155_c1<float, XI, PX>), grid, dim3(256),
+                           0, ctx->stream, (const float*)in,
+                           (const float*)wgt, bias, (float*)out, N, D, H,
+                           W, K);
+    CFX_CHECK(hipGetLastError());
+    double flops = 2.0 * 25.0 * K * (double)N * D * H * W;
+    if (prof_end(ctx, e0, CFX_K_CONV, flops)) return -1;
+    return 0;
+}
 
 extern "C" int cfx_upconv_2x2(cfx_ctx* ctx, const void* in, const void* wgt,
                               const float* bias, void* out, int N, int D,

@@ -55,18 +55,20 @@ class PyTorchEngine(EngineBase):
             # kernels (fastconv.py); CFX_FASTCONV=0 keeps MIOpen everywhere
             if os.environ.get('CFX_FASTCONV', '1') != '0':
                 idx = int(str(self.device).split(':')[-1])                     if ':' in str(self.device) else 0
-                from ..fastconv import accelerate_updown
-                if self.dtype == 'bfloat16':
+                from ..fastconv import (accelerate_conv_in,
+                                        accelerate_updown)
+                bf = self.dtype == 'bfloat16'
+                if bf:
                     from ..fastconv import maybe_accelerate_bf16
                     self.fastconv_count = maybe_accelerate_bf16(
                         self.model, idx)
-                    self.fastconv_count += accelerate_updown(
-                        self.model, idx, bf16=True)
                 else:
                     from ..fastconv import maybe_accelerate
                     self.fastconv_count = maybe_accelerate(self.model, idx)
-                    self.fastconv_count += accelerate_updown(
-                        self.model, idx, bf16=False)
+                self.fastconv_count += accelerate_updown(
+                    self.model, idx, bf16=bf)
+                self.fastconv_count += accelerate_conv_in(
+                    self.model, idx, bf16=bf)
         torch.backends.cudnn.benchmark = True
         self.pre_process = getattr(net_source, 'pre_process', None)
         self.post_process = getattr(net_source, 'post_process', None)

@@ -118,12 +118,16 @@ def maybe_accelerate(model: nn.Module, device_index: int = 0) -> int:
     returns the replacement count."""
     count = 0
     dev = f'cuda:{device_index}'
+    fused_out = set()  # detached originals (their convs must not re-count)
     for parent in list(model.modules()):  # snapshot: we mutate the tree
+        if id(parent) in fused_out:
+            continue
         for name, child in list(parent.named_children()):
             if _resblock_like(child):
                 fused = CfxResBlock(child, device_index).to(dev)
                 if _fusion_matches(child, fused):
                     setattr(parent, name, fused)
+                    fused_out.add(id(child))
                     count += 1
                     continue
                 # forward does something else: fall through to lone convs
@@ -230,12 +234,16 @@ def maybe_accelerate_bf16(model: nn.Module, device_index: int = 0) -> int:
         return bool(torch.allclose(got.float(), want.float(), rtol=0.05,
                                    atol=0.05))
 
+    fused_out = set()
     for parent in list(model.modules()):  # snapshot: we mutate the tree
+        if id(parent) in fused_out:
+            continue
         for name, child in list(parent.named_children()):
             if rb(child):
                 fused = CfxResBlockBF16(child, device_index).to(dev)
                 if matches(child, fused):
                     setattr(parent, name, fused)
+                    fused_out.add(id(child))
                     count += 1
                     continue
                 for cn in ('conv1', 'conv2'):
@@ -323,12 +331,21 @@ class CfxDownConv3d(nn.Module):
         return out
 
 
+# surgery allowlist (measured, profiles/updown_probe_r02.json): the hand
+# up-conv wins at in_channels 36 (7.1x bf16 / 2.1x f32) and 48 (tie bf16 /
+# 1.4x f32); it loses at 64 (32^2 extent) and the down shapes lose to
+# MIOpen everywhere (0.03-0.25x) -- those keep MIOpen; kernels stay
+# callable and GPU-tested for the record.
+UP_SURGERY_WIDTHS = (36, 48)
+
+
 def _up_eligible(m) -> bool:
     return (isinstance(m, nn.ConvTranspose3d)
             and m.kernel_size == (1, 2, 2) and m.stride == (1, 2, 2)
             and m.padding == (0, 0, 0) and m.output_padding == (0, 0, 0)
             and m.dilation == (1, 1, 1) and m.groups == 1
-            and m.in_channels <= 64 and m.out_channels <= 64)
+            and m.in_channels in UP_SURGERY_WIDTHS
+            and m.out_channels <= 64)
 
 
 def _down_eligible(m) -> bool:
@@ -361,9 +378,78 @@ def accelerate_updown(model: nn.Module, device_index: int = 0,
             repl = None
             if _up_eligible(child):
                 repl = CfxUpConv3d(child, device_index, bf16=bf16).to(dev)
-            elif _down_eligible(child):
-                repl = CfxDownConv3d(child, device_index, bf16=bf16).to(dev)
             if repl is not None and _updown_matches(child, repl, dev, bf16):
                 setattr(parent, name, repl)
                 count += 1
+    return count
+
+
+class CfxConvIn155(nn.Module):
+    """Drop-in for nn.Conv3d(1, K, (1,5,5), padding=(0,2,2)) — the RSUNet
+    input conv. MIOpen's implicit GEMM degenerates on the single input
+    channel (measured 38.8 ms bf16 / 6.8 ms f32 per batch-24 launch vs
+    ~0.5 ms algorithmic); csrc/updown.hip runs it as a 2-D stencil."""
+
+    def __init__(self, conv: nn.Conv3d, device_index: int = 0,
+                 bf16: bool = False):
+        super().__init__()
+        self.K = conv.out_channels
+        self.device_index = device_index
+        self.bf16 = bf16
+        w = conv.weight.detach().float()      # (K, 1, 1, 5, 5)
+        dt = torch.bfloat16 if bf16 else torch.float32
+        self.register_buffer('wpack',
+                             w.reshape(self.K, 25).to(dt).contiguous())
+        if conv.bias is not None:
+            self.register_buffer('bias', conv.bias.detach().float())
+        else:
+            self.bias = None
+
+    def forward(self, x):
+        x = x.contiguous(memory_format=torch.channels_last_3d)
+        n, c, d, h, w = x.shape
+        assert c == 1
+        dt = torch.bfloat16 if self.bf16 else torch.float32
+        assert x.dtype == dt
+        out = torch.empty((n, self.K, d, h, w), dtype=dt, device=x.device,
+                          memory_format=torch.channels_last_3d)
+        get_cfx(self.device_index).conv155_c1(
+            x.data_ptr(), self.wpack.data_ptr(),
+            self.bias.data_ptr() if self.bias is not None else None,
+            out.data_ptr(), n, d, h, w, self.K, bf16=self.bf16)
+        return out
+
+
+def _conv155_eligible(m) -> bool:
+    return (isinstance(m, nn.Conv3d) and m.kernel_size == (1, 5, 5)
+            and m.stride == (1, 1, 1) and m.padding == (0, 2, 2)
+            and m.dilation == (1, 1, 1) and m.groups == 1
+            and m.in_channels == 1 and m.out_channels <= 32)
+
+
+@torch.no_grad()
+def _conv155_matches(orig, repl, dev, bf16):
+    x = torch.randn(1, 1, 3, 14, 19, device=dev)
+    if bf16:
+        x = x.to(torch.bfloat16)
+    x = x.contiguous(memory_format=torch.channels_last_3d)
+    want = orig.to(dev)(x).float()
+    got = repl(x).float()
+    tol = 0.05 if bf16 else 1e-4
+    return bool(torch.allclose(got, want, rtol=tol, atol=tol))
+
+
+def accelerate_conv_in(model: nn.Module, device_index: int = 0,
+                       bf16: bool = False) -> int:
+    """Swap the eligible single-channel (1,5,5) input convs; returns
+    count."""
+    count = 0
+    dev = f'cuda:{device_index}'
+    for parent in list(model.modules()):
+        for name, child in list(parent.named_children()):
+            if _conv155_eligible(child):
+                repl = CfxConvIn155(child, device_index, bf16=bf16).to(dev)
+                if _conv155_matches(child, repl, dev, bf16):
+                    setattr(parent, name, repl)
+                    count += 1
     return count

@@ -165,6 +165,12 @@ int cfx_upconv_2x2(cfx_ctx* ctx, const void* in, const void* wgt,
 int cfx_downconv_2x2(cfx_ctx* ctx, const void* in, const void* wgt,
                      const float* bias, void* out, int N, int D, int H,
                      int W, int C, int K, int is_bf16);
+/* single-channel (1,5,5) conv, pad (0,2,2) — RSUNet conv_in; in
+ * (N,D,H,W,1), out (N,D,H,W,K) NDHWC; wgt [K][25] row-major (dy*5+dx) in
+ * the compute dtype; K <= 32 */
+int cfx_conv155_c1(cfx_ctx* ctx, const void* in, const void* wgt,
+                   const float* bias, void* out, int N, int D, int H,
+                   int W, int K, int is_bf16);
 /* the 32x32x2-MFMA variant (C == K == 28 instantiated) */
 int cfx_conv3_ndhwc_w32(cfx_ctx* ctx, const float* in, const float* wgt,
                         const float* bias, const float* residual,

@@ -895,11 +895,26 @@ def test_rsunet_bf16_full_surgery_runs():
     m2 = copy.deepcopy(model)
     n_rb = maybe_accelerate_bf16(m2, 0)
     n_ud = accelerate_updown(m2, 0, bf16=True)
-    from chunkflow_amd.fastconv import BF16_WIDTHS
+    from chunkflow_amd.fastconv import BF16_WIDTHS, UP_SURGERY_WIDTHS
     want_rb = 2 * len([w for w in (28, 36, 48) if w in BF16_WIDTHS])
     assert n_rb == want_rb, (n_rb, want_rb)  # enc+dec blocks per width
-    assert n_ud == 6, n_ud   # 3 down + 3 up
+    assert n_ud == len(UP_SURGERY_WIDTHS), n_ud  # measured-winning ups
     with torch.no_grad():
         got = m2(x).float()
     # sigmoid output in [0,1]: absolute tolerance
     torch.testing.assert_close(got, want, rtol=0.1, atol=0.03)
+
+
+@pytest.mark.parametrize('bf16', [False, True])
+def test_conv155_c1_vs_torch(bf16):
+    """Single-channel (1,5,5) input-conv stencil vs torch/MIOpen."""
+    from chunkflow_amd.fastconv import CfxConvIn155
+    torch.manual_seed(9)
+    conv = torch.nn.Conv3d(1, 28, (1, 5, 5), padding=(0, 2, 2)).cuda()
+    x = torch.randn(2, 1, 3, 37, 83, device='cuda')
+    dt = torch.bfloat16 if bf16 else torch.float32
+    xc = x.to(dt).contiguous(memory_format=torch.channels_last_3d)
+    want = conv.to(dt)(xc).float()
+    got = CfxConvIn155(conv, 0, bf16=bf16).cuda()(xc).float()
+    tol = 0.05 if bf16 else 1e-4
+    torch.testing.assert_close(got, want, rtol=tol, atol=tol)
