@@ -199,22 +199,25 @@ __global__ __launch_bounds__(256, 2) void k_downconv2(
 // 5 input rows staged to LDS, per-thread weights in registers, PX
 // contiguous x positions per thread so each input value loads once.
 // out[n,z,y,x,k] = bias[k] + sum_{dy,dx} in[n,z,y+dy-2,x+dx-2] * w[k][tap]
-template <typename T, int XI, int PX>
+template <typename T, int XI, int KC>  // KC: compile-time K (0 = runtime)
 __global__ __launch_bounds__(256, 2) void k_conv155_c1(
     const T* __restrict__ in, const T* __restrict__ wgt,
     const float* __restrict__ bias, T* __restrict__ out, int N, int D,
-    int H, int W, int K) {
+    int H, int W, int K_rt) {
+    const int K = KC > 0 ? KC : K_rt;
+    // thread = one output position, computing ALL K: the K outputs of one
+    // position are contiguous in NDHWC, so each thread's stores are a
+    // K-element run and a wave's stores coalesce (the k-parallel variant
+    // wrote 2-byte strided singles and was store-issue-bound, 5.3 ms)
     __shared__ float s_in[5][XI + 4];
-    const int tid = threadIdx.x;
-    const int kslot = tid & 31;
-    const int pslot = tid >> 5;          // 8 position-slots
+    __shared__ float s_w[32 * 28];        // rows padded 25 -> 28 (16 B)
+    const int tid = threadIdx.x;          // position slot (XI == 256)
     const int nz = blockIdx.z;
     const int y = blockIdx.y;
     const int x0 = blockIdx.x * XI;
 
-    // stage 5 input rows (zero-padded at y/x edges), batched
     const long long plane = (long long)nz * H;
-    for (int idx = tid; idx < 5 * (XI + 4); idx += 256) {
+    for (int idx = tid; idx < 5 * (XI + 4); idx += XI) {
         const int dy = idx / (XI + 4);
         const int xl = idx % (XI + 4);
         const int gy = y + dy - 2;
@@ -222,37 +225,65 @@ __global__ __launch_bounds__(256, 2) void k_conv155_c1(
         const bool ok = gy >= 0 && gy < H && gx >= 0 && gx < W;
         s_in[dy][xl] = ok ? ldf(&in[(plane + gy) * W + gx]) : 0.f;
     }
-    float wreg[25];
-#pragma unroll
-    for (int t = 0; t < 25; ++t)
-        wreg[t] = kslot < K ? ldf(&wgt[kslot * 25 + t]) : 0.f;
+    for (int idx = tid; idx < K * 28; idx += XI) {
+        const int t = idx % 28;
+        const int k = idx / 28;
+        s_w[idx] = t < 25 ? ldf(&wgt[k * 25 + t]) : 0.f;
+    }
     __syncthreads();
 
-    if (kslot >= K) return;
-    float acc[PX];
-    const float bj = bias ? bias[kslot] : 0.f;
-#pragma unroll
-    for (int p = 0; p < PX; ++p) acc[p] = bj;
-    // per-thread x window into registers: rows 5 x (PX + 4)
-    float v[5][PX + 4];
-    const int xl0 = pslot * PX;          // contiguous block per thread
+    const int gx = x0 + tid;
+    if (gx >= W) return;
+    // the 5x5 window of this position, registers
+    float v[25];
 #pragma unroll
     for (int dy = 0; dy < 5; ++dy)
 #pragma unroll
-        for (int i = 0; i < PX + 4; ++i) v[dy][i] = s_in[dy][xl0 + i];
+        for (int dx = 0; dx < 5; ++dx)
+            v[dy * 5 + dx] = s_in[dy][tid + dx];
+    T* op = out + (plane + y) * (long long)W * K + (long long)gx * K;
+    if (KC > 0) {
+        // compile-time K: accumulators in registers, one contiguous
+        // K-run stored in 8-byte chunks (the largest alignment every
+        // K * sizeof(T) row stride guarantees)
+        T tmp[KC > 0 ? KC : 1];
 #pragma unroll
-    for (int dy = 0; dy < 5; ++dy)
+        for (int k = 0; k < KC; ++k) {
+            float acc = bias ? bias[k] : 0.f;
 #pragma unroll
-        for (int dx = 0; dx < 5; ++dx) {
-            const float w = wreg[dy * 5 + dx];
-#pragma unroll
-            for (int p = 0; p < PX; ++p) acc[p] += v[dy][p + dx] * w;
+            for (int t4 = 0; t4 < 24; t4 += 4) {
+                const float4 w4 =
+                    *reinterpret_cast<const float4*>(&s_w[k * 28 + t4]);
+                acc += v[t4] * w4.x + v[t4 + 1] * w4.y +
+                       v[t4 + 2] * w4.z + v[t4 + 3] * w4.w;
+            }
+            acc += v[24] * s_w[k * 28 + 24];
+            tmp[k] = (T)acc;
         }
-    const long long orow = (plane + y) * (long long)W * K;
+        constexpr int NB = (int)(KC * sizeof(T));
+        char* dst = reinterpret_cast<char*>(op);
+        const char* src = reinterpret_cast<const char*>(tmp);
 #pragma unroll
-    for (int p = 0; p < PX; ++p) {
-        const int gx = x0 + xl0 + p;
-        if (gx < W) stf(&out[orow + (long long)gx * K + kslot], acc[p]);
+        for (int ofs = 0; ofs + 8 <= NB; ofs += 8)
+            *reinterpret_cast<uint64_t*>(dst + ofs) =
+                *reinterpret_cast<const uint64_t*>(src + ofs);
+#pragma unroll
+        for (int ofs = NB & ~7; ofs < NB; ofs += (int)sizeof(T))
+            *reinterpret_cast<T*>(dst + ofs) =
+                *reinterpret_cast<const T*>(src + ofs);
+    } else {
+        for (int k = 0; k < K; ++k) {
+            float acc = bias ? bias[k] : 0.f;
+#pragma unroll
+            for (int t4 = 0; t4 < 24; t4 += 4) {
+                const float4 w4 =
+                    *reinterpret_cast<const float4*>(&s_w[k * 28 + t4]);
+                acc += v[t4] * w4.x + v[t4 + 1] * w4.y +
+                       v[t4 + 2] * w4.z + v[t4 + 3] * w4.w;
+            }
+            acc += v[24] * s_w[k * 28 + 24];
+            stf(&op[k], acc);
+        }
     }
 }
 
@@ -265,17 +296,27 @@ extern "C" int cfx_conv155_c1(cfx_ctx* ctx, const void* in, const void* wgt,
         g_err = "cfx_conv155_c1: K <= 32 supported";
         return -1;
     }
-    constexpr int XI = 64, PX = 8;
+    constexpr int XI = 256;
     dim3 grid((W + XI - 1) / XI, H, (unsigned)(N * D));
     hipEvent_t e0;
     if (prof_begin(ctx, &e0)) return -1;
-    if (is_bf16)
-        hipLaunchKernelGGL((k_conv155_c1<cfx_bf16, XI, PX>), grid,
-                           dim3(256), 0, ctx->stream, (const cfx_bf16*)in,
+    if (is_bf16 && K == 28)
+        hipLaunchKernelGGL((k_conv155_c1<cfx_bf16, XI, 28>), grid,
+                           dim3(XI), 0, ctx->stream, (const cfx_bf16*)in,
                            (const cfx_bf16*)wgt, bias, (cfx_bf16*)out, N,
                            D, H, W, K);
+    else if (is_bf16)
+        hipLaunchKernelGGL((k_conv155_c1<cfx_bf16, XI, 0>), grid,
+                           dim3(XI), 0, ctx->stream, (const cfx_bf16*)in,
+                           (const cfx_bf16*)wgt, bias, (cfx_bf16*)out, N,
+                           D, H, W, K);
+    else if (K == 28)
+        hipLaunchKernelGGL((k_conv155_c1<float, XI, 28>), grid, dim3(XI),
+                           0, ctx->stream, (const float*)in,
+                           (const float*)wgt, bias, (float*)out, N, D, H,
+                           W, K);
     else
-        hipLaunchKernelGGL((k_conv155_c1<float, XI, PX>), grid, dim3(256),
+        hipLaunchKernelGGL((k_conv155_c1<float, XI, 0>), grid, dim3(XI),
                            0, ctx->stream, (const float*)in,
                            (const float*)wgt, bias, (float*)out, N, D, H,
                            W, K);
