@@ -32,6 +32,30 @@ template <>
 __device__ __forceinline__ float ldf<cfx_bf16>(const cfx_bf16* p) {
     return __bfloat162float(*p);
 }
+__device__ __forceinline__ float cvf(float v) { return v; }
+__device__ __forceinline__ float cvf(cfx_bf16 v) {
+    return __bfloat162float(v);
+}
+
+// 4-wide raw vector per dtype (__hip_bfloat16 is a struct, so its raw
+// bits travel as ushort lanes)
+template <typename T>
+struct vec4;
+template <>
+struct vec4<float> {
+    typedef float type __attribute__((ext_vector_type(4)));
+    static __device__ __forceinline__ float get(type v, int i) {
+        return v[i];
+    }
+};
+template <>
+struct vec4<cfx_bf16> {
+    typedef unsigned short type __attribute__((ext_vector_type(4)));
+    static __device__ __forceinline__ float get(type v, int i) {
+        return __bfloat162float(
+            __hip_bfloat16(__hip_bfloat16_raw{(unsigned short)v[i]}));
+    }
+};
 template <typename T>
 __device__ __forceinline__ void stf(T* p, float v);
 template <>
@@ -242,35 +266,67 @@ __global__ __launch_bounds__(256, 2) void k_conv155_c1(
         for (int dx = 0; dx < 5; ++dx)
             v[dy * 5 + dx] = s_in[dy][tid + dx];
     T* op = out + (plane + y) * (long long)W * K + (long long)gx * K;
-    if (KC > 0) {
-        // compile-time K: accumulators in registers, one contiguous
-        // K-run stored in 8-byte chunks (the largest alignment every
-        // K * sizeof(T) row stride guarantees)
-        T tmp[KC > 0 ? KC : 1];
-#pragma unroll
-        for (int k = 0; k < KC; ++k) {
-            float acc = bias ? bias[k] : 0.f;
+    if (KC > 0 && KC % 4 == 0) {
+        // compile-time K in groups of 4: scalar accumulators packed into
+        // 8-byte vector stores. No register array address-taking (spills
+        // to scratch) and no full k unroll (28x7 hoisted float4 weight
+        // reads blow the register file) — both measured ~10x slower.
+        typedef unsigned int uint2v __attribute__((ext_vector_type(2)));
+        typedef float float2v __attribute__((ext_vector_type(2)));
+#pragma unroll 1
+        for (int k = 0; k < KC; k += 4) {
+            float a0 = bias ? bias[k] : 0.f;
+            float a1 = bias ? bias[k + 1] : 0.f;
+            float a2 = bias ? bias[k + 2] : 0.f;
+            float a3 = bias ? bias[k + 3] : 0.f;
 #pragma unroll
             for (int t4 = 0; t4 < 24; t4 += 4) {
-                const float4 w4 =
-                    *reinterpret_cast<const float4*>(&s_w[k * 28 + t4]);
-                acc += v[t4] * w4.x + v[t4 + 1] * w4.y +
-                       v[t4 + 2] * w4.z + v[t4 + 3] * w4.w;
+                const float4 w0 = *reinterpret_cast<const float4*>(
+                    &s_w[k * 28 + t4]);
+                const float4 w1 = *reinterpret_cast<const float4*>(
+                    &s_w[(k + 1) * 28 + t4]);
+                const float4 w2 = *reinterpret_cast<const float4*>(
+                    &s_w[(k + 2) * 28 + t4]);
+                const float4 w3 = *reinterpret_cast<const float4*>(
+                    &s_w[(k + 3) * 28 + t4]);
+                a0 += v[t4] * w0.x + v[t4 + 1] * w0.y + v[t4 + 2] * w0.z +
+                      v[t4 + 3] * w0.w;
+                a1 += v[t4] * w1.x + v[t4 + 1] * w1.y + v[t4 + 2] * w1.z +
+                      v[t4 + 3] * w1.w;
+                a2 += v[t4] * w2.x + v[t4 + 1] * w2.y + v[t4 + 2] * w2.z +
+                      v[t4 + 3] * w2.w;
+                a3 += v[t4] * w3.x + v[t4 + 1] * w3.y + v[t4 + 2] * w3.z +
+                      v[t4 + 3] * w3.w;
             }
-            acc += v[24] * s_w[k * 28 + 24];
-            tmp[k] = (T)acc;
+            a0 += v[24] * s_w[k * 28 + 24];
+            a1 += v[24] * s_w[(k + 1) * 28 + 24];
+            a2 += v[24] * s_w[(k + 2) * 28 + 24];
+            a3 += v[24] * s_w[(k + 3) * 28 + 24];
+            if (sizeof(T) == 2) {
+                uint2v u;
+                u.x = (unsigned)((__hip_bfloat16_raw)__float2bfloat16(
+                          a0)).x |
+                      ((unsigned)((__hip_bfloat16_raw)__float2bfloat16(
+                           a1)).x << 16);
+                u.y = (unsigned)((__hip_bfloat16_raw)__float2bfloat16(
+                          a2)).x |
+                      ((unsigned)((__hip_bfloat16_raw)__float2bfloat16(
+                           a3)).x << 16);
+                *reinterpret_cast<uint2v*>(reinterpret_cast<char*>(op) +
+                                           (size_t)k * 2) = u;
+            } else {
+                float2v f01, f23;
+                f01.x = a0;
+                f01.y = a1;
+                f23.x = a2;
+                f23.y = a3;
+                float2v* d =
+                    reinterpret_cast<float2v*>(reinterpret_cast<char*>(op) +
+                                               (size_t)k * 4);
+                d[0] = f01;
+                d[1] = f23;
+            }
         }
-        constexpr int NB = (int)(KC * sizeof(T));
-        char* dst = reinterpret_cast<char*>(op);
-        const char* src = reinterpret_cast<const char*>(tmp);
-#pragma unroll
-        for (int ofs = 0; ofs + 8 <= NB; ofs += 8)
-            *reinterpret_cast<uint64_t*>(dst + ofs) =
-                *reinterpret_cast<const uint64_t*>(src + ofs);
-#pragma unroll
-        for (int ofs = NB & ~7; ofs < NB; ofs += (int)sizeof(T))
-            *reinterpret_cast<T*>(dst + ofs) =
-                *reinterpret_cast<const T*>(src + ofs);
     } else {
         for (int k = 0; k < K; ++k) {
             float acc = bias ? bias[k] : 0.f;
@@ -287,7 +343,107 @@ __global__ __launch_bounds__(256, 2) void k_conv155_c1(
     }
 }
 
+// ---- output conv (1,5,5), pad (0,2,2), few output channels ---------------
+// RSUNet's conv_out (28 -> 3). MIOpen's bf16 implicit GEMM degenerates on
+// the K=3 output dim (measured 38.6 ms per batch-24 launch). Thread =
+// one output position computing all K: 5 input rows staged to LDS in the
+// raw dtype, weights [K][25][28-padded] f32 in LDS, f32 accumulation.
+template <typename T, int XI, int CC, int KO>
+__global__ __launch_bounds__(XI, 2) void k_conv155_out(
+    const T* __restrict__ in, const T* __restrict__ wgt,
+    const float* __restrict__ bias, T* __restrict__ out, int N, int D,
+    int H, int W) {
+    __shared__ T s_in[5][XI + 4][CC];
+    __shared__ float s_w[KO][25][28];     // c rows padded to 28 (16 B)
+    const int tid = threadIdx.x;
+    const int nz = blockIdx.z;
+    const int y = blockIdx.y;
+    const int x0 = blockIdx.x * XI;
+
+    const long long plane = (long long)nz * H;
+    const int C4 = CC / 4;
+    typedef typename vec4<T>::type tx4;
+    for (int idx = tid; idx < 5 * (XI + 4) * C4; idx += XI) {
+        const int c4 = idx % C4;
+        const int xl = (idx / C4) % (XI + 4);
+        const int dy = idx / (C4 * (XI + 4));
+        const int gy = y + dy - 2;
+        const int gx = x0 + xl - 2;
+        const bool ok = gy >= 0 && gy < H && gx >= 0 && gx < W;
+        *reinterpret_cast<tx4*>(&s_in[dy][xl][c4 * 4]) =
+            ok ? *reinterpret_cast<const tx4*>(
+                     &in[((plane + gy) * W + gx) * (long long)CC + c4 * 4])
+               : tx4{};
+    }
+    for (int idx = tid; idx < KO * 25 * 28; idx += XI) {
+        const int c = idx % 28;
+        const int t = (idx / 28) % 25;
+        const int k = idx / (28 * 25);
+        s_w[k][t][c] = c < CC ? ldf(&wgt[(k * 25 + t) * CC + c]) : 0.f;
+    }
+    __syncthreads();
+
+    const int gx = x0 + tid;
+    if (gx >= W) return;
+    float acc[KO];
+#pragma unroll
+    for (int k = 0; k < KO; ++k) acc[k] = bias ? bias[k] : 0.f;
+#pragma unroll 1
+    for (int tap = 0; tap < 25; ++tap) {
+        const int dy = tap / 5, dx = tap % 5;
+        const T* ip = &s_in[dy][tid + dx][0];
+#pragma unroll
+        for (int c4 = 0; c4 < C4; ++c4) {
+            const tx4 v4 = *reinterpret_cast<const tx4*>(&ip[c4 * 4]);
+            const float f0 = vec4<T>::get(v4, 0);
+            const float f1 = vec4<T>::get(v4, 1);
+            const float f2 = vec4<T>::get(v4, 2);
+            const float f3 = vec4<T>::get(v4, 3);
+#pragma unroll
+            for (int k = 0; k < KO; ++k) {
+                const float4 w4 = *reinterpret_cast<const float4*>(
+                    &s_w[k][tap][c4 * 4]);
+                acc[k] += f0 * w4.x + f1 * w4.y + f2 * w4.z + f3 * w4.w;
+            }
+        }
+    }
+    T* op = out + (plane + y) * (long long)W * KO + (long long)gx * KO;
+#pragma unroll
+    for (int k = 0; k < KO; ++k) stf(&op[k], acc[k]);
+}
+
 }  // namespace
+
+extern "C" int cfx_conv155_out(cfx_ctx* ctx, const void* in,
+                               const void* wgt, const float* bias,
+                               void* out, int N, int D, int H, int W,
+                               int C, int K, int is_bf16) {
+    if (C != 28 || K != 3) {
+        g_err = "cfx_conv155_out: only C == 28, K == 3 instantiated";
+        return -1;
+    }
+    hipEvent_t e0;
+    if (prof_begin(ctx, &e0)) return -1;
+    if (is_bf16) {
+        constexpr int XI = 256;
+        dim3 grid((W + XI - 1) / XI, H, (unsigned)(N * D));
+        hipLaunchKernelGGL((k_conv155_out<cfx_bf16, XI, 28, 3>), grid,
+                           dim3(XI), 0, ctx->stream, (const cfx_bf16*)in,
+                           (const cfx_bf16*)wgt, bias, (cfx_bf16*)out, N,
+                           D, H, W);
+    } else {
+        constexpr int XI = 128;  // f32 LDS: 5 x 132 x 28 x 4 B
+        dim3 grid((W + XI - 1) / XI, H, (unsigned)(N * D));
+        hipLaunchKernelGGL((k_conv155_out<float, XI, 28, 3>), grid,
+                           dim3(XI), 0, ctx->stream, (const float*)in,
+                           (const float*)wgt, bias, (float*)out, N, D, H,
+                           W);
+    }
+    CFX_CHECK(hipGetLastError());
+    double flops = 2.0 * 25.0 * C * K * (double)N * D * H * W;
+    if (prof_end(ctx, e0, CFX_K_CONV, flops)) return -1;
+    return 0;
+}
 
 extern "C" int cfx_conv155_c1(cfx_ctx* ctx, const void* in, const void* wgt,
                               const float* bias, void* out, int N, int D,

@@ -453,3 +453,74 @@ def accelerate_conv_in(model: nn.Module, device_index: int = 0,
                     setattr(parent, name, repl)
                     count += 1
     return count
+
+
+class CfxConvOut155(nn.Module):
+    """Drop-in for nn.Conv3d(C, K<=3-ish, (1,5,5), padding=(0,2,2)) — the
+    RSUNet output conv. MIOpen's bf16 implicit GEMM degenerates on the
+    tiny K (measured 38.6 ms per batch-24 launch vs ~2 ms here)."""
+
+    def __init__(self, conv: nn.Conv3d, device_index: int = 0,
+                 bf16: bool = False):
+        super().__init__()
+        self.C = conv.in_channels
+        self.K = conv.out_channels
+        self.device_index = device_index
+        self.bf16 = bf16
+        w = conv.weight.detach().float()      # (K, C, 1, 5, 5)
+        pack = w[:, :, 0].permute(0, 2, 3, 1).reshape(self.K, 25, self.C)
+        dt = torch.bfloat16 if bf16 else torch.float32
+        self.register_buffer('wpack', pack.to(dt).contiguous())
+        if conv.bias is not None:
+            self.register_buffer('bias', conv.bias.detach().float())
+        else:
+            self.bias = None
+
+    def forward(self, x):
+        x = x.contiguous(memory_format=torch.channels_last_3d)
+        n, c, d, h, w = x.shape
+        dt = torch.bfloat16 if self.bf16 else torch.float32
+        assert x.dtype == dt and c == self.C
+        out = torch.empty((n, self.K, d, h, w), dtype=dt, device=x.device,
+                          memory_format=torch.channels_last_3d)
+        get_cfx(self.device_index).conv155_out(
+            x.data_ptr(), self.wpack.data_ptr(),
+            self.bias.data_ptr() if self.bias is not None else None,
+            out.data_ptr(), n, d, h, w, self.C, self.K, bf16=self.bf16)
+        return out
+
+
+def _conv155_out_eligible(m) -> bool:
+    return (isinstance(m, nn.Conv3d) and m.kernel_size == (1, 5, 5)
+            and m.stride == (1, 1, 1) and m.padding == (0, 2, 2)
+            and m.dilation == (1, 1, 1) and m.groups == 1
+            and m.in_channels == 28 and m.out_channels == 3)
+
+
+def accelerate_conv_out(model: nn.Module, device_index: int = 0,
+                        bf16: bool = False) -> int:
+    """Swap the eligible few-channel (1,5,5) output convs (bf16 wins big;
+    f32 MIOpen is already fine there — callers gate)."""
+    count = 0
+    dev = f'cuda:{device_index}'
+    for parent in list(model.modules()):
+        for name, child in list(parent.named_children()):
+            if _conv155_out_eligible(child):
+                repl = CfxConvOut155(child, device_index, bf16=bf16).to(dev)
+                if _conv155_matches_generic(child, repl, dev, bf16,
+                                            channels=repl.C):
+                    setattr(parent, name, repl)
+                    count += 1
+    return count
+
+
+@torch.no_grad()
+def _conv155_matches_generic(orig, repl, dev, bf16, channels=1):
+    x = torch.randn(1, channels, 3, 14, 19, device=dev)
+    if bf16:
+        x = x.to(torch.bfloat16)
+    x = x.contiguous(memory_format=torch.channels_last_3d)
+    want = orig.to(dev)(x).float()
+    got = repl(x).float()
+    tol = 0.05 if bf16 else 1e-4
+    return bool(torch.allclose(got, want, rtol=tol, atol=tol))

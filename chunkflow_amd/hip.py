@@ -27,6 +27,7 @@ SYMBOLS = [
     'cfx_conv3_ndhwc', 'cfx_conv3_ndhwc_w32',
     'cfx_conv3_ndhwc_zring', 'cfx_conv3_ndhwc_bf16',
     'cfx_upconv_2x2', 'cfx_downconv_2x2', 'cfx_conv155_c1',
+    'cfx_conv155_out',
     'cfx_profile_enable', 'cfx_profile_reset',
     'cfx_profile_get',
 ]
@@ -290,6 +291,17 @@ class CfxContext:
             ctypes.c_void_p(out_ptr), ctypes.c_int(n), ctypes.c_int(d),
             ctypes.c_int(h), ctypes.c_int(w), ctypes.c_int(k),
             ctypes.c_int(1 if bf16 else 0)), 'cfx_conv155_c1')
+
+    def conv155_out(self, in_ptr, wgt_ptr, bias_ptr, out_ptr, n, d, h, w,
+                    c, k, bf16=False):
+        self._chk(self.lib.cfx_conv155_out(
+            ctypes.c_void_p(self.ctx), ctypes.c_void_p(in_ptr),
+            ctypes.c_void_p(wgt_ptr),
+            ctypes.c_void_p(bias_ptr) if bias_ptr else None,
+            ctypes.c_void_p(out_ptr), ctypes.c_int(n), ctypes.c_int(d),
+            ctypes.c_int(h), ctypes.c_int(w), ctypes.c_int(c),
+            ctypes.c_int(k), ctypes.c_int(1 if bf16 else 0)),
+            'cfx_conv155_out')
 
     # --- image normalization -------------------------------------------------
     def hist_u8(self, in_ptr, n_per_sec, nsec, hist_ptr):

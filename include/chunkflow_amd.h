@@ -171,6 +171,12 @@ int cfx_downconv_2x2(cfx_ctx* ctx, const void* in, const void* wgt,
 int cfx_conv155_c1(cfx_ctx* ctx, const void* in, const void* wgt,
                    const float* bias, void* out, int N, int D, int H,
                    int W, int K, int is_bf16);
+/* few-output-channel (1,5,5) conv, pad (0,2,2) — RSUNet conv_out; in
+ * (N,D,H,W,C), out (N,D,H,W,K) NDHWC; wgt [K][25][C] (k, dy*5+dx, c) in
+ * the compute dtype; C == 28, K == 3 instantiated */
+int cfx_conv155_out(cfx_ctx* ctx, const void* in, const void* wgt,
+                    const float* bias, void* out, int N, int D, int H,
+                    int W, int C, int K, int is_bf16);
 /* the 32x32x2-MFMA variant (C == K == 28 instantiated) */
 int cfx_conv3_ndhwc_w32(cfx_ctx* ctx, const float* in, const float* wgt,
                         const float* bias, const float* residual,

@@ -918,3 +918,18 @@ def test_conv155_c1_vs_torch(bf16):
     got = CfxConvIn155(conv, 0, bf16=bf16).cuda()(xc).float()
     tol = 0.05 if bf16 else 1e-4
     torch.testing.assert_close(got, want, rtol=tol, atol=tol)
+
+
+@pytest.mark.parametrize('bf16', [False, True])
+def test_conv155_out_vs_torch(bf16):
+    """Few-output-channel (1,5,5) output-conv kernel vs torch/MIOpen."""
+    from chunkflow_amd.fastconv import CfxConvOut155
+    torch.manual_seed(10)
+    conv = torch.nn.Conv3d(28, 3, (1, 5, 5), padding=(0, 2, 2)).cuda()
+    x = torch.randn(2, 28, 3, 41, 77, device='cuda')
+    dt = torch.bfloat16 if bf16 else torch.float32
+    xc = x.to(dt).contiguous(memory_format=torch.channels_last_3d)
+    want = conv.to(dt)(xc).float()
+    got = CfxConvOut155(conv, 0, bf16=bf16).cuda()(xc).float()
+    tol = 0.05 if bf16 else 1e-4
+    torch.testing.assert_close(got, want, rtol=tol, atol=tol)

@@ -12,7 +12,8 @@ import torch
 import torch.nn as nn
 
 from chunkflow_amd.fastconv import (CfxUpConv3d, CfxDownConv3d,
-                                    CfxConv3dBF16, CfxConvIn155)
+                                    CfxConv3dBF16, CfxConvIn155,
+                                    CfxConvOut155)
 
 torch.backends.cudnn.benchmark = True
 cl = torch.channels_last_3d
@@ -97,6 +98,26 @@ def probe_conv155(N, D, H, W, bf16):
     print(name, RES[name], flush=True)
 
 
+def probe_conv155_out(N, D, H, W, bf16):
+    torch.manual_seed(0)
+    conv = nn.Conv3d(28, 3, (1, 5, 5), padding=(0, 2, 2)).cuda()
+    repl = CfxConvOut155(conv, 0, bf16=bf16).cuda()
+    dt = torch.bfloat16 if bf16 else torch.float32
+    tconv = conv.to(dt).to(memory_format=cl)
+    x = torch.randn(N, 28, D, H, W, device='cuda').to(dt).contiguous(
+        memory_format=cl)
+    with torch.no_grad():
+        want = tconv(x).float()
+        got = repl(x).float()
+        err = (got - want).abs().max().item()
+        t_my = timeit(lambda: repl(x))
+        t_to = timeit(lambda: tconv(x))
+    name = f'conv155_out_{"bf16" if bf16 else "f32"}_{N}x{D}x{H}x{W}'
+    RES[name] = {'mine_ms': t_my * 1e3, 'torch_ms': t_to * 1e3,
+                 'speedup': t_to / t_my, 'max_err': err}
+    print(name, RES[name], flush=True)
+
+
 def main():
     N, D = 12, 20  # config-2-like depth; config-5 uses N=24 D=32
     # RSUNet up/down shapes (H, W = input dims of the op)
@@ -114,6 +135,8 @@ def main():
     probe_sliced(48, 24, 32, 64, 64)
     probe_conv155(12, 20, 256, 256, False)
     probe_conv155(24, 32, 256, 256, True)
+    probe_conv155_out(12, 20, 256, 256, False)
+    probe_conv155_out(24, 32, 256, 256, True)
     os.makedirs('gpurun_out', exist_ok=True)
     with open('gpurun_out/updown_probe.json', 'w') as f:
         json.dump(RES, f, indent=1)
