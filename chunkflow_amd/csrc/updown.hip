@@ -349,33 +349,39 @@ __global__ __launch_bounds__(256, 2) void k_conv155_c1(
 // one output position computing all K: 5 input rows staged to LDS in the
 // raw dtype, weights [K][25][28-padded] f32 in LDS, f32 accumulation.
 template <typename T, int XI, int CC, int KO>
-__global__ __launch_bounds__(XI, 2) void k_conv155_out(
+__global__ __launch_bounds__(256, 2) void k_conv155_out(
     const T* __restrict__ in, const T* __restrict__ wgt,
     const float* __restrict__ bias, T* __restrict__ out, int N, int D,
     int H, int W) {
-    __shared__ T s_in[5][XI + 4][CC];
+    // One WG = 4 output rows x XI x-positions: the 8 staged input rows
+    // are shared by the 4 rows (2 LDS rows per output instead of 5 --
+    // the single-row version was LDS-bound at 1 WG/CU and latency-bound).
+    // 256 threads = 4 waves, wave = local y; lane covers 2 x positions.
+    __shared__ T s_in[8][XI + 4][CC];
     __shared__ float s_w[KO][25][28];     // c rows padded to 28 (16 B)
     const int tid = threadIdx.x;
+    const int wave = tid >> 6;
+    const int lane = tid & 63;
     const int nz = blockIdx.z;
-    const int y = blockIdx.y;
+    const int y0 = blockIdx.y * 4;
     const int x0 = blockIdx.x * XI;
 
     const long long plane = (long long)nz * H;
     const int C4 = CC / 4;
     typedef typename vec4<T>::type tx4;
-    for (int idx = tid; idx < 5 * (XI + 4) * C4; idx += XI) {
+    for (int idx = tid; idx < 8 * (XI + 4) * C4; idx += 256) {
         const int c4 = idx % C4;
         const int xl = (idx / C4) % (XI + 4);
-        const int dy = idx / (C4 * (XI + 4));
-        const int gy = y + dy - 2;
+        const int r = idx / (C4 * (XI + 4));
+        const int gy = y0 + r - 2;
         const int gx = x0 + xl - 2;
         const bool ok = gy >= 0 && gy < H && gx >= 0 && gx < W;
-        *reinterpret_cast<tx4*>(&s_in[dy][xl][c4 * 4]) =
+        *reinterpret_cast<tx4*>(&s_in[r][xl][c4 * 4]) =
             ok ? *reinterpret_cast<const tx4*>(
                      &in[((plane + gy) * W + gx) * (long long)CC + c4 * 4])
                : tx4{};
     }
-    for (int idx = tid; idx < KO * 25 * 28; idx += XI) {
+    for (int idx = tid; idx < KO * 25 * 28; idx += 256) {
         const int c = idx % 28;
         const int t = (idx / 28) % 25;
         const int k = idx / (28 * 25);
@@ -383,33 +389,47 @@ __global__ __launch_bounds__(XI, 2) void k_conv155_out(
     }
     __syncthreads();
 
-    const int gx = x0 + tid;
-    if (gx >= W) return;
-    float acc[KO];
+    const int y = y0 + wave;
+    if (y >= H) return;
+    float acc[2][KO];
 #pragma unroll
-    for (int k = 0; k < KO; ++k) acc[k] = bias ? bias[k] : 0.f;
+    for (int p = 0; p < 2; ++p)
+#pragma unroll
+        for (int k = 0; k < KO; ++k) acc[p][k] = bias ? bias[k] : 0.f;
 #pragma unroll 1
     for (int tap = 0; tap < 25; ++tap) {
         const int dy = tap / 5, dx = tap % 5;
-        const T* ip = &s_in[dy][tid + dx][0];
-#pragma unroll
+#pragma unroll 1
         for (int c4 = 0; c4 < C4; ++c4) {
-            const tx4 v4 = *reinterpret_cast<const tx4*>(&ip[c4 * 4]);
-            const float f0 = vec4<T>::get(v4, 0);
-            const float f1 = vec4<T>::get(v4, 1);
-            const float f2 = vec4<T>::get(v4, 2);
-            const float f3 = vec4<T>::get(v4, 3);
+            float4 w4k[KO];
 #pragma unroll
-            for (int k = 0; k < KO; ++k) {
-                const float4 w4 = *reinterpret_cast<const float4*>(
+            for (int k = 0; k < KO; ++k)
+                w4k[k] = *reinterpret_cast<const float4*>(
                     &s_w[k][tap][c4 * 4]);
-                acc[k] += f0 * w4.x + f1 * w4.y + f2 * w4.z + f3 * w4.w;
+#pragma unroll
+            for (int p = 0; p < 2; ++p) {
+                const tx4 v4 = *reinterpret_cast<const tx4*>(
+                    &s_in[wave + dy][lane * 2 + p + dx][c4 * 4]);
+                const float f0 = vec4<T>::get(v4, 0);
+                const float f1 = vec4<T>::get(v4, 1);
+                const float f2 = vec4<T>::get(v4, 2);
+                const float f3 = vec4<T>::get(v4, 3);
+#pragma unroll
+                for (int k = 0; k < KO; ++k)
+                    acc[p][k] += f0 * w4k[k].x + f1 * w4k[k].y +
+                                 f2 * w4k[k].z + f3 * w4k[k].w;
             }
         }
     }
-    T* op = out + (plane + y) * (long long)W * KO + (long long)gx * KO;
 #pragma unroll
-    for (int k = 0; k < KO; ++k) stf(&op[k], acc[k]);
+    for (int p = 0; p < 2; ++p) {
+        const int gx = x0 + lane * 2 + p;
+        if (gx >= W) continue;
+        T* op =
+            out + (plane + y) * (long long)W * KO + (long long)gx * KO;
+#pragma unroll
+        for (int k = 0; k < KO; ++k) stf(&op[k], acc[p][k]);
+    }
 }
 
 }  // namespace
@@ -424,21 +444,18 @@ extern "C" int cfx_conv155_out(cfx_ctx* ctx, const void* in,
     }
     hipEvent_t e0;
     if (prof_begin(ctx, &e0)) return -1;
-    if (is_bf16) {
-        constexpr int XI = 256;
-        dim3 grid((W + XI - 1) / XI, H, (unsigned)(N * D));
+    constexpr int XI = 128;  // wave covers 2*64 x; 4 rows per WG
+    dim3 grid((W + XI - 1) / XI, (H + 3) / 4, (unsigned)(N * D));
+    if (is_bf16)
         hipLaunchKernelGGL((k_conv155_out<cfx_bf16, XI, 28, 3>), grid,
-                           dim3(XI), 0, ctx->stream, (const cfx_bf16*)in,
+                           dim3(256), 0, ctx->stream, (const cfx_bf16*)in,
                            (const cfx_bf16*)wgt, bias, (cfx_bf16*)out, N,
                            D, H, W);
-    } else {
-        constexpr int XI = 128;  // f32 LDS: 5 x 132 x 28 x 4 B
-        dim3 grid((W + XI - 1) / XI, H, (unsigned)(N * D));
+    else
         hipLaunchKernelGGL((k_conv155_out<float, XI, 28, 3>), grid,
-                           dim3(XI), 0, ctx->stream, (const float*)in,
+                           dim3(256), 0, ctx->stream, (const float*)in,
                            (const float*)wgt, bias, (float*)out, N, D, H,
                            W);
-    }
     CFX_CHECK(hipGetLastError());
     double flops = 2.0 * 25.0 * C * K * (double)N * D * H * W;
     if (prof_end(ctx, e0, CFX_K_CONV, flops)) return -1;
