@@ -1746,6 +1746,214 @@ __global__ __launch_bounds__(512, 1) void k_conv3_zring_bf16_pl2(
 }
 
 
+// Wide bf16 ring ("w"): 256 threads / 4 waves, each wave computing TWO
+// y-rows per z so the B (weight) fragment read and the per-z loop/barrier
+// overhead amortize over 108 MFMAs instead of 54. LDS drops the +8
+// padding: rows are stride-32 (64 B) with an XOR swizzle on the 16-byte
+// chunk index (chunk ^= (row>>2) & 3), which spreads the 4-lane
+// same-bank classes of each b128 lane group across 4 chunks — measured
+// conflict-free reasoning in DESIGN.md §10 r2. Ring 65.3 KB + wall
+// 55.3 KB = 120.6 KB (the padded layout at this shape needs 182 KB).
+// Accumulation order per output element is IDENTICAL to _pl.
+template <int C, int K, int TY, int TX>
+__global__ __launch_bounds__(256, 1) void k_conv3_zring_bf16_w(
+    const cfx_bf16* __restrict__ in, const cfx_bf16* __restrict__ wgt,
+    const float* __restrict__ bias, const cfx_bf16* __restrict__ res,
+    cfx_bf16* __restrict__ out, int N, int D, int H, int W, int do_elu) {
+    constexpr int CP = 32;
+    constexpr int RS = 32;               // row stride, elements (64 B)
+    constexpr int SX = TX + 2;
+    constexpr int SY = TY + 2;
+    typedef cfx_bf16 bf16x4 __attribute__((ext_vector_type(4)));
+    static_assert(C <= CP && K <= 32 && TY == 8 && TX == 32, "");
+
+    __shared__ cfx_bf16 ring[3 * SY * SX * RS];
+    __shared__ cfx_bf16 wall[27 * 32 * RS];
+
+    const int n = blockIdx.z;
+    const int y0 = blockIdx.y * TY;
+    const int x0 = blockIdx.x * TX;
+    const int tid = threadIdx.x;
+    const int wave = tid >> 6;           // 0..3 -> y-row pair
+    const int lane = tid & 63;
+
+    // swizzled element offset within a row: chunk16 ^= (row >> 2) & 3
+    auto sw = [](int row, int el) {
+        return row * RS + ((((el >> 3) ^ (row >> 2)) & 3) << 3) + (el & 7);
+    };
+
+    for (int idx = tid; idx < 27 * 32 * CP; idx += 256) {
+        const int c = idx % CP;
+        const int j = (idx / CP) % 32;
+        const int tap = idx / (CP * 32);
+        wall[sw(tap * 32 + j, c)] = wgt[(tap * 32 + j) * 32 + c];
+    }
+
+    const bool xy_interior = y0 >= 1 && y0 + TY + 1 <= H && x0 >= 1 &&
+                             x0 + TX + 1 <= W;
+    constexpr int C4 = CP / 4;
+    constexpr int STOT = SY * SX * C4;
+    constexpr int LV = (STOT + 255) / 256;
+
+    auto plane_load = [&](int P, bf16x4 (&vals)[LV], bool (&keep)[LV]) {
+        const bool zin = P >= 0 && P < D;
+        const bool interior = zin && xy_interior;
+#pragma unroll
+        for (int li = 0; li < LV; ++li) {
+            const int idx = min(tid + li * 256, STOT - 1);
+            const int c4 = idx % C4;
+            const int v = idx / C4;
+            const int gy = y0 + v / SX - 1;
+            const int gx = x0 + v % SX - 1;
+            const bool cok = c4 * 4 < C;
+            const bool ok = cok && zin &&
+                            (interior || (gy >= 0 && gy < H && gx >= 0 &&
+                                          gx < W));
+            keep[li] = ok;
+            vals[li] = *reinterpret_cast<const bf16x4*>(
+                in + ((((long long)n * D + (zin ? P : 0)) * H +
+                       (ok ? gy : 0)) * W + (ok ? gx : 0)) * C +
+                (cok ? c4 * 4 : 0));
+        }
+    };
+    auto plane_store = [&](int P, const bf16x4 (&vals)[LV],
+                           const bool (&keep)[LV]) {
+        const int slot = ((P + 1) % 3 + 3) % 3;
+#pragma unroll
+        for (int li = 0; li < LV; ++li) {
+            const int idx = tid + li * 256;
+            if (idx >= STOT) break;
+            const int c4 = idx % C4;
+            const int v = idx / C4;
+            *reinterpret_cast<bf16x4*>(
+                &ring[sw((slot * SY + v / SX) * SX + v % SX, c4 * 4)]) =
+                keep[li] ? vals[li] : bf16x4{};
+        }
+    };
+
+    {
+        bf16x4 v0[LV], v1[LV];
+        bool k0[LV], k1[LV];
+        plane_load(-1, v0, k0);
+        plane_load(0, v1, k1);
+        plane_store(-1, v0, k0);
+        plane_store(0, v1, k1);
+    }
+    __syncthreads();
+
+    const int ax = lane & 31;
+    const int khalf = (lane >> 5) * 8;
+
+    for (int z = 0; z < D; ++z) {
+        f32x16 acc0 = {};
+        f32x16 acc1 = {};
+        int slots[3];
+#pragma unroll
+        for (int dzi = 0; dzi < 3; ++dzi)
+            slots[dzi] = ((z + dzi) % 3 + 3) % 3;
+        auto addrA = [&](int p, int row2) {  // row2: 0/1 within the pair
+            const int tap = p >> 1, kk = p & 1;
+            const int dzi = tap / 9, tl = tap % 9;
+            const int dy = tl / 3 - 1, dx = tl % 3 - 1;
+            const int r = (slots[dzi] * SY + 1 + 2 * wave + row2 + dy) *
+                              SX + (1 + dx) + ax;
+            return reinterpret_cast<const bf16x8*>(
+                &ring[sw(r, khalf + kk * 16)]);
+        };
+        auto addrB = [&](int p) {
+            const int tap = p >> 1, kk = p & 1;
+            return reinterpret_cast<const bf16x8*>(
+                &wall[sw(tap * 32 + ax, khalf + kk * 16)]);
+        };
+        constexpr int PD = 3;
+
+        bf16x4 vals[LV];
+        bool keep[LV];
+        plane_load(z + 1, vals, keep);
+        {
+            bf16x8 a0buf[PD], a1buf[PD], bbuf[PD];
+#pragma unroll
+            for (int p = 0; p < PD; ++p) {
+                a0buf[p] = *addrA(p, 0);
+                a1buf[p] = *addrA(p, 1);
+                bbuf[p] = *addrB(p);
+            }
+#pragma unroll
+            for (int p = 0; p < 36; ++p) {  // dzi 0,1: planes z-1, z
+                const int si = p % PD;
+                acc0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                    a0buf[si], bbuf[si], acc0, 0, 0, 0);
+                acc1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                    a1buf[si], bbuf[si], acc1, 0, 0, 0);
+                __builtin_amdgcn_sched_barrier(0);
+                if (p + PD < 36) {
+                    a0buf[si] = *addrA(p + PD, 0);
+                    a1buf[si] = *addrA(p + PD, 1);
+                    bbuf[si] = *addrB(p + PD);
+                }
+                __builtin_amdgcn_sched_barrier(0);
+            }
+            plane_store(z + 1, vals, keep);
+            __syncthreads();
+#pragma unroll
+            for (int p = 36; p < 36 + PD; ++p) {
+                a0buf[p % PD] = *addrA(p, 0);
+                a1buf[p % PD] = *addrA(p, 1);
+                bbuf[p % PD] = *addrB(p);
+            }
+#pragma unroll
+            for (int p = 36; p < 54; ++p) {  // dzi 2: plane z + 1
+                const int si = p % PD;
+                acc0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                    a0buf[si], bbuf[si], acc0, 0, 0, 0);
+                acc1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                    a1buf[si], bbuf[si], acc1, 0, 0, 0);
+                __builtin_amdgcn_sched_barrier(0);
+                if (p + PD < 54) {
+                    a0buf[si] = *addrA(p + PD, 0);
+                    a1buf[si] = *addrA(p + PD, 1);
+                    bbuf[si] = *addrB(p + PD);
+                }
+                __builtin_amdgcn_sched_barrier(0);
+            }
+        }
+
+        const int j = lane & 31;
+#pragma unroll
+        for (int row2 = 0; row2 < 2; ++row2) {
+            const int gy = y0 + 2 * wave + row2;
+            if (gy >= H || j >= K) continue;
+            const f32x16& acc = row2 ? acc1 : acc0;
+            const float bj = bias ? bias[j] : 0.f;
+            cfx_bf16 rv[16];
+            if (res) {
+#pragma unroll
+                for (int r = 0; r < 16; ++r) {
+                    const int row =
+                        (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
+                    const int gx = min(x0 + row, W - 1);
+                    rv[r] = res[
+                        ((((long long)n * D + z) * H + gy) * W + gx) * K +
+                        j];
+                }
+            }
+#pragma unroll
+            for (int r = 0; r < 16; ++r) {
+                const int row = (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
+                const int gx = x0 + row;
+                if (gx >= W) continue;
+                long long o =
+                    ((((long long)n * D + z) * H + gy) * W + gx) * K + j;
+                float v = acc[r] + bj;
+                if (res) v += (float)rv[r];
+                if (do_elu) v = v > 0.f ? v : expm1f(v);
+                out[o] = (cfx_bf16)v;
+            }
+        }
+    }
+}
+
+
 // Sliced bf16 ring: the _pl kernel generalized to a c-slice [c0, c0+CL)
 // of a wider channel dimension (runtime stride CS) and a j-tile
 // [j0, j0+32) of a wider K (runtime stride KS). Widths 36 and 48 run as
@@ -2169,7 +2377,13 @@ extern "C" int cfx_conv3_ndhwc_bf16(cfx_ctx* ctx, const void* in,
         const char* e = getenv("CFX_BF16_MODE");  // phase ablation (timing)
         return e ? atoi(e) : 0;
     }();
-    if (use_pl == 3)
+    if (use_pl == 4)
+        hipLaunchKernelGGL((k_conv3_zring_bf16_w<28, 28, 8, 32>), grid,
+                           dim3(256), 0, ctx->stream, (const cfx_bf16*)in,
+                           (const cfx_bf16*)wgt, bias,
+                           (const cfx_bf16*)residual, (cfx_bf16*)out, N, D,
+                           H, W, do_elu);
+    else if (use_pl == 3)
         hipLaunchKernelGGL((k_conv3_zring_bf16_pl2<28, 28, 8, 32>), grid,
                            dim3(512), 0, ctx->stream, (const cfx_bf16*)in,
                            (const cfx_bf16*)wgt, bias,
