@@ -69,11 +69,11 @@ class PyTorchEngine(EngineBase):
                     self.model, idx, bf16=bf)
                 self.fastconv_count += accelerate_conv_in(
                     self.model, idx, bf16=bf)
-                if bf:
-                    # bf16-only: MIOpen's f32 conv_out is already fast
-                    from ..fastconv import accelerate_conv_out
-                    self.fastconv_count += accelerate_conv_out(
-                        self.model, idx, bf16=True)
+                # conv_out kernel beats MIOpen at both dtypes (4.8x bf16,
+                # 1.4x f32 -- profiles/updown_probe_r02.json)
+                from ..fastconv import accelerate_conv_out
+                self.fastconv_count += accelerate_conv_out(
+                    self.model, idx, bf16=bf)
         torch.backends.cudnn.benchmark = True
         self.pre_process = getattr(net_source, 'pre_process', None)
         self.post_process = getattr(net_source, 'post_process', None)
