@@ -59,10 +59,14 @@ def parse_args():
                    default='float32')
     p.add_argument('--no-cpu-baseline', action='store_true')
     p.add_argument('--cpu-baseline-seconds', type=float, default=15.0)
+    p.add_argument('--cpu-baseline-full', action='store_true',
+                   help='run the CPU baseline over ALL patches (no '
+                        'extrapolation; ~5-6 min)')
     return p.parse_args()
 
 
-def cpu_baseline(args, chunk_u8: np.ndarray, budget_s: float) -> dict:
+def cpu_baseline(args, chunk_u8: np.ndarray, budget_s: float,
+                 full: bool = False) -> dict:
     """The oracle (CPU restatement of the reference path, torch-CPU conv,
     batch 1 like the reference pytorch engine) timed on a bounded sample of
     the same workload; voxels/sec extrapolated from the patch fraction."""
@@ -101,7 +105,7 @@ def cpu_baseline(args, chunk_u8: np.ndarray, budget_s: float) -> dict:
         res = forward(np.ascontiguousarray(buf)) * mask
         blend_into(out, (0, 0, 0), res[0], (oz, oy, ox))
         done += 1
-        if time.perf_counter() - t0 > budget_s and done >= 2:
+        if not full and time.perf_counter() - t0 > budget_s and done >= 2:
             break
     elapsed = time.perf_counter() - t0
     vox = chunk_u8.size * (done / n_total)
@@ -112,22 +116,26 @@ def cpu_baseline(args, chunk_u8: np.ndarray, budget_s: float) -> dict:
         'kind': 'port',
         'sample': f'{done}/{n_total} patches of the 512^3 config-2 workload,'
                   f' oracle numpy tiler/blend + torch-CPU '
-                  f'{args.engine} at batch 1, {elapsed:.1f}s',
+                  f'{args.engine} at batch 1, {elapsed:.1f}s'
+                  + (' (FULL RUN, no extrapolation)' if full else ''),
     }
 
 
-def read_pmc_traffic() -> float:
-    """Per-launch HBM bytes of the blend kernel from a committed rocprofv3
-    PMC measurement (profiles/pmc_traffic.json), or None."""
+def read_pmc_traffic():
+    """(bytes, provenance) of the blend kernel's per-launch HBM traffic
+    from a committed rocprofv3 PMC measurement (profiles/pmc_traffic.json;
+    collected offline — rocprofv3 PMC passes cannot run inside the timed
+    bench), or (None, None)."""
     path = os.path.join(REPO, 'profiles', 'pmc_traffic.json')
     if not os.path.exists(path):
-        return None
+        return None, None
     try:
         with open(path) as f:
             d = json.load(f)
-        return float(d['blend_bytes_per_launch'])
+        return (float(d['blend_bytes_per_launch']),
+                d.get('provenance', 'profiles/pmc_traffic.json'))
     except Exception:
-        return None
+        return None, None
 
 
 def main():
@@ -198,7 +206,7 @@ def main():
     cfx.profile_enable(True)
     # the fastconv kernels run on their own context; profile it too
     conv_ctx = None
-    if args.engine == 'rsunet' and args.dtype == 'float32':
+    if args.engine == 'rsunet':
         try:
             from chunkflow_amd.fastconv import _CTX
             conv_ctx = _CTX.get(local_rank)
@@ -235,15 +243,22 @@ def main():
     voxels = float(cz) * cy * cx * world * args.steps
     value = voxels / elapsed
 
-    traffic = read_pmc_traffic()
+    traffic, traffic_src = read_pmc_traffic()
     conv_roofline = None
     if conv and conv['count'] > 0 and conv['total_ms'] > 0:
         tflops = conv['bytes'] / (conv['total_ms'] / 1e3) / 1e12
+        # aggregate useful FLOP rate over ALL hand conv kernels (rings +
+        # up-convs + (1,5,5) convs); peak = the dtype's dense MFMA peak
+        # (MI355X_MICROARCH.md; bf16 includes HBM-bound stream kernels in
+        # the mix, so frac understates the rings alone -- per-kernel
+        # numbers live in profiles/updown_probe_r02.json)
+        peak = 157.3 if args.dtype == 'float32' else 2500.0
         conv_roofline = {
             'bound': 'mfma', 'achieved_tflops': tflops,
-            'peak_tflops': 157.3,  # f32 MFMA peak (MI355X_MICROARCH.md)
-            'frac': tflops / 157.3, 'launches': conv['count'],
-            'kernel': 'k_conv3_zring (fastconv, widths 28/36)',
+            'peak_tflops': peak,
+            'frac': tflops / peak, 'launches': conv['count'],
+            'kernel': 'hand conv kernels (zring/upconv/conv155), '
+                      + args.dtype,
         }
     roofline = None
     if blend['count'] > 0 and blend['total_ms'] > 0:
@@ -255,6 +270,7 @@ def main():
             'unit': 'GB/s',
             'frac': achieved / HBM_PEAK_BYTES_PER_S,
             'traffic': traffic,
+            'traffic_source': traffic_src,
             'kernel': 'k_blend',
             'launches': blend['count'],
             'algorithmic_bytes_per_launch': blend['bytes'] / blend['count'],
@@ -264,7 +280,9 @@ def main():
 
     cpu = None
     if not args.no_cpu_baseline and world == 1:
-        cpu = cpu_baseline(args, host_chunk.array, args.cpu_baseline_seconds)
+        cpu = cpu_baseline(args, host_chunk.array,
+                           args.cpu_baseline_seconds,
+                           full=args.cpu_baseline_full)
 
     result = {
         'metric': 'output_voxels_per_sec',
