@@ -230,8 +230,10 @@ def main():
     blend = cfx.profile_get('blend')
     # the hand MFMA conv kernels (fastconv) publish FLOPs in the bytes slot
     conv = None
+    conv_stream = None
     if conv_ctx is not None:
         conv = conv_ctx.profile_get('conv')
+        conv_stream = conv_ctx.profile_get('conv_stream')
         conv_ctx.profile_enable(False)
     cfx.profile_enable(False)
 
@@ -247,19 +249,25 @@ def main():
     conv_roofline = None
     if conv and conv['count'] > 0 and conv['total_ms'] > 0:
         tflops = conv['bytes'] / (conv['total_ms'] / 1e3) / 1e12
-        # aggregate useful FLOP rate over ALL hand conv kernels (rings +
-        # up-convs + (1,5,5) convs); peak = the dtype's dense MFMA peak
-        # (MI355X_MICROARCH.md; bf16 includes HBM-bound stream kernels in
-        # the mix, so frac understates the rings alone -- per-kernel
-        # numbers live in profiles/updown_probe_r02.json)
+        # useful FLOP rate of the MFMA ring convs alone (the HBM-stream
+        # convs report separately below); peak = the dtype's dense MFMA
+        # peak (MI355X_MICROARCH.md)
         peak = 157.3 if args.dtype == 'float32' else 2500.0
         conv_roofline = {
             'bound': 'mfma', 'achieved_tflops': tflops,
             'peak_tflops': peak,
             'frac': tflops / peak, 'launches': conv['count'],
-            'kernel': 'hand conv kernels (zring/upconv/conv155), '
-                      + args.dtype,
+            'kernel': 'MFMA z-ring convs (fastconv), ' + args.dtype,
         }
+        if conv_stream and conv_stream['count'] > 0 \
+                and conv_stream['total_ms'] > 0:
+            conv_roofline['stream_convs'] = {
+                'bound': 'hbm',
+                'achieved_tflops': conv_stream['bytes']
+                / (conv_stream['total_ms'] / 1e3) / 1e12,
+                'launches': conv_stream['count'],
+                'kernel': 'upconv/conv155 HBM-stream convs',
+            }
     roofline = None
     if blend['count'] > 0 and blend['total_ms'] > 0:
         achieved = blend['bytes'] / (blend['total_ms'] / 1e3)

@@ -458,7 +458,7 @@ extern "C" int cfx_conv155_out(cfx_ctx* ctx, const void* in,
                            W);
     CFX_CHECK(hipGetLastError());
     double flops = 2.0 * 25.0 * C * K * (double)N * D * H * W;
-    if (prof_end(ctx, e0, CFX_K_CONV, flops)) return -1;
+    if (prof_end(ctx, e0, CFX_K_CONV_STREAM, flops)) return -1;
     return 0;
 }
 
@@ -495,7 +495,7 @@ extern "C" int cfx_conv155_c1(cfx_ctx* ctx, const void* in, const void* wgt,
                            W, K);
     CFX_CHECK(hipGetLastError());
     double flops = 2.0 * 25.0 * K * (double)N * D * H * W;
-    if (prof_end(ctx, e0, CFX_K_CONV, flops)) return -1;
+    if (prof_end(ctx, e0, CFX_K_CONV_STREAM, flops)) return -1;
     return 0;
 }
 
@@ -523,7 +523,7 @@ extern "C" int cfx_upconv_2x2(cfx_ctx* ctx, const void* in, const void* wgt,
                            bias, (float*)out, N, D, H, W, C, K);
     CFX_CHECK(hipGetLastError());
     double flops = 2.0 * 4.0 * C * K * (double)N * D * H * W;
-    if (prof_end(ctx, e0, CFX_K_CONV, flops)) return -1;
+    if (prof_end(ctx, e0, CFX_K_CONV_STREAM, flops)) return -1;
     return 0;
 }
 
@@ -556,6 +556,6 @@ extern "C" int cfx_downconv_2x2(cfx_ctx* ctx, const void* in,
                            bias, (float*)out, N, D, H, W, C, K);
     CFX_CHECK(hipGetLastError());
     double flops = 2.0 * 4.0 * C * K * (double)N * D * (H / 2) * (W / 2);
-    if (prof_end(ctx, e0, CFX_K_CONV, flops)) return -1;
+    if (prof_end(ctx, e0, CFX_K_CONV_STREAM, flops)) return -1;
     return 0;
 }

@@ -35,6 +35,7 @@ SYMBOLS = [
 KERNEL_IDS = {
     'blend': 0, 'extract': 1, 'normalize': 2, 'cast': 3, 'reciprocal': 4,
     'maskmul': 5, 'crop': 6, 'max': 7, 'myelin': 8, 'cc': 9, 'conv': 10,
+    'conv_stream': 11,
 }
 
 _lib = None

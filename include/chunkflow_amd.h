@@ -196,7 +196,8 @@ enum cfx_kernel_id {
     CFX_K_MYELIN = 8,
     CFX_K_CC = 9,
     CFX_K_CONV = 10,
-    CFX_K_COUNT = 11
+    CFX_K_CONV_STREAM = 11,  /* up/down-sample + (1,5,5) stream convs */
+    CFX_K_COUNT = 12
 };
 int cfx_profile_enable(cfx_ctx* ctx, int enable);
 int cfx_profile_reset(cfx_ctx* ctx);
