@@ -933,3 +933,23 @@ def test_conv155_out_vs_torch(bf16):
     got = CfxConvOut155(conv, 0, bf16=bf16).cuda()(xc).float()
     tol = 0.05 if bf16 else 1e-4
     torch.testing.assert_close(got, want, rtol=tol, atol=tol)
+
+
+def test_blend_reference_order_gpu(golden, monkeypatch):
+    """The per-patch reference-order blend mode on GPU matches the grouped
+    default at 1e-6 and the reference golden at 1e-5."""
+    from chunkflow_amd.chunk import Chunk
+    _, arrays = golden
+
+    def run():
+        inf = _hip_inferencer(framework='identity', num_output_channels=3,
+                              batch_size=3, mask_output_chunk=True)
+        return np.asarray(
+            inf(Chunk(arrays['e2e_input_u8'].copy())).numpy().array)
+
+    base = run()
+    monkeypatch.setenv('CFX_BLEND_REFORDER', '1')
+    ref_order = run()
+    np.testing.assert_allclose(ref_order, base, rtol=1e-6, atol=1e-7)
+    np.testing.assert_allclose(ref_order, arrays['e2e_identity_out'],
+                               rtol=1e-5, atol=1e-6)

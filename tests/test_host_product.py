@@ -497,3 +497,28 @@ def test_rsunet_64x256x256_cpu_plumbing_golden(golden, golden_dir):
         out.ravel()[arrays['rsunet_64x256x256_sample_idx']],
         arrays['rsunet_64x256x256_sample_val'], rtol=1e-5, atol=1e-5)
     assert abs(out.astype(np.float64).sum() - case['sum_f64']) < 1.0
+
+
+def test_blend_reference_order_switch(golden, monkeypatch):
+    """CFX_BLEND_REFORDER=1 blends one patch per call in the reference's
+    exact order; result must match the default grouped blending at 1e-6
+    and the reference golden exactly as the default does."""
+    if GPU:
+        pytest.skip('CPU plumbing path')
+    from chunkflow_amd.inferencer import Inferencer
+    _, arrays = golden
+
+    def run():
+        inf = Inferencer(None, None, (10, 32, 32),
+                         output_patch_overlap=(2, 8, 8),
+                         framework='identity', num_output_channels=3,
+                         batch_size=3, mask_output_chunk=True)
+        return np.asarray(
+            inf(Chunk(arrays['e2e_input_u8'].copy())).numpy().array)
+
+    base = run()
+    monkeypatch.setenv('CFX_BLEND_REFORDER', '1')
+    ref_order = run()
+    np.testing.assert_allclose(ref_order, base, rtol=1e-6, atol=1e-7)
+    np.testing.assert_allclose(ref_order, arrays['e2e_identity_out'],
+                               rtol=1e-6, atol=1e-7)
