@@ -953,3 +953,35 @@ def test_blend_reference_order_gpu(golden, monkeypatch):
     np.testing.assert_allclose(ref_order, base, rtol=1e-6, atol=1e-7)
     np.testing.assert_allclose(ref_order, arrays['e2e_identity_out'],
                                rtol=1e-5, atol=1e-6)
+
+
+def test_cli_worker_stitch_gpu_world1(tmp_path, monkeypatch):
+    """The config-3 CLI worker pipeline (generate-tasks sharding + stitch
+    operator) on GPU at world 1: inference runs the HIP path per task and
+    the stitch places each task's sub-volume correctly."""
+    from click.testing import CliRunner
+    from chunkflow_amd.flow import main
+    for k in ('RANK', 'WORLD_SIZE', 'MASTER_ADDR', 'MASTER_PORT'):
+        monkeypatch.delenv(k, raising=False)
+    out = tmp_path / 'vol.npy'
+    r = CliRunner().invoke(main, [
+        'generate-tasks', '-c', '16', '40', '48',
+        '--roi-size', '16', '40', '96',
+        'create-chunk',
+        'inference', '-s', '8', '16', '16',
+        '--output-patch-overlap', '2', '4', '4',
+        '--framework', 'identity', '--batch-size', '4',
+        '--num-output-channels', '3', '--mask-output-chunk',
+        'stitch', '-f', str(out)], catch_exceptions=False)
+    assert r.exit_code == 0, r.output
+    vol = np.load(out)
+    assert vol.shape == (3, 16, 40, 96)
+    # identity inference of the sin chunk == chunk/255 at every position;
+    # create-chunk from bbox uses the same sin for both tasks
+    from chunkflow_amd.chunk import Chunk
+    sin = np.asarray(Chunk.create(size=(16, 40, 48), dtype='uint8',
+                                  pattern='sin').array)
+    ref = sin.astype(np.float32) / 255.0
+    for t in range(2):
+        np.testing.assert_allclose(vol[0, :, :, t * 48:(t + 1) * 48], ref,
+                                   rtol=1e-5, atol=1e-6)
