@@ -2570,12 +2570,15 @@ extern "C" int cfx_conv3_ndhwc_bf16(cfx_ctx* ctx, const void* in,
         return -1;
     }
     dim3 grid((W + 31) / 32, (H + 7) / 8, (unsigned)N);
-    // pipelined default: with batched staging loads the register-held
-    // overlap pays off — 500.2 TF vs 458.4 plain (an earlier "neutral"
-    // verdict came from per-load vmcnt drains masking the difference)
+    // default 6 = pipelined + alternating accumulators (539.9 TF vs
+    // 528.6 for _pl; the r2 ladder tried 4 structural variants — pl2
+    // prefetch-double-buffer neutral, wide-swizzled 280 (1 wave/SIMD),
+    // alternation +2% — the ring is pinned at ~530-540 TF by something
+    // other than staging, LDS conflicts, or the same-acc issue cliff;
+    // DESIGN.md §10-r2 item 6)
     static const int use_pl = [] {
         const char* e = getenv("CFX_ZRING_PL");
-        return e ? atoi(e) : 1;
+        return e ? atoi(e) : 6;
     }();
     hipEvent_t e0;
     if (prof_begin(ctx, &e0)) return -1;
