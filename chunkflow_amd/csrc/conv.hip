@@ -2257,7 +2257,8 @@ __global__ __launch_bounds__(512, 1) void k_conv3_zring_bf16_s(
     const int khalf = (lane >> 5) * 8;
 
     for (int z = 0; z < D; ++z) {
-        f32x16 acc = {};
+        f32x16 acc = {};   // even pairs; accB odd (same-acc-cliff dodge,
+        f32x16 accB = {};  // see k_conv3_zring_bf16_a)
         const cfx_bf16* planes[3];
 #pragma unroll
         for (int dzi = 0; dzi < 3; ++dzi)
@@ -2292,8 +2293,12 @@ __global__ __launch_bounds__(512, 1) void k_conv3_zring_bf16_s(
 #pragma unroll
             for (int p = 0; p < NP1; ++p) {
                 const int si = p % PD;
-                acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-                    abuf[si], bbuf[si], acc, 0, 0, 0);
+                if (p & 1)
+                    accB = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                        abuf[si], bbuf[si], accB, 0, 0, 0);
+                else
+                    acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                        abuf[si], bbuf[si], acc, 0, 0, 0);
                 __builtin_amdgcn_sched_barrier(0);
                 if (p + PD < NP1) {
                     abuf[si] = *addrA(p + PD);
@@ -2311,8 +2316,12 @@ __global__ __launch_bounds__(512, 1) void k_conv3_zring_bf16_s(
 #pragma unroll
             for (int p = NP1; p < NP2; ++p) {
                 const int si = p % PD;
-                acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-                    abuf[si], bbuf[si], acc, 0, 0, 0);
+                if (p & 1)
+                    accB = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                        abuf[si], bbuf[si], accB, 0, 0, 0);
+                else
+                    acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                        abuf[si], bbuf[si], acc, 0, 0, 0);
                 __builtin_amdgcn_sched_barrier(0);
                 if (p + PD < NP2) {
                     abuf[si] = *addrA(p + PD);
@@ -2322,6 +2331,7 @@ __global__ __launch_bounds__(512, 1) void k_conv3_zring_bf16_s(
             }
         }
 
+        acc = acc + accB;
         const int gy = y0 + wave;
         const int j = lane & 31;
         if (gy < H && j0 + j < Ktot) {

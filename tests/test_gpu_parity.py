@@ -985,3 +985,66 @@ def test_cli_worker_stitch_gpu_world1(tmp_path, monkeypatch):
     for t in range(2):
         np.testing.assert_allclose(vol[0, :, :, t * 48:(t + 1) * 48], ref,
                                    rtol=1e-5, atol=1e-6)
+
+
+def test_upconv_fuzz_shapes():
+    """Random-shape fuzz of the up-conv kernel vs torch, f32 (odd sizes,
+    varied channels)."""
+    from chunkflow_amd.fastconv import CfxUpConv3d
+    rng = np.random.RandomState(11)
+    for _ in range(6):
+        C = int(rng.choice([3, 17, 36, 48, 64]))
+        K = int(rng.choice([1, 5, 28, 33, 64]))
+        n = int(rng.randint(1, 3))
+        d, h, w = (int(rng.randint(1, 5)), int(rng.randint(1, 20)),
+                   int(rng.randint(1, 70)))
+        conv = torch.nn.ConvTranspose3d(C, K, (1, 2, 2),
+                                        stride=(1, 2, 2)).cuda()
+        x = torch.randn(n, C, d, h, w, device='cuda') \
+            .contiguous(memory_format=torch.channels_last_3d)
+        want = conv(x)
+        got = CfxUpConv3d(conv, 0, bf16=False).cuda()(x)
+        torch.testing.assert_close(got, want, rtol=1e-4, atol=1e-4)
+
+
+def test_conv155_fuzz_shapes():
+    """Random-shape fuzz of the (1,5,5) conv_in/conv_out kernels, f32."""
+    from chunkflow_amd.fastconv import CfxConvIn155, CfxConvOut155
+    rng = np.random.RandomState(12)
+    for _ in range(5):
+        K = int(rng.choice([1, 7, 28, 32]))
+        n, d = int(rng.randint(1, 3)), int(rng.randint(1, 4))
+        h, w = int(rng.randint(1, 30)), int(rng.randint(1, 300))
+        conv = torch.nn.Conv3d(1, K, (1, 5, 5), padding=(0, 2, 2)).cuda()
+        x = torch.randn(n, 1, d, h, w, device='cuda') \
+            .contiguous(memory_format=torch.channels_last_3d)
+        torch.testing.assert_close(
+            CfxConvIn155(conv, 0).cuda()(x), conv(x),
+            rtol=1e-4, atol=1e-4)
+    for _ in range(3):
+        n, d = int(rng.randint(1, 3)), int(rng.randint(1, 4))
+        h, w = int(rng.randint(1, 30)), int(rng.randint(1, 300))
+        conv = torch.nn.Conv3d(28, 3, (1, 5, 5), padding=(0, 2, 2)).cuda()
+        x = torch.randn(n, 28, d, h, w, device='cuda') \
+            .contiguous(memory_format=torch.channels_last_3d)
+        torch.testing.assert_close(
+            CfxConvOut155(conv, 0).cuda()(x), conv(x),
+            rtol=1e-4, atol=1e-4)
+
+
+def test_sliced_ring_fuzz_shapes():
+    """Random-shape fuzz of the sliced bf16 ring (C=36/48) vs torch f32
+    at bf16 tolerance."""
+    from chunkflow_amd.fastconv import CfxConv3dBF16
+    rng = np.random.RandomState(13)
+    for _ in range(4):
+        C = int(rng.choice([36, 48]))
+        n, d = int(rng.randint(1, 3)), int(rng.randint(1, 6))
+        h, w = int(rng.randint(1, 40)), int(rng.randint(1, 40))
+        conv = torch.nn.Conv3d(C, C, 3, padding=1).cuda()
+        x = (torch.randn(n, C, d, h, w, device='cuda') * 0.3)
+        want = conv(x).float()
+        got = CfxConv3dBF16(conv, 0).cuda()._run(
+            x.to(torch.bfloat16)
+            .contiguous(memory_format=torch.channels_last_3d)).float()
+        torch.testing.assert_close(got, want, rtol=0.1, atol=0.05)
