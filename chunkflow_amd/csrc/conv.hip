@@ -2160,6 +2160,226 @@ __global__ __launch_bounds__(256, 1) void k_conv3_zring_bf16_w(
 }
 
 
+// 16x16x32 ring ("t"): same staging/ring/wall as _pl but the mainloop
+// uses v_mfma_f32_16x16x32_bf16, whose K=32 covers ALL channels in one
+// op — ONE b128 LDS read per MFMA instead of two. The r2 phase ablation
+//  showed the 32x32x16 loop's 2 reads/MFMA saturate the
+// LDS array at exactly the MFMA-pipe time (3456 cyc each per z per CU),
+// so the 32-wide ring is LDS-read co-bound; 16x16x32 trades a 20% lower
+// MFMA peak rate (~5 vs ~8 cyc/CU per-flop-adjusted) for halving the LDS
+// pressure. Per wave: 2 M-tiles (x) x 2 N-tiles (j) of f32x4, which also
+// alternates accumulators between consecutive MFMAs for free.
+// Fragment layouts probe-verified (tools/mfma_layout_probe.py, 16x16x32):
+// A/B: row/col = lane&15, k = (lane>>4)*8 + e; C/D: col = lane&15,
+// row = (lane>>4)*4 + reg.
+template <int C, int K, int TY, int TX>
+__global__ __launch_bounds__(512, 1) void k_conv3_zring_bf16_t(
+    const cfx_bf16* __restrict__ in, const cfx_bf16* __restrict__ wgt,
+    const float* __restrict__ bias, const cfx_bf16* __restrict__ res,
+    cfx_bf16* __restrict__ out, int N, int D, int H, int W, int do_elu) {
+    constexpr int CP = 32;
+    constexpr int PCB = CP + 8;
+    constexpr int SX = TX + 2;
+    constexpr int SY = TY + 2;
+    typedef cfx_bf16 bf16x4 __attribute__((ext_vector_type(4)));
+    typedef float f32x4 __attribute__((ext_vector_type(4)));
+    static_assert(C <= CP && K <= 32 && TX == 32, "");
+
+    __shared__ cfx_bf16 ring[3 * SY * SX * PCB];
+    __shared__ cfx_bf16 wall[27 * 32 * PCB];
+
+    const int n = blockIdx.z;
+    const int y0 = blockIdx.y * TY;
+    const int x0 = blockIdx.x * TX;
+    const int tid = threadIdx.x;
+    const int wave = tid >> 6;
+    const int lane = tid & 63;
+
+    for (int idx = tid; idx < 27 * 32 * CP; idx += 512) {
+        const int c = idx % CP;
+        const int j = (idx / CP) % 32;
+        const int tap = idx / (CP * 32);
+        wall[(tap * 32 + j) * PCB + c] = wgt[(tap * 32 + j) * 32 + c];
+    }
+
+    const bool xy_interior = y0 >= 1 && y0 + TY + 1 <= H && x0 >= 1 &&
+                             x0 + TX + 1 <= W;
+    constexpr int C4 = CP / 4;
+    constexpr int STOT = SY * SX * C4;
+    constexpr int LV = (STOT + 511) / 512;
+
+    auto plane_load = [&](int P, bf16x4 (&vals)[LV], bool (&keep)[LV]) {
+        const bool zin = P >= 0 && P < D;
+        const bool interior = zin && xy_interior;
+#pragma unroll
+        for (int li = 0; li < LV; ++li) {
+            const int idx = min(tid + li * 512, STOT - 1);
+            const int c4 = idx % C4;
+            const int v = idx / C4;
+            const int gy = y0 + v / SX - 1;
+            const int gx = x0 + v % SX - 1;
+            const bool cok = c4 * 4 < C;
+            const bool ok = cok && zin &&
+                            (interior || (gy >= 0 && gy < H && gx >= 0 &&
+                                          gx < W));
+            keep[li] = ok;
+            vals[li] = *reinterpret_cast<const bf16x4*>(
+                in + ((((long long)n * D + (zin ? P : 0)) * H +
+                       (ok ? gy : 0)) * W + (ok ? gx : 0)) * C +
+                (cok ? c4 * 4 : 0));
+        }
+    };
+    auto plane_store = [&](int P, const bf16x4 (&vals)[LV],
+                           const bool (&keep)[LV]) {
+        const int slot = ((P + 1) % 3 + 3) % 3;
+#pragma unroll
+        for (int li = 0; li < LV; ++li) {
+            const int idx = tid + li * 512;
+            if (idx >= STOT) break;
+            const int c4 = idx % C4;
+            const int v = idx / C4;
+            *reinterpret_cast<bf16x4*>(
+                &ring[((slot * SY + v / SX) * SX + v % SX) * PCB +
+                      c4 * 4]) = keep[li] ? vals[li] : bf16x4{};
+        }
+    };
+
+    {
+        bf16x4 v0[LV], v1[LV];
+        bool k0[LV], k1[LV];
+        plane_load(-1, v0, k0);
+        plane_load(0, v1, k1);
+        plane_store(-1, v0, k0);
+        plane_store(0, v1, k1);
+    }
+    __syncthreads();
+
+    const int li16 = lane & 15;          // fragment row/col
+    const int kc = (lane >> 4) * 8;      // fragment k-chunk (channels)
+
+    for (int z = 0; z < D; ++z) {
+        f32x4 acc00 = {}, acc01 = {}, acc10 = {}, acc11 = {};
+        const cfx_bf16* planes[3];
+#pragma unroll
+        for (int dzi = 0; dzi < 3; ++dzi)
+            planes[dzi] = &ring[(((z + dzi) % 3 + 3) % 3) * SY * SX * PCB];
+        auto addrA = [&](int tap, int m) {
+            const int dzi = tap / 9, tl = tap % 9;
+            const int dy = tl / 3 - 1, dx = tl % 3 - 1;
+            return reinterpret_cast<const bf16x8*>(
+                &planes[dzi][((1 + wave + dy) * SX + (1 + dx) + m * 16 +
+                              li16) * PCB + kc]);
+        };
+        auto addrB = [&](int tap, int nt) {
+            return reinterpret_cast<const bf16x8*>(
+                &wall[(tap * 32 + nt * 16 + li16) * PCB + kc]);
+        };
+        constexpr int PD = 3;
+
+        bf16x4 vals[LV];
+        bool keep[LV];
+        plane_load(z + 1, vals, keep);
+        {
+            bf16x8 a0[PD], a1[PD], b0[PD], b1[PD];
+#pragma unroll
+            for (int p = 0; p < PD; ++p) {
+                a0[p] = *addrA(p, 0);
+                a1[p] = *addrA(p, 1);
+                b0[p] = *addrB(p, 0);
+                b1[p] = *addrB(p, 1);
+            }
+#pragma unroll
+            for (int tap = 0; tap < 18; ++tap) {  // dzi 0,1
+                const int si = tap % PD;
+                acc00 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    a0[si], b0[si], acc00, 0, 0, 0);
+                acc01 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    a0[si], b1[si], acc01, 0, 0, 0);
+                acc10 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    a1[si], b0[si], acc10, 0, 0, 0);
+                acc11 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    a1[si], b1[si], acc11, 0, 0, 0);
+                __builtin_amdgcn_sched_barrier(0);
+                if (tap + PD < 18) {
+                    a0[si] = *addrA(tap + PD, 0);
+                    a1[si] = *addrA(tap + PD, 1);
+                    b0[si] = *addrB(tap + PD, 0);
+                    b1[si] = *addrB(tap + PD, 1);
+                }
+                __builtin_amdgcn_sched_barrier(0);
+            }
+            plane_store(z + 1, vals, keep);
+            __syncthreads();
+#pragma unroll
+            for (int tap = 18; tap < 18 + PD; ++tap) {
+                a0[tap % PD] = *addrA(tap, 0);
+                a1[tap % PD] = *addrA(tap, 1);
+                b0[tap % PD] = *addrB(tap, 0);
+                b1[tap % PD] = *addrB(tap, 1);
+            }
+#pragma unroll
+            for (int tap = 18; tap < 27; ++tap) {  // dzi 2
+                const int si = tap % PD;
+                acc00 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    a0[si], b0[si], acc00, 0, 0, 0);
+                acc01 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    a0[si], b1[si], acc01, 0, 0, 0);
+                acc10 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    a1[si], b0[si], acc10, 0, 0, 0);
+                acc11 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    a1[si], b1[si], acc11, 0, 0, 0);
+                __builtin_amdgcn_sched_barrier(0);
+                if (tap + PD < 27) {
+                    a0[si] = *addrA(tap + PD, 0);
+                    a1[si] = *addrA(tap + PD, 1);
+                    b0[si] = *addrB(tap + PD, 0);
+                    b1[si] = *addrB(tap + PD, 1);
+                }
+                __builtin_amdgcn_sched_barrier(0);
+            }
+        }
+
+        const int gy = y0 + wave;
+        if (gy < H) {
+            const long long rowbase =
+                (((long long)n * D + z) * H + gy) * (long long)W * K;
+#pragma unroll
+            for (int m = 0; m < 2; ++m) {
+#pragma unroll
+                for (int nt = 0; nt < 2; ++nt) {
+                    const int j = nt * 16 + li16;
+                    if (j >= K) continue;
+                    const f32x4& acc = m == 0 ? (nt == 0 ? acc00 : acc01)
+                                             : (nt == 0 ? acc10 : acc11);
+                    const float bj = bias ? bias[j] : 0.f;
+                    cfx_bf16 rv[4];
+                    if (res) {
+#pragma unroll
+                        for (int r = 0; r < 4; ++r) {
+                            const int row =
+                                m * 16 + (lane >> 4) * 4 + r;
+                            const int gx = min(x0 + row, W - 1);
+                            rv[r] = res[rowbase + (long long)gx * K + j];
+                        }
+                    }
+#pragma unroll
+                    for (int r = 0; r < 4; ++r) {
+                        const int row = m * 16 + (lane >> 4) * 4 + r;
+                        const int gx = x0 + row;
+                        if (gx >= W) continue;
+                        float v = acc[r] + bj;
+                        if (res) v += (float)rv[r];
+                        if (do_elu) v = v > 0.f ? v : expm1f(v);
+                        out[rowbase + (long long)gx * K + j] =
+                            (cfx_bf16)v;
+                    }
+                }
+            }
+        }
+    }
+}
+
+
 // Sliced bf16 ring: the _pl kernel generalized to a c-slice [c0, c0+CL)
 // of a wider channel dimension (runtime stride CS) and a j-tile
 // [j0, j0+32) of a wider K (runtime stride KS). Widths 36 and 48 run as
@@ -2596,7 +2816,13 @@ extern "C" int cfx_conv3_ndhwc_bf16(cfx_ctx* ctx, const void* in,
         const char* e = getenv("CFX_BF16_MODE");  // phase ablation (timing)
         return e ? atoi(e) : 0;
     }();
-    if (use_pl == 6)
+    if (use_pl == 7)
+        hipLaunchKernelGGL((k_conv3_zring_bf16_t<28, 28, 8, 32>), grid,
+                           dim3(512), 0, ctx->stream, (const cfx_bf16*)in,
+                           (const cfx_bf16*)wgt, bias,
+                           (const cfx_bf16*)residual, (cfx_bf16*)out, N, D,
+                           H, W, do_elu);
+    else if (use_pl == 6)
         hipLaunchKernelGGL((k_conv3_zring_bf16_a<28, 28, 8, 32>), grid,
                            dim3(512), 0, ctx->stream, (const cfx_bf16*)in,
                            (const cfx_bf16*)wgt, bias,
