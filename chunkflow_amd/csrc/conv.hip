@@ -1547,7 +1547,8 @@ __global__ __launch_bounds__(512, 1) void k_conv3_zring_bf16_pl(
 
 // accumulator-alternating variant (CFX_ZRING_PL=6): targets the
 // same-accumulator issue cliff (see comment in the z loop)
-template <int C, int K, int TY, int TX>
+template <int C, int K, int TY, int TX, int PDX = 4,
+          int SB = 1, int EPI = 1>
 __global__ __launch_bounds__(512, 1) void k_conv3_zring_bf16_a(
     const cfx_bf16* __restrict__ in, const cfx_bf16* __restrict__ wgt,
     const float* __restrict__ bias, const cfx_bf16* __restrict__ res,
@@ -1664,7 +1665,7 @@ __global__ __launch_bounds__(512, 1) void k_conv3_zring_bf16_a(
             return reinterpret_cast<const bf16x8*>(
                 &wall[(tap * 32 + ax) * PCB + khalf + kk * 16]);
         };
-        constexpr int PD = 4;
+        constexpr int PD = PDX;
 
         bf16x4 vals[LV];
         bool keep[LV];
@@ -1685,12 +1686,12 @@ __global__ __launch_bounds__(512, 1) void k_conv3_zring_bf16_a(
                 else
                     accA = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
                         abuf[si], bbuf[si], accA, 0, 0, 0);
-                __builtin_amdgcn_sched_barrier(0);
+                if (SB) __builtin_amdgcn_sched_barrier(0);
                 if (p + PD < 36) {
                     abuf[si] = *addrA(p + PD);
                     bbuf[si] = *addrB(p + PD);
                 }
-                __builtin_amdgcn_sched_barrier(0);
+                if (SB) __builtin_amdgcn_sched_barrier(0);
             }
             plane_store(z + 1, vals, keep);
             __syncthreads();
@@ -1708,16 +1709,21 @@ __global__ __launch_bounds__(512, 1) void k_conv3_zring_bf16_a(
                 else
                     accA = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
                         abuf[si], bbuf[si], accA, 0, 0, 0);
-                __builtin_amdgcn_sched_barrier(0);
+                if (SB) __builtin_amdgcn_sched_barrier(0);
                 if (p + PD < 54) {
                     abuf[si] = *addrA(p + PD);
                     bbuf[si] = *addrB(p + PD);
                 }
-                __builtin_amdgcn_sched_barrier(0);
+                if (SB) __builtin_amdgcn_sched_barrier(0);
             }
         }
 
         const f32x16 acc = accA + accB;
+        if (!EPI) {  // timing ablation: keep acc alive, skip the real
+                     // epilogue (WRONG results; CFX_ZRING_PL=10 only)
+            out[tid] = (cfx_bf16)(acc[0] + acc[15]);
+            continue;
+        }
         const int gy = y0 + wave;
         const int j = lane & 31;
         if (gy < H && j < K) {
@@ -2816,7 +2822,25 @@ extern "C" int cfx_conv3_ndhwc_bf16(cfx_ctx* ctx, const void* in,
         const char* e = getenv("CFX_BF16_MODE");  // phase ablation (timing)
         return e ? atoi(e) : 0;
     }();
-    if (use_pl == 7)
+    if (use_pl == 8)       // ablation: deeper LDS prefetch
+        hipLaunchKernelGGL((k_conv3_zring_bf16_a<28, 28, 8, 32, 8>), grid,
+                           dim3(512), 0, ctx->stream, (const cfx_bf16*)in,
+                           (const cfx_bf16*)wgt, bias,
+                           (const cfx_bf16*)residual, (cfx_bf16*)out, N, D,
+                           H, W, do_elu);
+    else if (use_pl == 9)  // ablation: no sched_barriers
+        hipLaunchKernelGGL((k_conv3_zring_bf16_a<28, 28, 8, 32, 4, 0>),
+                           grid, dim3(512), 0, ctx->stream,
+                           (const cfx_bf16*)in, (const cfx_bf16*)wgt, bias,
+                           (const cfx_bf16*)residual, (cfx_bf16*)out, N, D,
+                           H, W, do_elu);
+    else if (use_pl == 10)  // ablation: cheap epilogue (WRONG results)
+        hipLaunchKernelGGL((k_conv3_zring_bf16_a<28, 28, 8, 32, 4, 1, 0>),
+                           grid, dim3(512), 0, ctx->stream,
+                           (const cfx_bf16*)in, (const cfx_bf16*)wgt, bias,
+                           (const cfx_bf16*)residual, (cfx_bf16*)out, N, D,
+                           H, W, do_elu);
+    else if (use_pl == 7)
         hipLaunchKernelGGL((k_conv3_zring_bf16_t<28, 28, 8, 32>), grid,
                            dim3(512), 0, ctx->stream, (const cfx_bf16*)in,
                            (const cfx_bf16*)wgt, bias,
