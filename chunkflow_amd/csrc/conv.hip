@@ -1561,10 +1561,14 @@ __global__ __launch_bounds__(512, 1) void k_conv3_zring_bf16_a(
     constexpr int C4 = CP / 4;
     constexpr int LV = (SY * SX * C4 + 511) / 512;
     typedef cfx_bf16 bf16x4 __attribute__((ext_vector_type(4)));
+    typedef cfx_bf16 bf16x8w __attribute__((ext_vector_type(8)));
     static_assert(C <= CP && K <= 32, "");
 
     __shared__ cfx_bf16 ring[3 * SY * SX * PCB];
     __shared__ cfx_bf16 wall[27 * 32 * PCB];
+    // wave-private transpose scratch for the EPI==2 epilogue:
+    // [wave][x-row 16][j 32] bf16 (1 KB per wave; no cross-wave sync)
+    __shared__ cfx_bf16 oscr[8][16][32];
 
     const int n = blockIdx.z;
     const int y0 = blockIdx.y * TY;
@@ -1722,6 +1726,56 @@ __global__ __launch_bounds__(512, 1) void k_conv3_zring_bf16_a(
         if (!EPI) {  // timing ablation: keep acc alive, skip the real
                      // epilogue (WRONG results; CFX_ZRING_PL=10 only)
             out[tid] = (cfx_bf16)(acc[0] + acc[15]);
+            continue;
+        }
+        if (EPI == 2) {
+            // transposed epilogue: the j-per-lane layout stores 16
+            // scattered 2-byte elements per lane per z; those 128 store
+            // instructions per CU per z measured as 40% of the launch
+            // (PL=10 ablation). Bounce the tile through
+            // wave-private LDS and store x-major: one b128/b64 store
+            // per lane per z, coalesced. Residual reads ride the same
+            // wide path.
+            const int gy = y0 + wave;
+            const int j = lane & 31;
+            const float bj = bias ? bias[j] : 0.f;
+#pragma unroll
+            for (int r = 0; r < 16; ++r) {
+                const int row = (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
+                float v = acc[r] + bj;
+                oscr[wave][row][j] = (cfx_bf16)v;
+            }
+            const int xr = lane >> 2;           // x row 0..15
+            const int ch = lane & 3;            // 16-byte chunk 0..3
+            const int gx = x0 + xr;
+            if (gy < H && gx < W) {
+                const int j0 = ch * 8;
+                long long o = ((((long long)n * D + z) * H + gy) * W +
+                               gx) * K + j0;
+                if (j0 + 8 <= K) {
+                    bf16x8w v = *reinterpret_cast<const bf16x8w*>(
+                        &oscr[wave][xr][j0]);
+#pragma unroll
+                    for (int e = 0; e < 8; ++e) {
+                        float t = (float)v[e];
+                        if (res) t += (float)res[o + e];
+                        if (do_elu) t = t > 0.f ? t : expm1f(t);
+                        v[e] = (cfx_bf16)t;
+                    }
+                    *reinterpret_cast<bf16x8w*>(out + o) = v;
+                } else if (j0 < K) {            // tail chunk (K=28: 4 j)
+                    bf16x4 v = *reinterpret_cast<const bf16x4*>(
+                        &oscr[wave][xr][j0]);
+#pragma unroll
+                    for (int e = 0; e < 4; ++e) {
+                        float t = (float)v[e];
+                        if (res) t += (float)res[o + e];
+                        if (do_elu) t = t > 0.f ? t : expm1f(t);
+                        v[e] = (cfx_bf16)t;
+                    }
+                    *reinterpret_cast<bf16x4*>(out + o) = v;
+                }
+            }
             continue;
         }
         const int gy = y0 + wave;
@@ -2830,6 +2884,12 @@ extern "C" int cfx_conv3_ndhwc_bf16(cfx_ctx* ctx, const void* in,
                            H, W, do_elu);
     else if (use_pl == 9)  // ablation: no sched_barriers
         hipLaunchKernelGGL((k_conv3_zring_bf16_a<28, 28, 8, 32, 4, 0>),
+                           grid, dim3(512), 0, ctx->stream,
+                           (const cfx_bf16*)in, (const cfx_bf16*)wgt, bias,
+                           (const cfx_bf16*)residual, (cfx_bf16*)out, N, D,
+                           H, W, do_elu);
+    else if (use_pl == 11)  // transposed wide-store epilogue
+        hipLaunchKernelGGL((k_conv3_zring_bf16_a<28, 28, 8, 32, 4, 1, 2>),
                            grid, dim3(512), 0, ctx->stream,
                            (const cfx_bf16*)in, (const cfx_bf16*)wgt, bias,
                            (const cfx_bf16*)residual, (cfx_bf16*)out, N, D,
