@@ -1732,48 +1732,54 @@ __global__ __launch_bounds__(512, 1) void k_conv3_zring_bf16_a(
             // transposed epilogue: the j-per-lane layout stores 16
             // scattered 2-byte elements per lane per z; those 128 store
             // instructions per CU per z measured as 40% of the launch
-            // (PL=10 ablation). Bounce the tile through
-            // wave-private LDS and store x-major: one b128/b64 store
-            // per lane per z, coalesced. Residual reads ride the same
-            // wide path.
+            // (PL=10 ablation). Bounce the 32x32 tile
+            // through wave-private LDS in two 16-row halves (the full
+            // tile would need 16 KB of scratch and bust the 160 KB LDS
+            // budget) and store x-major: one coalesced b128/b64 store
+            // per lane per half. Residual reads ride the same wide
+            // path; (acc+bias) rounds through bf16 before the res add
+            // (bf16-engine tolerance, not bit-identical to _pl).
             const int gy = y0 + wave;
             const int j = lane & 31;
             const float bj = bias ? bias[j] : 0.f;
 #pragma unroll
-            for (int r = 0; r < 16; ++r) {
-                const int row = (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
-                float v = acc[r] + bj;
-                oscr[wave][row][j] = (cfx_bf16)v;
-            }
-            const int xr = lane >> 2;           // x row 0..15
-            const int ch = lane & 3;            // 16-byte chunk 0..3
-            const int gx = x0 + xr;
-            if (gy < H && gx < W) {
-                const int j0 = ch * 8;
-                long long o = ((((long long)n * D + z) * H + gy) * W +
-                               gx) * K + j0;
-                if (j0 + 8 <= K) {
-                    bf16x8w v = *reinterpret_cast<const bf16x8w*>(
-                        &oscr[wave][xr][j0]);
+            for (int h = 0; h < 2; ++h) {
 #pragma unroll
-                    for (int e = 0; e < 8; ++e) {
-                        float t = (float)v[e];
-                        if (res) t += (float)res[o + e];
-                        if (do_elu) t = t > 0.f ? t : expm1f(t);
-                        v[e] = (cfx_bf16)t;
-                    }
-                    *reinterpret_cast<bf16x8w*>(out + o) = v;
-                } else if (j0 < K) {            // tail chunk (K=28: 4 j)
-                    bf16x4 v = *reinterpret_cast<const bf16x4*>(
-                        &oscr[wave][xr][j0]);
+                for (int r = 8 * h; r < 8 * h + 8; ++r) {
+                    const int row = (r & 3) + 8 * ((r >> 2) & 1) +
+                                    4 * (lane >> 5);
+                    oscr[wave][row][j] = (cfx_bf16)(acc[r] + bj);
+                }
+                const int xr = lane >> 2;       // x row 0..15 in half
+                const int ch = lane & 3;        // 16-byte chunk 0..3
+                const int gx = x0 + 16 * h + xr;
+                if (gy < H && gx < W) {
+                    const int j0 = ch * 8;
+                    long long o = ((((long long)n * D + z) * H + gy) *
+                                   W + gx) * K + j0;
+                    if (j0 + 8 <= K) {
+                        bf16x8w v = *reinterpret_cast<const bf16x8w*>(
+                            &oscr[wave][xr][j0]);
 #pragma unroll
-                    for (int e = 0; e < 4; ++e) {
-                        float t = (float)v[e];
-                        if (res) t += (float)res[o + e];
-                        if (do_elu) t = t > 0.f ? t : expm1f(t);
-                        v[e] = (cfx_bf16)t;
+                        for (int e = 0; e < 8; ++e) {
+                            float t = (float)v[e];
+                            if (res) t += (float)res[o + e];
+                            if (do_elu) t = t > 0.f ? t : expm1f(t);
+                            v[e] = (cfx_bf16)t;
+                        }
+                        *reinterpret_cast<bf16x8w*>(out + o) = v;
+                    } else if (j0 < K) {        // tail chunk (K=28)
+                        bf16x4 v = *reinterpret_cast<const bf16x4*>(
+                            &oscr[wave][xr][j0]);
+#pragma unroll
+                        for (int e = 0; e < 4; ++e) {
+                            float t = (float)v[e];
+                            if (res) t += (float)res[o + e];
+                            if (do_elu) t = t > 0.f ? t : expm1f(t);
+                            v[e] = (cfx_bf16)t;
+                        }
+                        *reinterpret_cast<bf16x4*>(out + o) = v;
                     }
-                    *reinterpret_cast<bf16x4*>(out + o) = v;
                 }
             }
             continue;
