@@ -2474,6 +2474,7 @@ __global__ __launch_bounds__(512, 1) void k_conv3_zring_bf16_s(
 
     __shared__ cfx_bf16 ring[3 * SY * SX * PCB];
     __shared__ cfx_bf16 wall[27 * 32 * PCB];
+    __shared__ cfx_bf16 oscr[8][16][32];  // transposed-epilogue scratch
 
     const int n = blockIdx.z;
     const int y0 = blockIdx.y * TY;
@@ -2618,34 +2619,58 @@ __global__ __launch_bounds__(512, 1) void k_conv3_zring_bf16_s(
         }
 
         acc = acc + accB;
+        // transposed wide-store epilogue (see k_conv3_zring_bf16_a
+        // EPI==2): two 16-row halves through wave-private LDS, one
+        // coalesced b128/b64 store per lane per half. rem per 8-j chunk
+        // is always 8, 4 or <=0 for K in {28, 36, 48}.
         const int gy = y0 + wave;
         const int j = lane & 31;
-        if (gy < H && j0 + j < Ktot) {
-            const float bj = bias ? bias[j0 + j] : 0.f;
-            cfx_bf16 rv[16];
-            if (res) {
+        if (gy < H) {
+            const float bj =
+                (bias && j0 + j < Ktot) ? bias[j0 + j] : 0.f;
 #pragma unroll
-                for (int r = 0; r < 16; ++r) {
-                    const int row =
-                        (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
-                    const int gx = min(x0 + row, W - 1);
-                    rv[r] = res[
-                        (((long long)n * D + z) * H + gy) * (long long)W *
-                            KS + (long long)gx * KS + j0 + j];
+            for (int h = 0; h < 2; ++h) {
+#pragma unroll
+                for (int r = 8 * h; r < 8 * h + 8; ++r) {
+                    const int row = (r & 3) + 8 * ((r >> 2) & 1) +
+                                    4 * (lane >> 5);
+                    oscr[wave][row][j] = (cfx_bf16)(acc[r] + bj);
                 }
-            }
+                const int xr = lane >> 2;
+                const int ch = lane & 3;
+                const int gx = x0 + 16 * h + xr;
+                const int rem = Ktot - j0 - ch * 8;
+                if (gx < W && rem >= 4) {
+                    const int j0c = ch * 8;
+                    long long o = (((long long)n * D + z) * H + gy) *
+                                      (long long)W * KS +
+                                  (long long)gx * KS + j0 + j0c;
+                    typedef cfx_bf16 bf16x8s
+                        __attribute__((ext_vector_type(8)));
+                    if (rem >= 8) {
+                        bf16x8s v = *reinterpret_cast<const bf16x8s*>(
+                            &oscr[wave][xr][j0c]);
 #pragma unroll
-            for (int r = 0; r < 16; ++r) {
-                const int row = (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
-                const int gx = x0 + row;
-                if (gx >= W) continue;
-                long long o = (((long long)n * D + z) * H + gy) *
-                                  (long long)W * KS +
-                              (long long)gx * KS + j0 + j;
-                float v = acc[r] + bj;
-                if (res) v += (float)rv[r];
-                if (do_elu) v = v > 0.f ? v : expm1f(v);
-                out[o] = (cfx_bf16)v;
+                        for (int e = 0; e < 8; ++e) {
+                            float t = (float)v[e];
+                            if (res) t += (float)res[o + e];
+                            if (do_elu) t = t > 0.f ? t : expm1f(t);
+                            v[e] = (cfx_bf16)t;
+                        }
+                        *reinterpret_cast<bf16x8s*>(out + o) = v;
+                    } else {
+                        bf16x4 v = *reinterpret_cast<const bf16x4*>(
+                            &oscr[wave][xr][j0c]);
+#pragma unroll
+                        for (int e = 0; e < 4; ++e) {
+                            float t = (float)v[e];
+                            if (res) t += (float)res[o + e];
+                            if (do_elu) t = t > 0.f ? t : expm1f(t);
+                            v[e] = (cfx_bf16)t;
+                        }
+                        *reinterpret_cast<bf16x4*>(out + o) = v;
+                    }
+                }
             }
         }
     }
@@ -2874,7 +2899,7 @@ extern "C" int cfx_conv3_ndhwc_bf16(cfx_ctx* ctx, const void* in,
     // DESIGN.md §10-r2 item 6)
     static const int use_pl = [] {
         const char* e = getenv("CFX_ZRING_PL");
-        return e ? atoi(e) : 6;
+        return e ? atoi(e) : 11;
     }();
     hipEvent_t e0;
     if (prof_begin(ctx, &e0)) return -1;
