@@ -143,11 +143,12 @@ def maybe_accelerate(model: nn.Module, device_index: int = 0) -> int:
     return count
 
 
-# bf16 ring widths that BEAT MIOpen bf16 in steady state (measured,
-# profiles/updown_probe_r02.json): C=28 534 TF (3.5x), C=36 sliced 205 vs
-# 195 TF. C=48 sliced measured-rejected: 269-295 TF vs MIOpen's 348-355 on
-# the 64^2 extent (kernel kept callable + tested for the record).
-BF16_WIDTHS = (28, 36)
+# bf16 ring widths on the hand kernels (measured,
+# profiles/updown_probe_r02.json): C=28 594-624 TF (4x MIOpen), C=36
+# sliced 249 TF (1.31x). C=48 sliced is 318 vs MIOpen's 344 raw, but the
+# fused ResBlock epilogue removes the ELU/add elementwise passes MIOpen
+# needs, so the BLOCK-level comparison decides (bench A/B r02).
+BF16_WIDTHS = (28, 36, 48)
 
 
 class CfxConv3dBF16(nn.Module):
