@@ -7,6 +7,8 @@ convs, ~85% of the affinity UNet's FLOPs. Everything else stays on
 MIOpen. Enabled in the pytorch engine via CFX_FASTCONV (see engine);
 weights are repacked once to the kernel's (27, C, K) tap-major layout.
 """
+import os
+
 import torch
 import torch.nn as nn
 
@@ -148,7 +150,9 @@ def maybe_accelerate(model: nn.Module, device_index: int = 0) -> int:
 # sliced 249 TF (1.31x). C=48 sliced is 318 vs MIOpen's 344 raw, but the
 # fused ResBlock epilogue removes the ELU/add elementwise passes MIOpen
 # needs, so the BLOCK-level comparison decides (bench A/B r02).
-BF16_WIDTHS = (28, 36, 48)
+BF16_WIDTHS = tuple(
+    int(w) for w in os.environ.get('CFX_BF16_WIDTHS', '28,36,48').split(',')
+    if w)
 
 
 class CfxConv3dBF16(nn.Module):
