@@ -147,11 +147,12 @@ def maybe_accelerate(model: nn.Module, device_index: int = 0) -> int:
 
 # bf16 ring widths on the hand kernels (measured,
 # profiles/updown_probe_r02.json): C=28 594-624 TF (4x MIOpen), C=36
-# sliced 249 TF (1.31x). C=48 sliced is 318 vs MIOpen's 344 raw, but the
-# fused ResBlock epilogue removes the ELU/add elementwise passes MIOpen
-# needs, so the BLOCK-level comparison decides (bench A/B r02).
+# sliced 249 TF (1.31x). C=48 sliced reaches 318 vs MIOpen's 344 raw and
+# even with the fused ResBlock epilogue the same-box bench A/B reads
+# 292.4/293.5 (C48 on) vs 297.1/296.5 (off) MV/s -- measured-rejected,
+# kernel kept callable behind CFX_BF16_WIDTHS=28,36,48.
 BF16_WIDTHS = tuple(
-    int(w) for w in os.environ.get('CFX_BF16_WIDTHS', '28,36,48').split(',')
+    int(w) for w in os.environ.get('CFX_BF16_WIDTHS', '28,36').split(',')
     if w)
 
 
