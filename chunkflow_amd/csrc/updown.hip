@@ -399,7 +399,9 @@ __global__ __launch_bounds__(256, 2) void k_conv155_out(
 #pragma unroll 1
     for (int tap = 0; tap < 25; ++tap) {
         const int dy = tap / 5, dx = tap % 5;
-#pragma unroll 1
+        // unroll 2 gives the dependent LDS reads cross-iteration ILP
+        // (full unroll hoists 7x3 float4 weight reads and spills)
+#pragma unroll 2
         for (int c4 = 0; c4 < C4; ++c4) {
             float4 w4k[KO];
 #pragma unroll
