@@ -1048,3 +1048,38 @@ def test_sliced_ring_fuzz_shapes():
             x.to(torch.bfloat16)
             .contiguous(memory_format=torch.channels_last_3d)).float()
         torch.testing.assert_close(got, want, rtol=0.1, atol=0.05)
+
+
+@pytest.mark.timeout(600)
+def test_zring_bf16_variant_sweep():
+    """Every bf16 ring variant behind CFX_ZRING_PL computes the same conv
+    (vs torch bf16). The env is read once per process, so each variant
+    runs in a subprocess. PL=10 is a timing-only ablation and excluded."""
+    import subprocess
+    import sys
+    code = (
+        'import sys, torch; sys.path.insert(0, ".")\n'
+        'from chunkflow_amd.fastconv import CfxConv3dBF16\n'
+        'torch.manual_seed(0)\n'
+        'conv = torch.nn.Conv3d(28, 28, 3, padding=1).cuda()\n'
+        'm = CfxConv3dBF16(conv).cuda()\n'
+        'x = (torch.randn(2, 28, 5, 37, 41, device="cuda") * 0.3)'
+        '.to(torch.bfloat16)'
+        '.contiguous(memory_format=torch.channels_last_3d)\n'
+        'r = (torch.randn(2, 28, 5, 37, 41, device="cuda") * 0.3)'
+        '.to(torch.bfloat16)'
+        '.contiguous(memory_format=torch.channels_last_3d)\n'
+        'want = torch.nn.functional.elu(\n'
+        '    torch.nn.functional.conv3d(x, conv.weight.to(torch.bfloat16)'
+        '.to(memory_format=torch.channels_last_3d),'
+        ' conv.bias.to(torch.bfloat16), padding=1) + r).float()\n'
+        'got = m._run(x, residual=r, elu=True).float()\n'
+        'torch.testing.assert_close(got, want, rtol=0.06, atol=0.06)\n'
+        'print("OK")\n')
+    for pl in ('0', '1', '2', '3', '6', '8', '9', '11'):
+        env = dict(os.environ, CFX_ZRING_PL=pl)
+        p = subprocess.run([sys.executable, '-c', code], env=env,
+                           capture_output=True, text=True, timeout=240,
+                           cwd=REPO)
+        assert p.returncode == 0 and 'OK' in p.stdout, \
+            (pl, p.stdout[-500:], p.stderr[-500:])
