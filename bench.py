@@ -121,19 +121,26 @@ def cpu_baseline(args, chunk_u8: np.ndarray, budget_s: float,
     }
 
 
-def read_pmc_traffic():
+def read_pmc_traffic(args):
     """(bytes, provenance) of the blend kernel's per-launch HBM traffic
-    from a committed rocprofv3 PMC measurement (profiles/pmc_traffic.json;
-    collected offline — rocprofv3 PMC passes cannot run inside the timed
-    bench), or (None, None)."""
-    path = os.path.join(REPO, 'profiles', 'pmc_traffic.json')
+    from a committed rocprofv3 PMC measurement (collected offline —
+    rocprofv3 PMC passes cannot run inside the timed bench). The value is
+    only attached when the run's workload MATCHES the mix it was measured
+    on (per-launch blend sizes differ between configs); otherwise
+    (None, reason)."""
+    if args.dtype == 'float32' and tuple(args.patch_size) == (20, 256, 256):
+        path = os.path.join(REPO, 'profiles', 'pmc_traffic.json')
+    elif args.dtype == 'bfloat16' and             tuple(args.patch_size) == (32, 256, 256):
+        path = os.path.join(REPO, 'profiles', 'pmc_traffic_bf16.json')
+    else:
+        return None, 'no PMC measurement for this workload mix'
     if not os.path.exists(path):
-        return None, None
+        return None, 'no PMC measurement for this workload mix'
     try:
         with open(path) as f:
             d = json.load(f)
         return (float(d['blend_bytes_per_launch']),
-                d.get('provenance', 'profiles/pmc_traffic.json'))
+                d.get('provenance', path))
     except Exception:
         return None, None
 
@@ -245,7 +252,7 @@ def main():
     voxels = float(cz) * cy * cx * world * args.steps
     value = voxels / elapsed
 
-    traffic, traffic_src = read_pmc_traffic()
+    traffic, traffic_src = read_pmc_traffic(args)
     conv_roofline = None
     if conv and conv['count'] > 0 and conv['total_ms'] > 0:
         tflops = conv['bytes'] / (conv['total_ms'] / 1e3) / 1e12

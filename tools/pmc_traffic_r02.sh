@@ -6,11 +6,13 @@
 # gfx950 FETCH_SIZE reports HALF the bytes of a wide coalesced read.
 repo=$(pwd)
 cd /tmp && export TMPDIR=/tmp && cd "$repo"
+EXTRA="${BENCH_ARGS:-}"
+TAG="${MIX_TAG:-}"
 for ctr in FETCH_SIZE WRITE_SIZE; do
   timeout 500 rocprofv3 --pmc $ctr --output-format csv \
-    -d gpurun_out/pmc_traffic_$ctr -- \
-    python bench.py --steps 1 --warmup 1 --no-cpu-baseline \
-    > gpurun_out/pmc_traffic_$ctr.log 2>&1
+    -d gpurun_out/pmc_traffic$TAG-$ctr -- \
+    python bench.py --steps 1 --warmup 1 --no-cpu-baseline $EXTRA \
+    > gpurun_out/pmc_traffic$TAG-$ctr.log 2>&1
   echo "$ctr rc=$?"
 done
 python - << 'PYEOF'
@@ -19,7 +21,9 @@ res = {}
 for ctr in ('FETCH_SIZE', 'WRITE_SIZE'):
     agg = collections.defaultdict(float)
     cnt = collections.defaultdict(int)
-    for f in glob.glob(f'gpurun_out/pmc_traffic_{ctr}/**/*.csv',
+    import os
+    mixtag = os.environ.get('MIX_TAG', '')
+    for f in glob.glob(f'gpurun_out/pmc_traffic{mixtag}-{ctr}/**/*.csv',
                        recursive=True):
         with open(f) as fh:
             for row in csv.DictReader(fh):
@@ -34,7 +38,9 @@ for ctr in ('FETCH_SIZE', 'WRITE_SIZE'):
                         cnt[tag] += 1
     res[ctr] = {k: {'kb_total': agg[k], 'launches': cnt[k],
                     'kb_avg': agg[k] / max(cnt[k], 1)} for k in agg}
-with open('gpurun_out/pmc_traffic_r02.json', 'w') as f:
+import os
+with open('gpurun_out/pmc_traffic_r02%s.json' % os.environ.get('MIX_TAG', ''),
+          'w') as f:
     json.dump(res, f, indent=1)
 print(json.dumps(res, indent=1))
 PYEOF
