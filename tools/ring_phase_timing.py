@@ -25,7 +25,8 @@ cfx.conv3_ndhwc_bf16(x.data_ptr(), w.data_ptr(), None, None,
                      out.data_ptr(), N, D, H, W, C, C)
 torch.cuda.synchronize()
 n_wg = (W // 32) * (H // 8) * N
-rec = out.view(torch.int64).flatten()[:n_wg * 6].view(n_wg, 6).cpu()
+flat = out.contiguous().flatten()
+rec = flat[:n_wg * 6 * 4].view(torch.int64).view(n_wg, 6).cpu()
 rec = rec.double()
 names = ['plane_load_issue', 'mfma_phase1(36)', 'store+barrier',
          'mfma_phase2(18)', 'epilogue(scattered)', 'total']
