@@ -15,7 +15,8 @@ cl = torch.channels_last_3d
 x = torch.randn(N, C, D, H, W, device='cuda').to(torch.bfloat16) \
     .contiguous(memory_format=cl)
 w = torch.randn(27, 32, 32).to(torch.bfloat16).cuda().contiguous()
-out = torch.zeros_like(x)
+out = torch.zeros(N * C * D * H * W, dtype=torch.bfloat16,
+                  device='cuda')  # flat raw buffer (kernel sees a pointer)
 cfx = get_cfx(0)
 for _ in range(3):  # warm
     cfx.conv3_ndhwc_bf16(x.data_ptr(), w.data_ptr(), None, None,
@@ -25,8 +26,7 @@ cfx.conv3_ndhwc_bf16(x.data_ptr(), w.data_ptr(), None, None,
                      out.data_ptr(), N, D, H, W, C, C)
 torch.cuda.synchronize()
 n_wg = (W // 32) * (H // 8) * N
-flat = out.contiguous().flatten()
-rec = flat[:n_wg * 6 * 4].view(torch.int64).view(n_wg, 6).cpu()
+rec = out[:n_wg * 6 * 4].view(torch.int64).view(n_wg, 6).cpu()
 rec = rec.double()
 names = ['plane_load_issue', 'mfma_phase1(36)', 'store+barrier',
          'mfma_phase2(18)', 'epilogue(scattered)', 'total']
