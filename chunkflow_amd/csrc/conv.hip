@@ -1561,7 +1561,7 @@ __global__ __launch_bounds__(512, 1) void k_conv3_zring_bf16_pl(
 // accumulator-alternating variant (CFX_ZRING_PL=6): targets the
 // same-accumulator issue cliff (see comment in the z loop)
 template <int C, int K, int TY, int TX, int PDX = 4,
-          int SB = 1, int EPI = 1>
+          int SB = 1, int EPI = 1, int STG = 0>
 __global__ __launch_bounds__(512, 1) void k_conv3_zring_bf16_a(
     const cfx_bf16* __restrict__ in, const cfx_bf16* __restrict__ wgt,
     const float* __restrict__ bias, const cfx_bf16* __restrict__ res,
@@ -1648,6 +1648,12 @@ __global__ __launch_bounds__(512, 1) void k_conv3_zring_bf16_a(
         plane_store(0, v1, k1);
     }
     __syncthreads();
+    // STG=1: one-iteration-ahead staging — plane z+1 is loaded during
+    // iteration z-1, so the store below never waits on its own loads
+    // (the phase timing showed 24% of the z period in store+barrier)
+    bf16x4 va[LV];
+    bool ka[LV];
+    if (STG) plane_load(1, va, ka);
 
     const int ax = lane & 31;
     const int khalf = (lane >> 5) * 8;
@@ -1686,7 +1692,10 @@ __global__ __launch_bounds__(512, 1) void k_conv3_zring_bf16_a(
 
         bf16x4 vals[LV];
         bool keep[LV];
-        plane_load(z + 1, vals, keep);
+        if (STG)
+            plane_load(z + 2, vals, keep);  // lands during NEXT iteration
+        else
+            plane_load(z + 1, vals, keep);
         {
             bf16x8 abuf[PD], bbuf[PD];
 #pragma unroll
@@ -1710,7 +1719,10 @@ __global__ __launch_bounds__(512, 1) void k_conv3_zring_bf16_a(
                 }
                 if (SB) __builtin_amdgcn_sched_barrier(0);
             }
-            plane_store(z + 1, vals, keep);
+            if (STG)
+                plane_store(z + 1, va, ka);
+            else
+                plane_store(z + 1, vals, keep);
             __syncthreads();
 #pragma unroll
             for (int p = 36; p < 36 + PD; ++p) {
@@ -1735,6 +1747,13 @@ __global__ __launch_bounds__(512, 1) void k_conv3_zring_bf16_a(
             }
         }
 
+        if (STG) {
+#pragma unroll
+            for (int li = 0; li < LV; ++li) {
+                va[li] = vals[li];
+                ka[li] = keep[li];
+            }
+        }
         const f32x16 acc = accA + accB;
         if (!EPI) {  // timing ablation: keep acc alive, skip the real
                      // epilogue (WRONG results; CFX_ZRING_PL=10 only)
@@ -3171,6 +3190,12 @@ extern "C" int cfx_conv3_ndhwc_bf16(cfx_ctx* ctx, const void* in,
                            (const cfx_bf16*)in, (const cfx_bf16*)wgt, bias,
                            (const cfx_bf16*)residual, (cfx_bf16*)out, N, D,
                            H, W, do_elu);
+    else if (use_pl == 13)  // transposed epilogue + iteration-ahead staging
+        hipLaunchKernelGGL(
+            (k_conv3_zring_bf16_a<28, 28, 8, 32, 4, 1, 2, 1>), grid,
+            dim3(512), 0, ctx->stream, (const cfx_bf16*)in,
+            (const cfx_bf16*)wgt, bias, (const cfx_bf16*)residual,
+            (cfx_bf16*)out, N, D, H, W, do_elu);
     else if (use_pl == 12)  // s_memtime phase timing (WRONG results)
         hipLaunchKernelGGL((k_conv3_zring_bf16_tm<28, 28, 8, 32>), grid,
                            dim3(512), 0, ctx->stream, (const cfx_bf16*)in,
