@@ -3170,7 +3170,7 @@ extern "C" int cfx_conv3_ndhwc_bf16(cfx_ctx* ctx, const void* in,
     // DESIGN.md §10-r2 item 6)
     static const int use_pl = [] {
         const char* e = getenv("CFX_ZRING_PL");
-        return e ? atoi(e) : 11;
+        return e ? atoi(e) : 13;
     }();
     hipEvent_t e0;
     if (prof_begin(ctx, &e0)) return -1;

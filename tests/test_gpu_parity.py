@@ -1076,7 +1076,7 @@ def test_zring_bf16_variant_sweep():
         'got = m._run(x, residual=r, elu=True).float()\n'
         'torch.testing.assert_close(got, want, rtol=0.06, atol=0.06)\n'
         'print("OK")\n')
-    for pl in ('0', '1', '2', '3', '6', '8', '9', '11'):
+    for pl in ('0', '1', '2', '3', '6', '8', '9', '11', '13'):
         env = dict(os.environ, CFX_ZRING_PL=pl)
         p = subprocess.run([sys.executable, '-c', code], env=env,
                            capture_output=True, text=True, timeout=240,
