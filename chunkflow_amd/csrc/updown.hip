@@ -348,7 +348,7 @@ __global__ __launch_bounds__(256, 2) void k_conv155_c1(
 // the K=3 output dim (measured 38.6 ms per batch-24 launch). Thread =
 // one output position computing all K: 5 input rows staged to LDS in the
 // raw dtype, weights [K][25][28-padded] f32 in LDS, f32 accumulation.
-template <typename T, int XI, int CC, int KO>
+template <typename T, int XI, int CC, int KO, int MODE = 0>
 __global__ __launch_bounds__(256, 2) void k_conv155_out(
     const T* __restrict__ in, const T* __restrict__ wgt,
     const float* __restrict__ bias, T* __restrict__ out, int N, int D,
@@ -369,7 +369,8 @@ __global__ __launch_bounds__(256, 2) void k_conv155_out(
     const long long plane = (long long)nz * H;
     const int C4 = CC / 4;
     typedef typename vec4<T>::type tx4;
-    for (int idx = tid; idx < 8 * (XI + 4) * C4; idx += 256) {
+    for (int idx = tid; MODE != 2 && idx < 8 * (XI + 4) * C4;
+         idx += 256) {
         const int c4 = idx % C4;
         const int xl = (idx / C4) % (XI + 4);
         const int r = idx / (C4 * (XI + 4));
@@ -381,7 +382,7 @@ __global__ __launch_bounds__(256, 2) void k_conv155_out(
                      &in[((plane + gy) * W + gx) * (long long)CC + c4 * 4])
                : tx4{};
     }
-    for (int idx = tid; idx < KO * 25 * 28; idx += 256) {
+    for (int idx = tid; MODE != 2 && idx < KO * 25 * 28; idx += 256) {
         const int c = idx % 28;
         const int t = (idx / 28) % 25;
         const int k = idx / (28 * 25);
@@ -397,7 +398,7 @@ __global__ __launch_bounds__(256, 2) void k_conv155_out(
 #pragma unroll
         for (int k = 0; k < KO; ++k) acc[p][k] = bias ? bias[k] : 0.f;
 #pragma unroll 1
-    for (int tap = 0; tap < 25; ++tap) {
+    for (int tap = 0; tap < 25 && MODE != 1; ++tap) {  // MODE 1: skip
         const int dy = tap / 5, dx = tap % 5;
         // unroll 2 gives the dependent LDS reads cross-iteration ILP
         // (full unroll hoists 7x3 float4 weight reads and spills)
@@ -448,7 +449,21 @@ extern "C" int cfx_conv155_out(cfx_ctx* ctx, const void* in,
     if (prof_begin(ctx, &e0)) return -1;
     constexpr int XI = 128;  // wave covers 2*64 x; 4 rows per WG
     dim3 grid((W + XI - 1) / XI, (H + 3) / 4, (unsigned)(N * D));
-    if (is_bf16)
+    static const int om = [] {
+        const char* e = getenv("CFX_CONVOUT_MODE");  // timing ablation
+        return e ? atoi(e) : 0;
+    }();
+    if (is_bf16 && om == 1)
+        hipLaunchKernelGGL((k_conv155_out<cfx_bf16, XI, 28, 3, 1>), grid,
+                           dim3(256), 0, ctx->stream, (const cfx_bf16*)in,
+                           (const cfx_bf16*)wgt, bias, (cfx_bf16*)out, N,
+                           D, H, W);
+    else if (is_bf16 && om == 2)
+        hipLaunchKernelGGL((k_conv155_out<cfx_bf16, XI, 28, 3, 2>), grid,
+                           dim3(256), 0, ctx->stream, (const cfx_bf16*)in,
+                           (const cfx_bf16*)wgt, bias, (cfx_bf16*)out, N,
+                           D, H, W);
+    else if (is_bf16)
         hipLaunchKernelGGL((k_conv155_out<cfx_bf16, XI, 28, 3>), grid,
                            dim3(256), 0, ctx->stream, (const cfx_bf16*)in,
                            (const cfx_bf16*)wgt, bias, (cfx_bf16*)out, N,
