@@ -397,50 +397,30 @@ __global__ __launch_bounds__(256, 2) void k_conv155_out(
     for (int p = 0; p < 2; ++p)
 #pragma unroll
         for (int k = 0; k < KO; ++k) acc[p][k] = bias ? bias[k] : 0.f;
-    // software-pipelined over the flattened (tap, c4) iteration space:
-    // next iteration's 3 weight float4s + 2 input tx4s load into the
-    // alternate register set while the current one computes — without
-    // this every iteration sat in a ~50-cycle lgkmcnt wait (phase
-    // ablation: the whole 8.26 ms launch was this loop; staging +
-    // epilogue together are 0.09 ms)
-    if (MODE != 1) {
-        constexpr int NIT = 25 * C4;
-        auto addr_w = [&](int it, int k) {
-            return reinterpret_cast<const float4*>(
-                &s_w[k][it / C4][(it % C4) * 4]);
-        };
-        auto addr_v = [&](int it, int p) {
-            const int tap = it / C4, c4 = it % C4;
-            const int dy = tap / 5, dx = tap % 5;
-            return reinterpret_cast<const tx4*>(
-                &s_in[wave + dy][lane * 2 + p + dx][c4 * 4]);
-        };
-        float4 wb[2][KO];
-        tx4 vb[2][2];
-#pragma unroll
-        for (int k = 0; k < KO; ++k) wb[0][k] = *addr_w(0, k);
-        vb[0][0] = *addr_v(0, 0);
-        vb[0][1] = *addr_v(0, 1);
+#pragma unroll 1
+    for (int tap = 0; tap < 25 && MODE != 1; ++tap) {  // MODE 1: skip
+        const int dy = tap / 5, dx = tap % 5;
+        // unroll 2 gives the dependent LDS reads cross-iteration ILP
+        // (full unroll hoists 7x3 float4 weight reads and spills)
 #pragma unroll 2
-        for (int it = 0; it < NIT; ++it) {
-            const int cur = it & 1, nxt = cur ^ 1;
-            if (it + 1 < NIT) {
+        for (int c4 = 0; c4 < C4; ++c4) {
+            float4 w4k[KO];
 #pragma unroll
-                for (int k = 0; k < KO; ++k)
-                    wb[nxt][k] = *addr_w(it + 1, k);
-                vb[nxt][0] = *addr_v(it + 1, 0);
-                vb[nxt][1] = *addr_v(it + 1, 1);
-            }
+            for (int k = 0; k < KO; ++k)
+                w4k[k] = *reinterpret_cast<const float4*>(
+                    &s_w[k][tap][c4 * 4]);
 #pragma unroll
             for (int p = 0; p < 2; ++p) {
-                const float f0 = vec4<T>::get(vb[cur][p], 0);
-                const float f1 = vec4<T>::get(vb[cur][p], 1);
-                const float f2 = vec4<T>::get(vb[cur][p], 2);
-                const float f3 = vec4<T>::get(vb[cur][p], 3);
+                const tx4 v4 = *reinterpret_cast<const tx4*>(
+                    &s_in[wave + dy][lane * 2 + p + dx][c4 * 4]);
+                const float f0 = vec4<T>::get(v4, 0);
+                const float f1 = vec4<T>::get(v4, 1);
+                const float f2 = vec4<T>::get(v4, 2);
+                const float f3 = vec4<T>::get(v4, 3);
 #pragma unroll
                 for (int k = 0; k < KO; ++k)
-                    acc[p][k] += f0 * wb[cur][k].x + f1 * wb[cur][k].y +
-                                 f2 * wb[cur][k].z + f3 * wb[cur][k].w;
+                    acc[p][k] += f0 * w4k[k].x + f1 * w4k[k].y +
+                                 f2 * w4k[k].z + f3 * w4k[k].w;
             }
         }
     }
