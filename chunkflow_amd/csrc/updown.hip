@@ -447,10 +447,7 @@ extern "C" int cfx_conv155_out(cfx_ctx* ctx, const void* in,
     }
     hipEvent_t e0;
     if (prof_begin(ctx, &e0)) return -1;
-    // XI=64: halves the staged-row LDS (30.5+8.4 KB) so 4 WGs co-reside
-    // per CU (16 waves = 4/SIMD) — the loop is LDS-latency-bound and
-    // needs the wave overlap (XI=128 at 2 waves/SIMD ran 8.3 ms)
-    constexpr int XI = 64;
+    constexpr int XI = 128;  // wave covers 2*64 x; 4 rows per WG
     dim3 grid((W + XI - 1) / XI, (H + 3) / 4, (unsigned)(N * D));
     static const int om = [] {
         const char* e = getenv("CFX_CONVOUT_MODE");  // timing ablation
