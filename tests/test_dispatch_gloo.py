@@ -241,3 +241,37 @@ def test_shard_and_stitch_world8(tmp_path):
         np.testing.assert_array_equal(vol[:, :, :, i * 8:(i + 1) * 8],
                                       np.full((3, 4, 8, 8), i + 1,
                                               dtype=np.float32))
+
+
+def _empty_rank_worker(rank, world, port, result_path):
+    os.environ['MASTER_ADDR'] = '127.0.0.1'
+    os.environ['MASTER_PORT'] = str(port)
+    os.environ['RANK'] = str(rank)
+    os.environ['WORLD_SIZE'] = str(world)
+    import torch.distributed as dist
+    from chunkflow_amd.cartesian import BoundingBoxes
+    from chunkflow_amd.dispatch import (init_distributed, shard_tasks,
+                                        stitch_to_rank0)
+    init_distributed(backend='gloo')
+    bboxes = BoundingBoxes.from_manual_setup(
+        (4, 8, 8), roi_size=(4, 8, 16))  # 2 tasks, 3 ranks: rank 2 empty
+    local = {i: torch.full((3, 4, 8, 8), float(i + 1))
+             for i in shard_tasks(list(range(2)), rank, world)}
+    vol = stitch_to_rank0(bboxes, local, 3, rank, world, 'cpu')
+    if rank == 0:
+        np.save(result_path, vol.numpy())
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(180)
+def test_stitch_with_empty_rank(tmp_path):
+    """More ranks than tasks: the taskless rank must not deadlock the
+    gather (rank 0 posts no receives from it)."""
+    result = str(tmp_path / 'vol_empty.npy')
+    mp.spawn(_empty_rank_worker, args=(3, 29581, result), nprocs=3,
+             join=True)
+    vol = np.load(result)
+    assert vol.shape == (3, 4, 8, 16)
+    np.testing.assert_array_equal(vol[:, :, :, :8], 1.0)
+    np.testing.assert_array_equal(vol[:, :, :, 8:], 2.0)
