@@ -465,6 +465,9 @@ def stitch(tasks, name, input_chunk_name, output_volume_name, file_name,
             collected[idx] = (task.get('bbox'), t)
             last_task = task
             last_task['log']['timer'][name] = time() - start
+    if collected and any(b is None for b, _ in collected.values()):
+        raise ValueError('stitch needs bbox-carrying tasks — put '
+                         'generate-tasks at the head of the pipeline')
     rank, world = init_distributed(backend=backend)
     all_bboxes = state.get('all_bboxes')
     if all_bboxes is None:

@@ -175,3 +175,15 @@ def test_inference_dtype_bfloat16(tmp_path, golden_dir, golden):
     ref = golden[1]['e2e_pytorch_out']
     np.testing.assert_allclose(got, ref, rtol=0.15, atol=0.06)
     assert float(np.abs(got - ref).mean()) < 2e-2
+
+
+def test_stitch_requires_bbox_tasks(tmp_path):
+    """stitch without generate-tasks (no bbox on the task) fails loudly."""
+    from click.testing import CliRunner
+    from chunkflow_amd.flow import main
+    r = CliRunner().invoke(main, [
+        'create-chunk', '--size', '8', '8', '8',
+        'stitch', '--backend', 'gloo',
+        '-f', str(tmp_path / 'x.npy')])
+    assert r.exit_code != 0
+    assert isinstance(r.exception, ValueError)
