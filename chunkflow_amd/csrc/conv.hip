@@ -2055,7 +2055,8 @@ __global__ __launch_bounds__(512, 1) void k_conv3_zring_bf16_q2(
 
 
 
-// s_memtime phase-timing clone (CFX_ZRING_PL=12; WRONG results)
+// s_memtime phase-timing clone of the PL=13 default (CFX_ZRING_PL=12;
+// WRONG results)
 template <int C, int K, int TY, int TX>
 __global__ __launch_bounds__(512, 1) void k_conv3_zring_bf16_tm(
     const cfx_bf16* __restrict__ in, const cfx_bf16* __restrict__ wgt,
@@ -2084,7 +2085,6 @@ __global__ __launch_bounds__(512, 1) void k_conv3_zring_bf16_tm(
     const int tid = threadIdx.x;
     const int wave = tid >> 6;
     const int lane = tid & 63;
-    // phase timers (s_memtime ticks = shader cycles; +~11% overhead)
     unsigned long long t_load = 0, t_mfma1 = 0, t_store = 0,
                        t_mfma2 = 0, t_epi = 0, t_mark;
 #define TM_MARK() t_mark = __builtin_amdgcn_s_memtime()
@@ -2148,6 +2148,12 @@ __global__ __launch_bounds__(512, 1) void k_conv3_zring_bf16_tm(
         plane_store(0, v1, k1);
     }
     __syncthreads();
+    // STG=1: one-iteration-ahead staging — plane z+1 is loaded during
+    // iteration z-1, so the store below never waits on its own loads
+    // (the phase timing showed 24% of the z period in store+barrier)
+    bf16x4 va[LV];
+    bool ka[LV];
+    plane_load(1, va, ka);
 
     const int ax = lane & 31;
     const int khalf = (lane >> 5) * 8;
@@ -2187,7 +2193,7 @@ __global__ __launch_bounds__(512, 1) void k_conv3_zring_bf16_tm(
         bf16x4 vals[LV];
         bool keep[LV];
         TM_MARK();
-        plane_load(z + 1, vals, keep);
+        plane_load(z + 2, vals, keep);
         TM_ACC(t_load);
         TM_MARK();
         {
@@ -2215,7 +2221,7 @@ __global__ __launch_bounds__(512, 1) void k_conv3_zring_bf16_tm(
             }
             TM_ACC(t_mfma1);
             TM_MARK();
-            plane_store(z + 1, vals, keep);
+            plane_store(z + 1, va, ka);
             __syncthreads();
             TM_ACC(t_store);
             TM_MARK();
@@ -2241,10 +2247,71 @@ __global__ __launch_bounds__(512, 1) void k_conv3_zring_bf16_tm(
                 __builtin_amdgcn_sched_barrier(0);
             }
         }
+
         TM_ACC(t_mfma2);
         TM_MARK();
-
+#pragma unroll
+        for (int li = 0; li < LV; ++li) {
+            va[li] = vals[li];
+            ka[li] = keep[li];
+        }
         const f32x16 acc = accA + accB;
+                if (1) {
+            // transposed epilogue: the j-per-lane layout stores 16
+            // scattered 2-byte elements per lane per z; those 128 store
+            // instructions per CU per z measured as 40% of the launch
+            // (PL=10 ablation). Bounce the 32x32 tile
+            // through wave-private LDS in two 16-row halves (the full
+            // tile would need 16 KB of scratch and bust the 160 KB LDS
+            // budget) and store x-major: one coalesced b128/b64 store
+            // per lane per half. Residual reads ride the same wide
+            // path; (acc+bias) rounds through bf16 before the res add
+            // (bf16-engine tolerance, not bit-identical to _pl).
+            const int gy = y0 + wave;
+            const int j = lane & 31;
+            const float bj = bias ? bias[j] : 0.f;
+#pragma unroll
+            for (int h = 0; h < 2; ++h) {
+#pragma unroll
+                for (int r = 8 * h; r < 8 * h + 8; ++r) {
+                    const int row = (r & 3) + 8 * ((r >> 2) & 1) +
+                                    4 * (lane >> 5);
+                    oscr[wave][row][j] = (cfx_bf16)(acc[r] + bj);
+                }
+                const int xr = lane >> 2;       // x row 0..15 in half
+                const int ch = lane & 3;        // 16-byte chunk 0..3
+                const int gx = x0 + 16 * h + xr;
+                if (gy < H && gx < W) {
+                    const int j0 = ch * 8;
+                    long long o = ((((long long)n * D + z) * H + gy) *
+                                   W + gx) * K + j0;
+                    if (j0 + 8 <= K) {
+                        bf16x8w v = *reinterpret_cast<const bf16x8w*>(
+                            &oscr[wave][xr][j0]);
+#pragma unroll
+                        for (int e = 0; e < 8; ++e) {
+                            float t = (float)v[e];
+                            if (res) t += (float)res[o + e];
+                            if (do_elu) t = t > 0.f ? t : expm1f(t);
+                            v[e] = (cfx_bf16)t;
+                        }
+                        *reinterpret_cast<bf16x8w*>(out + o) = v;
+                    } else if (j0 < K) {        // tail chunk (K=28)
+                        bf16x4 v = *reinterpret_cast<const bf16x4*>(
+                            &oscr[wave][xr][j0]);
+#pragma unroll
+                        for (int e = 0; e < 4; ++e) {
+                            float t = (float)v[e];
+                            if (res) t += (float)res[o + e];
+                            if (do_elu) t = t > 0.f ? t : expm1f(t);
+                            v[e] = (cfx_bf16)t;
+                        }
+                        *reinterpret_cast<bf16x4*>(out + o) = v;
+                    }
+                }
+            }
+            continue;
+        }
         const int gy = y0 + wave;
         const int j = lane & 31;
         if (gy < H && j < K) {
@@ -2276,8 +2343,6 @@ __global__ __launch_bounds__(512, 1) void k_conv3_zring_bf16_tm(
         }
         TM_ACC(t_epi);
     }
-    // dump: one record per workgroup from wave 0 lane 0 (timing mode --
-    // results in `out` are WRONG by design)
     if (tid == 0) {
         unsigned long long* dbg = reinterpret_cast<unsigned long long*>(out);
         const int wg = (blockIdx.z * gridDim.y + blockIdx.y) * gridDim.x +

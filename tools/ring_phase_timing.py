@@ -29,7 +29,7 @@ n_wg = (W // 32) * (H // 8) * N
 rec = out[:n_wg * 6 * 4].view(torch.int64).view(n_wg, 6).cpu()
 rec = rec.double()
 names = ['plane_load_issue', 'mfma_phase1(36)', 'store+barrier',
-         'mfma_phase2(18)', 'epilogue(scattered)', 'total']
+         'mfma_phase2(18)', 'epilogue+rotate', 'total']
 mean = rec.mean(dim=0) / D  # per z
 print(f'{n_wg} WGs, per-z cycles (wave-0 view, D={D}):')
 tot = float(mean[5])
