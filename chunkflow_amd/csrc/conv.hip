@@ -2988,6 +2988,211 @@ __global__ __launch_bounds__(512, 1) void k_conv3_zring_bf16_t(
 }
 
 
+
+// q3 (CFX_ZRING_PL=15): 4-slot PADDED ring (PCB=40, constant-offset
+// addressing preserved — the swizzled q2 lost more to per-lane address
+// VALU than its barrier saved) + the weight wall halved to the kk0
+// slices (PCB=24) with kk1 B-fragments streamed from global/L2 (the
+// 55 KB pack is L2-hot on every XCD). The slot stored in iteration z,
+// (z+3) % 4, is disjoint from every slot read in iterations z and z-1
+// after the single end-of-iteration barrier — which bounds wave skew
+// to < 1 iteration (the r2 phase timing put 40% of the non-epilogue z
+// period in the old mid-iteration store+barrier). Epilogue = the
+// transposed wide-store path; accumulators alternate.
+template <int C, int K, int TY, int TX>
+__global__ __launch_bounds__(512, 1) void k_conv3_zring_bf16_q3(
+    const cfx_bf16* __restrict__ in, const cfx_bf16* __restrict__ wgt,
+    const float* __restrict__ bias, const cfx_bf16* __restrict__ res,
+    cfx_bf16* __restrict__ out, int N, int D, int H, int W, int do_elu) {
+    constexpr int CP = 32;
+    constexpr int PCB = CP + 8;          // ring x-stride (conflict-free)
+    constexpr int WS = 24;               // wall row stride (16 used + 8)
+    constexpr int SX = TX + 2;
+    constexpr int SY = TY + 2;
+    constexpr int C4 = CP / 4;
+    constexpr int STOT = SY * SX * C4;
+    constexpr int LV = (STOT + 511) / 512;
+    typedef cfx_bf16 bf16x4 __attribute__((ext_vector_type(4)));
+    typedef cfx_bf16 bf16x8w __attribute__((ext_vector_type(8)));
+    static_assert(C <= CP && K <= 32, "");
+
+    __shared__ cfx_bf16 ring[4 * SY * SX * PCB];   // 108.8 KB
+    __shared__ cfx_bf16 wall[27 * 32 * WS];        // 41.5 KB (kk0 half)
+    __shared__ cfx_bf16 oscr[8][16][32];           // 8 KB
+
+    const int n = blockIdx.z;
+    const int y0 = blockIdx.y * TY;
+    const int x0 = blockIdx.x * TX;
+    const int tid = threadIdx.x;
+    const int wave = tid >> 6;
+    const int lane = tid & 63;
+
+    for (int idx = tid; idx < 27 * 32 * 16; idx += 512) {
+        const int c = idx % 16;
+        const int j = (idx / 16) % 32;
+        const int tap = idx / (16 * 32);
+        wall[(tap * 32 + j) * WS + c] = wgt[(tap * 32 + j) * 32 + c];
+    }
+
+    const bool xy_interior = y0 >= 1 && y0 + TY + 1 <= H && x0 >= 1 &&
+                             x0 + TX + 1 <= W;
+
+    auto plane_load = [&](int P, bf16x4 (&vals)[LV], bool (&keep)[LV]) {
+        const bool zin = P >= 0 && P < D;
+        const bool interior = zin && xy_interior;
+#pragma unroll
+        for (int li = 0; li < LV; ++li) {
+            const int idx = min(tid + li * 512, STOT - 1);
+            const int c4 = idx % C4;
+            const int v = idx / C4;
+            const int gy = y0 + v / SX - 1;
+            const int gx = x0 + v % SX - 1;
+            const bool cok = c4 * 4 < C;
+            const bool ok = cok && zin &&
+                            (interior || (gy >= 0 && gy < H && gx >= 0 &&
+                                          gx < W));
+            keep[li] = ok;
+            vals[li] = *reinterpret_cast<const bf16x4*>(
+                in + ((((long long)n * D + (zin ? P : 0)) * H +
+                       (ok ? gy : 0)) * W + (ok ? gx : 0)) * C +
+                (cok ? c4 * 4 : 0));
+        }
+    };
+    auto plane_store = [&](int P, const bf16x4 (&vals)[LV],
+                           const bool (&keep)[LV]) {
+        const int slot = ((P + 1) % 4 + 4) % 4;
+#pragma unroll
+        for (int li = 0; li < LV; ++li) {
+            const int idx = tid + li * 512;
+            if (idx >= STOT) break;
+            const int c4 = idx % C4;
+            const int v = idx / C4;
+            *reinterpret_cast<bf16x4*>(
+                &ring[((slot * SY + v / SX) * SX + v % SX) * PCB +
+                      c4 * 4]) = keep[li] ? vals[li] : bf16x4{};
+        }
+    };
+
+    {
+        bf16x4 v0[LV], v1[LV], v2[LV];
+        bool k0[LV], k1[LV], k2[LV];
+        plane_load(-1, v0, k0);
+        plane_load(0, v1, k1);
+        plane_load(1, v2, k2);
+        plane_store(-1, v0, k0);
+        plane_store(0, v1, k1);
+        plane_store(1, v2, k2);
+    }
+    __syncthreads();
+
+    const int ax = lane & 31;
+    const int khalf = (lane >> 5) * 8;
+
+    for (int z = 0; z < D; ++z) {
+        f32x16 accA = {};
+        f32x16 accB = {};
+        const cfx_bf16* planes[3];
+#pragma unroll
+        for (int dzi = 0; dzi < 3; ++dzi)
+            planes[dzi] =
+                &ring[(((z + dzi) % 4 + 4) % 4) * SY * SX * PCB];
+        auto addrA = [&](int p) {
+            const int tap = p >> 1, kk = p & 1;
+            const int dzi = tap / 9, tl = tap % 9;
+            const int dy = tl / 3 - 1, dx = tl % 3 - 1;
+            return reinterpret_cast<const bf16x8*>(
+                &planes[dzi][((1 + wave + dy) * SX + (1 + dx) + ax) * PCB +
+                             khalf + kk * 16]);
+        };
+        auto addrB = [&](int p) {  // kk0: LDS wall; kk1: global (L2-hot)
+            const int tap = p >> 1;
+            if (p & 1)
+                return reinterpret_cast<const bf16x8*>(
+                    &wgt[(tap * 32 + ax) * 32 + khalf + 16]);
+            return reinterpret_cast<const bf16x8*>(
+                &wall[(tap * 32 + ax) * WS + khalf]);
+        };
+        constexpr int PD = 4;
+
+        bf16x4 vals[LV];
+        bool keep[LV];
+        plane_load(z + 2, vals, keep);
+        {
+            bf16x8 abuf[PD], bbuf[PD];
+#pragma unroll
+            for (int p = 0; p < PD; ++p) {
+                abuf[p] = *addrA(p);
+                bbuf[p] = *addrB(p);
+            }
+#pragma unroll
+            for (int p = 0; p < 54; ++p) {
+                const int si = p % PD;
+                if (p & 1)
+                    accB = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                        abuf[si], bbuf[si], accB, 0, 0, 0);
+                else
+                    accA = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                        abuf[si], bbuf[si], accA, 0, 0, 0);
+                __builtin_amdgcn_sched_barrier(0);
+                if (p + PD < 54) {
+                    abuf[si] = *addrA(p + PD);
+                    bbuf[si] = *addrB(p + PD);
+                }
+                __builtin_amdgcn_sched_barrier(0);
+            }
+        }
+        plane_store(z + 2, vals, keep);  // slot (z+3)%4: disjoint
+
+        const f32x16 acc = accA + accB;
+        const int gy = y0 + wave;
+        const int j = lane & 31;
+        const float bj = (bias && j < K) ? bias[j] : 0.f;
+#pragma unroll
+        for (int h = 0; h < 2; ++h) {
+#pragma unroll
+            for (int r = 8 * h; r < 8 * h + 8; ++r) {
+                const int row = (r & 3) + 8 * ((r >> 2) & 1) +
+                                4 * (lane >> 5);
+                oscr[wave][row][j] = (cfx_bf16)(acc[r] + bj);
+            }
+            const int xr = lane >> 2;
+            const int ch = lane & 3;
+            const int gx = x0 + 16 * h + xr;
+            const int rem = K - ch * 8;
+            if (gy < H && gx < W && rem >= 4) {
+                const int j0c = ch * 8;
+                long long o = ((((long long)n * D + z) * H + gy) * W +
+                               gx) * K + j0c;
+                if (rem >= 8) {
+                    bf16x8w v = *reinterpret_cast<const bf16x8w*>(
+                        &oscr[wave][xr][j0c]);
+#pragma unroll
+                    for (int e = 0; e < 8; ++e) {
+                        float t = (float)v[e];
+                        if (res) t += (float)res[o + e];
+                        if (do_elu) t = t > 0.f ? t : expm1f(t);
+                        v[e] = (cfx_bf16)t;
+                    }
+                    *reinterpret_cast<bf16x8w*>(out + o) = v;
+                } else {
+                    bf16x4 v = *reinterpret_cast<const bf16x4*>(
+                        &oscr[wave][xr][j0c]);
+#pragma unroll
+                    for (int e = 0; e < 4; ++e) {
+                        float t = (float)v[e];
+                        if (res) t += (float)res[o + e];
+                        if (do_elu) t = t > 0.f ? t : expm1f(t);
+                        v[e] = (cfx_bf16)t;
+                    }
+                    *reinterpret_cast<bf16x4*>(out + o) = v;
+                }
+            }
+        }
+        __syncthreads();  // the only barrier per z
+    }
+}
+
+
 // Sliced bf16 ring: the _pl kernel generalized to a c-slice [c0, c0+CL)
 // of a wider channel dimension (runtime stride CS) and a j-tile
 // [j0, j0+32) of a wider K (runtime stride KS). Widths 36 and 48 run as
@@ -3463,6 +3668,12 @@ extern "C" int cfx_conv3_ndhwc_bf16(cfx_ctx* ctx, const void* in,
                            H, W, do_elu);
     else if (use_pl == 14)  // 4-slot swizzled ring, one barrier per z
         hipLaunchKernelGGL((k_conv3_zring_bf16_q2<28, 28, 8, 32>), grid,
+                           dim3(512), 0, ctx->stream, (const cfx_bf16*)in,
+                           (const cfx_bf16*)wgt, bias,
+                           (const cfx_bf16*)residual, (cfx_bf16*)out, N, D,
+                           H, W, do_elu);
+    else if (use_pl == 15)  // q3: 4-slot padded ring, half wall + L2 B
+        hipLaunchKernelGGL((k_conv3_zring_bf16_q3<28, 28, 8, 32>), grid,
                            dim3(512), 0, ctx->stream, (const cfx_bf16*)in,
                            (const cfx_bf16*)wgt, bias,
                            (const cfx_bf16*)residual, (cfx_bf16*)out, N, D,
