@@ -1658,6 +1658,13 @@ __global__ __launch_bounds__(512, 1) void k_conv3_zring_bf16_a(
     const int ax = lane & 31;
     const int khalf = (lane >> 5) * 8;
 
+    // EPI==3: each z's epilogue is DEFERRED into the next iteration's
+    // mainloop region (fences off there) so the compiler can interleave
+    // its ~60 VALU/LDS/store instructions into the MFMA issue shadows —
+    // the phase timing showed the per-z barrier phase-locks the waves
+    // and the pipe idles through the epilogue otherwise.
+    f32x16 prevA = {};
+    f32x16 prevB = {};
     for (int z = 0; z < D; ++z) {
         // TWO alternating accumulators: anything issued between two MFMAs
         // on the SAME accumulator costs a +43-cycle cliff (microarch
@@ -1688,6 +1695,52 @@ __global__ __launch_bounds__(512, 1) void k_conv3_zring_bf16_a(
             return reinterpret_cast<const bf16x8*>(
                 &wall[(tap * 32 + ax) * PCB + khalf + kk * 16]);
         };
+        auto epilogue = [&](int zz, const f32x16& acc) {
+            const int gy = y0 + wave;
+            const int j = lane & 31;
+            const float bj = (bias && j < K) ? bias[j] : 0.f;
+#pragma unroll
+            for (int h = 0; h < 2; ++h) {
+#pragma unroll
+                for (int r = 8 * h; r < 8 * h + 8; ++r) {
+                    const int row = (r & 3) + 8 * ((r >> 2) & 1) +
+                                    4 * (lane >> 5);
+                    oscr[wave][row][j] = (cfx_bf16)(acc[r] + bj);
+                }
+                const int xr = lane >> 2;
+                const int ch = lane & 3;
+                const int gx = x0 + 16 * h + xr;
+                const int rem = K - ch * 8;
+                if (gy < H && gx < W && rem >= 4) {
+                    const int j0c = ch * 8;
+                    long long o = ((((long long)n * D + zz) * H + gy) *
+                                   W + gx) * K + j0c;
+                    if (rem >= 8) {
+                        bf16x8w v = *reinterpret_cast<const bf16x8w*>(
+                            &oscr[wave][xr][j0c]);
+#pragma unroll
+                        for (int e = 0; e < 8; ++e) {
+                            float t = (float)v[e];
+                            if (res) t += (float)res[o + e];
+                            if (do_elu) t = t > 0.f ? t : expm1f(t);
+                            v[e] = (cfx_bf16)t;
+                        }
+                        *reinterpret_cast<bf16x8w*>(out + o) = v;
+                    } else {
+                        bf16x4 v = *reinterpret_cast<const bf16x4*>(
+                            &oscr[wave][xr][j0c]);
+#pragma unroll
+                        for (int e = 0; e < 4; ++e) {
+                            float t = (float)v[e];
+                            if (res) t += (float)res[o + e];
+                            if (do_elu) t = t > 0.f ? t : expm1f(t);
+                            v[e] = (cfx_bf16)t;
+                        }
+                        *reinterpret_cast<bf16x4*>(out + o) = v;
+                    }
+                }
+            }
+        };
         constexpr int PD = PDX;
 
         bf16x4 vals[LV];
@@ -1703,6 +1756,7 @@ __global__ __launch_bounds__(512, 1) void k_conv3_zring_bf16_a(
                 abuf[p] = *addrA(p);
                 bbuf[p] = *addrB(p);
             }
+            if (EPI == 3 && z > 0) epilogue(z - 1, prevA + prevB);
 #pragma unroll
             for (int p = 0; p < 36; ++p) {  // dzi 0,1: planes z-1, z
                 const int si = p % PD;
@@ -1753,6 +1807,11 @@ __global__ __launch_bounds__(512, 1) void k_conv3_zring_bf16_a(
                 va[li] = vals[li];
                 ka[li] = keep[li];
             }
+        }
+        if (EPI == 3) {
+            prevA = accA;
+            prevB = accB;
+            continue;
         }
         const f32x16 acc = accA + accB;
         if (!EPI) {  // timing ablation: keep acc alive, skip the real
@@ -1843,6 +1902,55 @@ __global__ __launch_bounds__(512, 1) void k_conv3_zring_bf16_a(
                 if (res) v += (float)rv[r];
                 if (do_elu) v = v > 0.f ? v : expm1f(v);
                 out[o] = (cfx_bf16)v;
+            }
+        }
+    }
+    if (EPI == 3) {
+        // last z's epilogue (the in-loop lambda is out of scope here;
+        // inline copy with zz = D-1)
+        const f32x16 acc = prevA + prevB;
+        const int gy = y0 + wave;
+        const int j = lane & 31;
+        const float bj = (bias && j < K) ? bias[j] : 0.f;
+#pragma unroll
+        for (int h = 0; h < 2; ++h) {
+#pragma unroll
+            for (int r = 8 * h; r < 8 * h + 8; ++r) {
+                const int row = (r & 3) + 8 * ((r >> 2) & 1) +
+                                4 * (lane >> 5);
+                oscr[wave][row][j] = (cfx_bf16)(acc[r] + bj);
+            }
+            const int xr = lane >> 2;
+            const int ch = lane & 3;
+            const int gx = x0 + 16 * h + xr;
+            const int rem = K - ch * 8;
+            if (gy < H && gx < W && rem >= 4) {
+                const int j0c = ch * 8;
+                long long o = ((((long long)n * D + (D - 1)) * H + gy) *
+                               W + gx) * K + j0c;
+                if (rem >= 8) {
+                    bf16x8w v = *reinterpret_cast<const bf16x8w*>(
+                        &oscr[wave][xr][j0c]);
+#pragma unroll
+                    for (int e = 0; e < 8; ++e) {
+                        float t = (float)v[e];
+                        if (res) t += (float)res[o + e];
+                        if (do_elu) t = t > 0.f ? t : expm1f(t);
+                        v[e] = (cfx_bf16)t;
+                    }
+                    *reinterpret_cast<bf16x8w*>(out + o) = v;
+                } else {
+                    bf16x4 v = *reinterpret_cast<const bf16x4*>(
+                        &oscr[wave][xr][j0c]);
+#pragma unroll
+                    for (int e = 0; e < 4; ++e) {
+                        float t = (float)v[e];
+                        if (res) t += (float)res[o + e];
+                        if (do_elu) t = t > 0.f ? t : expm1f(t);
+                        v[e] = (cfx_bf16)t;
+                    }
+                    *reinterpret_cast<bf16x4*>(out + o) = v;
+                }
             }
         }
     }
@@ -3672,6 +3780,18 @@ extern "C" int cfx_conv3_ndhwc_bf16(cfx_ctx* ctx, const void* in,
                            (const cfx_bf16*)wgt, bias,
                            (const cfx_bf16*)residual, (cfx_bf16*)out, N, D,
                            H, W, do_elu);
+    else if (use_pl == 16)  // deferred epilogue, fences off (interleave)
+        hipLaunchKernelGGL(
+            (k_conv3_zring_bf16_a<28, 28, 8, 32, 4, 0, 3, 1>), grid,
+            dim3(512), 0, ctx->stream, (const cfx_bf16*)in,
+            (const cfx_bf16*)wgt, bias, (const cfx_bf16*)residual,
+            (cfx_bf16*)out, N, D, H, W, do_elu);
+    else if (use_pl == 17)  // deferred epilogue, fences ON (control)
+        hipLaunchKernelGGL(
+            (k_conv3_zring_bf16_a<28, 28, 8, 32, 4, 1, 3, 1>), grid,
+            dim3(512), 0, ctx->stream, (const cfx_bf16*)in,
+            (const cfx_bf16*)wgt, bias, (const cfx_bf16*)residual,
+            (cfx_bf16*)out, N, D, H, W, do_elu);
     else if (use_pl == 15)  // q3: 4-slot padded ring, half wall + L2 B
         hipLaunchKernelGGL((k_conv3_zring_bf16_q3<28, 28, 8, 32>), grid,
                            dim3(512), 0, ctx->stream, (const cfx_bf16*)in,
