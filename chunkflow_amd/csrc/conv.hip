@@ -1,6 +1,23 @@
 // Hand-written MFMA 3x3x3 convolution for gfx950 — the RSUNet ResBlock
 // conv (C == K in {28, 36, 48, 64}, stride 1, pad 1, NDHWC f32).
 //
+// VARIANT INDEX (this file is a measured ladder, not dead code; every
+// rejected variant stays callable behind its env switch and is covered
+// by the sweep test in tests/test_gpu_parity.py; numbers in DESIGN.md
+// §10/§10-r2):
+//   f32: k_conv3 (slab), k_conv3_w32 (32x32x2), k_conv3_zring(+_pl)
+//        [SHIPPED, 96.8 TF C=28], k_conv3_zring_dw (C=48, rejected 37 TF)
+//   bf16 C=28 (CFX_ZRING_PL): 0 plain 458 | 1 _pl pipelined 528 |
+//        2 _q 4-slot PCB36 104 (LDS conflicts) | 3 _pl2 prefetch 527 |
+//        4 _w 2-row swizzled 280 (1 wave/SIMD) | 6 _a alt-acc 540 |
+//        7 _t 16x16x32 444 | 8/9 PD/fence ablations | 10/12 timing-only
+//        (WRONG results) | 11 _a+transposed epilogue 594-624 |
+//        13 +iteration-ahead staging [SHIPPED DEFAULT, 610-646] |
+//        14 _q2 4-slot swizzled 559 | 15 _q3 4-slot padded half-wall 622
+//        (neutral) | 16/17 deferred epilogue 590/602 (rejected)
+//   bf16 C=36/48: k_conv3_zring_bf16_s sliced c-half x j-tile schedule
+//        [SHIPPED for 36 at 249 TF; 48 measured-rejected vs MIOpen]
+//
 // MIOpen's ck-xdlops kernels reach ~58 TF/s f32 on these shapes; the small
 // channel depth (N-dim 28..64) starves generic implicit GEMM. This kernel
 // exploits the structure directly:
