@@ -230,7 +230,10 @@ def main():
     barrier_sync()
     elapsed = time.perf_counter() - t0
     if world > 1:
-        t = torch.tensor([elapsed], device=device)
+        # gloo (the CFX_DIST_BACKEND dress-rehearsal path) moves CPU
+        # tensors only; nccl/RCCL wants the device tensor
+        red_dev = 'cpu' if dist.get_backend() == 'gloo' else device
+        t = torch.tensor([elapsed], device=red_dev)
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.item())
 

@@ -29,7 +29,9 @@ def init_distributed(backend: str = None):
         return 0, 1
     if not dist.is_initialized():
         if backend is None:
-            backend = 'nccl' if torch.cuda.is_available() else 'gloo'
+            backend = os.environ.get(
+                'CFX_DIST_BACKEND',
+                'nccl' if torch.cuda.is_available() else 'gloo')
         rank = int(os.environ['RANK'])
         if backend == 'nccl':
             torch.cuda.set_device(int(os.environ.get('LOCAL_RANK', rank)))
@@ -56,7 +58,20 @@ def stitch_to_rank0(bboxes: List[BoundingBox], local_outputs: dict,
 
     P2P matching relies on per-source ordering, not tags (RCCL ignores
     tags): both sides walk the same global task order.
+
+    Backend note: gloo moves CPU tensors only, so under gloo any device
+    sub-volumes bounce through host memory and the stitched volume
+    returns on the original device — functionally identical, used for
+    dress-rehearsing the torchrun path on boxes where RCCL cannot form
+    the communicator (e.g. two ranks on one GPU).
     """
+    gloo_cuda = (dist.is_initialized() and world_size > 1
+                 and dist.get_backend() == 'gloo'
+                 and any(t.is_cuda for t in local_outputs.values()))
+    if gloo_cuda:
+        orig_device = device
+        local_outputs = {i: t.cpu() for i, t in local_outputs.items()}
+        device = 'cpu'
     if roi_start is None:
         roi_start = Cartesian(*(min(b.start[d] for b in bboxes)
                                 for d in range(3)))
@@ -97,7 +112,7 @@ def stitch_to_rank0(bboxes: List[BoundingBox], local_outputs: dict,
             s = bboxes[i].start - roi_start
             d = bboxes[i].shape
             volume[:, s.z:s.z + d.z, s.y:s.y + d.y, s.x:s.x + d.x] = buf
-        return volume
+        return volume.to(orig_device) if gloo_cuda else volume
     else:
         reqs = []
         for i, bbox in enumerate(bboxes):
