@@ -154,6 +154,9 @@ def main():
     if not torch.cuda.is_available():
         raise RuntimeError('bench.py requires a GPU (the CPU plumbing path '
                            'is not the product path)')
+    # modulo keeps the 1:1 mapping on a full node and lets the gloo
+    # dress-rehearsal run several ranks on a single-GPU box
+    local_rank = local_rank % torch.cuda.device_count()
     torch.cuda.set_device(local_rank)
     device = f'cuda:{local_rank}'
     torch.manual_seed(0)
