@@ -327,7 +327,9 @@ def main():
             'batch_size': args.batch_size,
             'engine': args.engine,
             'chunks_per_step': world,
-            'stitch': 'rccl-p2p-gather-to-rank0' if world > 1 else None,
+            'stitch': (('rccl' if dist.get_backend() == 'nccl'
+                        else dist.get_backend())
+                       + '-p2p-gather-to-rank0') if world > 1 else None,
         },
         'roofline': roofline,
         'cpu_baseline': cpu,
