@@ -111,3 +111,57 @@ def test_h5_roundtrip_random(d0, d1, d2, d3, dtype):
         p = f'{td}/r.h5'
         h5io.write_h5(p, {'main': arr})
         np.testing.assert_array_equal(h5io.read_h5(p)['main'], arr)
+
+
+@settings(max_examples=25, deadline=None,
+          suppress_health_check=[HealthCheck.too_slow])
+@given(st.integers(1, 6), st.integers(1, 4), st.integers(0, 2**31 - 1))
+def test_stitch_placement_property(ntasks, chunk_scale, seed):
+    """Property: for ANY 1-D task decomposition, stitch_to_rank0 (world 1)
+    places every task's block exactly at its bbox."""
+    import torch
+    from chunkflow_amd.cartesian import BoundingBoxes
+    from chunkflow_amd.dispatch import stitch_to_rank0
+    rng = np.random.RandomState(seed)
+    cz, cy, cx = 2, 3, 2 * chunk_scale
+    bboxes = BoundingBoxes.from_manual_setup(
+        (cz, cy, cx), roi_size=(cz, cy, cx * ntasks))
+    blocks = {i: torch.from_numpy(
+        rng.rand(2, cz, cy, cx).astype(np.float32)) for i in range(ntasks)}
+    vol = stitch_to_rank0(bboxes, dict(blocks), 2, 0, 1, 'cpu')
+    assert vol.shape == (2, cz, cy, cx * ntasks)
+    for i in range(ntasks):
+        np.testing.assert_array_equal(
+            vol[:, :, :, i * cx:(i + 1) * cx].numpy(), blocks[i].numpy())
+
+
+@settings(max_examples=25, deadline=None,
+          suppress_health_check=[HealthCheck.too_slow])
+@given(st.integers(0, 2**31 - 1), st.integers(2, 5))
+def test_equal_value_label_properties(seed, nvals):
+    """Properties of the cc3d-semantics labeler: (1) the label partition
+    refines the value partition (no component spans two values); (2)
+    labels are dense 1..N; (3) background stays 0; (4) per-value
+    partition matches scipy run on that value alone."""
+    from scipy import ndimage
+    from chunkflow_amd.connected import equal_value_label
+    rng = np.random.RandomState(seed)
+    arr = rng.randint(0, nvals, size=(5, 9, 8)).astype(np.uint32)
+    lab = equal_value_label(arr, 6)
+    assert (lab[arr == 0] == 0).all()
+    u = np.unique(lab)
+    u = u[u != 0]
+    assert np.array_equal(u, np.arange(1, len(u) + 1))
+    struct = ndimage.generate_binary_structure(3, 1)
+    for lb in u:
+        vals = np.unique(arr[lab == lb])
+        assert len(vals) == 1 and vals[0] != 0
+    for v in np.unique(arr):
+        if v == 0:
+            continue
+        ref, n = ndimage.label(arr == v, structure=struct)
+        got_ids = lab[arr == v]
+        # same number of components and a bijection between labelings
+        assert len(np.unique(got_ids)) == n
+        pairs = {(a, b) for a, b in zip(ref[arr == v], got_ids)}
+        assert len(pairs) == n
