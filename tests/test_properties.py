@@ -165,3 +165,27 @@ def test_equal_value_label_properties(seed, nvals):
         assert len(np.unique(got_ids)) == n
         pairs = {(a, b) for a, b in zip(ref[arr == v], got_ids)}
         assert len(pairs) == n
+
+
+@settings(max_examples=20, deadline=None,
+          suppress_health_check=[HealthCheck.too_slow])
+@given(st.tuples(st.integers(3, 10), st.integers(6, 24),
+                 st.integers(6, 24)),
+       st.integers(1, 2), st.integers(1, 5), st.integers(1, 5))
+def test_patch_mask_partition_of_unity(ps, oz, oy, ox):
+    """SURVEY A.2 invariants for ANY geometry: the bump patch mask's
+    interior [ov, ps-ov) is exactly 1.0 and its SUM equals the
+    stride-cell volume prod(ps - ov) — i.e. neighboring patch weights
+    form a partition of unity."""
+    from chunkflow_amd.patch_mask import make_patch_mask
+    ov = (min(oz, ps[0] // 2 - 1) or 1, min(oy, ps[1] // 2 - 1) or 1,
+          min(ox, ps[2] // 2 - 1) or 1)
+    if any(o < 1 or 2 * o >= p for o, p in zip(ov, ps)):
+        return
+    m = make_patch_mask(tuple(ps), ov)
+    inner = m[ov[0]:ps[0] - ov[0], ov[1]:ps[1] - ov[1],
+              ov[2]:ps[2] - ov[2]]
+    np.testing.assert_array_equal(inner, 1.0)
+    want = float(np.prod([p - o for p, o in zip(ps, ov)]))
+    assert abs(float(m.sum()) - want) / want < 1e-5
+    assert float(m.min()) > 0
